@@ -1,0 +1,52 @@
+"""ResNet-18 (2D) for the FedAvg/custom-reducer config (BASELINE.json #5).
+
+Written from the ResNet paper structure (basic blocks, 4 stages
+[2,2,2,2]); no torchvision dependency.
+"""
+import torch.nn as nn
+
+
+class BasicBlock(nn.Module):
+    expansion = 1
+
+    def __init__(self, cin, cout, stride=1):
+        super().__init__()
+        self.conv1 = nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
+        self.bn1 = nn.BatchNorm2d(cout)
+        self.conv2 = nn.Conv2d(cout, cout, 3, padding=1, bias=False)
+        self.bn2 = nn.BatchNorm2d(cout)
+        self.act = nn.ReLU(inplace=True)
+        self.down = None
+        if stride != 1 or cin != cout:
+            self.down = nn.Sequential(
+                nn.Conv2d(cin, cout, 1, stride=stride, bias=False),
+                nn.BatchNorm2d(cout))
+
+    def forward(self, x):
+        idt = x if self.down is None else self.down(x)
+        y = self.act(self.bn1(self.conv1(x)))
+        y = self.bn2(self.conv2(y))
+        return self.act(y + idt)
+
+
+class ResNet18(nn.Module):
+    def __init__(self, in_channels=3, num_class=10, widths=(64, 128, 256, 512)):
+        super().__init__()
+        self.stem = nn.Sequential(
+            nn.Conv2d(in_channels, widths[0], 7, stride=2, padding=3, bias=False),
+            nn.BatchNorm2d(widths[0]), nn.ReLU(inplace=True),
+            nn.MaxPool2d(3, stride=2, padding=1))
+        layers = []
+        cin = widths[0]
+        for i, w in enumerate(widths):
+            stride = 1 if i == 0 else 2
+            layers += [BasicBlock(cin, w, stride), BasicBlock(w, w)]
+            cin = w
+        self.stages = nn.Sequential(*layers)
+        self.pool = nn.AdaptiveAvgPool2d(1)
+        self.head = nn.Linear(cin, num_class)
+
+    def forward(self, x):
+        x = self.stem(x)
+        x = self.stages(x)
+        return self.head(self.pool(x).flatten(1))

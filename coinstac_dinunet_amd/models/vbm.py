@@ -1,0 +1,45 @@
+"""VBM 3D-CNN classifier (BASELINE.json configs 3/4 — the flagship bench).
+
+The reference's GPU example workload is a 3D CNN over ~64^3 (up to
+121x145x121) brain volumes with small channel counts.
+This is our own architecture sized for that shape: conv3x3x3 stacks with
+BN/ReLU and stride-2 downsampling, global average pool, linear head.
+
+MI355X notes: channels_last_3d (NDHWC) memory format is used on GPU so the
+conv kernels (ops Conv3d implicit-GEMM / MIOpen) see coalescible layouts;
+bf16 autocast-friendly (no ops that silently upcast).
+"""
+import torch
+import torch.nn as nn
+
+
+class _ConvBlock(nn.Module):
+    def __init__(self, cin, cout, stride=1):
+        super().__init__()
+        self.conv = nn.Conv3d(cin, cout, 3, stride=stride, padding=1, bias=False)
+        self.bn = nn.BatchNorm3d(cout)
+        self.act = nn.ReLU(inplace=True)
+
+    def forward(self, x):
+        return self.act(self.bn(self.conv(x)))
+
+
+class VBMNet(nn.Module):
+    """~9-layer 3D CNN: 1 -> widths[0] -> ... with stride-2 stages."""
+
+    def __init__(self, in_channels=1, num_class=2, widths=(32, 64, 128, 256)):
+        super().__init__()
+        stages = []
+        cin = in_channels
+        for i, w in enumerate(widths):
+            stages.append(_ConvBlock(cin, w, stride=1 if i == 0 else 2))
+            stages.append(_ConvBlock(w, w))
+            cin = w
+        self.features = nn.Sequential(*stages)
+        self.pool = nn.AdaptiveAvgPool3d(1)
+        self.head = nn.Linear(cin, num_class)
+
+    def forward(self, x):
+        x = self.features(x)
+        x = self.pool(x).flatten(1)
+        return self.head(x)

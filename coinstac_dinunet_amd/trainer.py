@@ -1,0 +1,85 @@
+"""COINSTAC-aware trainer: distributed validation/test + metric factory.
+
+API-parity: /root/reference/coinstac_dinunet/trainer.py:15-80 (COINNTrainer).
+Note the reference keeps nn/device/optimizer dicts inside cache so the
+stateless per-iteration invocations can find them again (trainer.py:17-20);
+we keep the same protocol so the loopback path and reference user code work,
+while the persistent RCCL engine simply holds one trainer per rank process.
+"""
+from abc import ABC as _ABC
+from os import sep as _sep
+
+from . import config as _conf
+from . import metrics as _metrics
+from .config.keys import Key, Mode
+from .nn.basetrainer import NNTrainer as _NNTrainer
+from .utils.utils import performance_improved_
+
+
+class COINNTrainer(_NNTrainer, _ABC):
+    def __init__(self, **kw):
+        super().__init__(**kw)
+        self.nn = self.cache.setdefault('nn', self.nn)
+        self.device = self.cache.setdefault('device', self.device)
+        self.optimizer = self.cache.setdefault('optimizer', self.optimizer)
+
+    def _save_if_better(self, epoch, val_metrics):
+        """During pretraining the improving weights ship to transferDirectory
+        as weights.tar so the remote can relay them to every site."""
+        out = {}
+        val_score = val_metrics.extract(self.cache['monitor_metric'])
+        if performance_improved_(epoch, val_score, self.cache):
+            out['weights_file'] = _conf.weights_file
+            self.save_checkpoint(
+                file_path=self.state['transferDirectory'] + _sep + out['weights_file'])
+        return out
+
+    def validation_distributed(self, dataset_cls=None):
+        out = {}
+        validation_dataset = self.data_handle.dataset.get('validation')
+        if validation_dataset and not isinstance(validation_dataset, list):
+            validation_dataset = [validation_dataset]
+        if validation_dataset:
+            avg, metrics = self.evaluation(mode='validation', save_pred=False,
+                                           dataset_list=validation_dataset,
+                                           use_padded_sampler=True)
+            out[Key.VALIDATION_SERIALIZABLE] = [{'averages': avg.serialize(),
+                                                 'metrics': metrics.serialize()}]
+        self.cache['cursor'] = 0
+        return out
+
+    def test_distributed(self, dataset_cls=None):
+        out = {}
+        import os as _os
+        best = self.cache['log_dir'] + _sep + self.cache['best_nn_state']
+        # Deviation from the reference (trainer.py:52): when validation never
+        # improved, no best checkpoint exists — test on current weights
+        # instead of crashing.
+        if _os.path.exists(best):
+            self.load_checkpoint(best)
+        test_dataset = self.data_handle.get_test_dataset(dataset_cls)
+        if test_dataset and not isinstance(test_dataset, list):
+            test_dataset = [test_dataset]
+        if test_dataset:
+            avg, metrics = self.evaluation(mode='test', save_pred=True,
+                                           dataset_list=test_dataset)
+            out[Key.TEST_SERIALIZABLE] = [{'averages': avg.serialize(),
+                                           'metrics': metrics.serialize()}]
+        return out
+
+    def set_monitor_metric(self):
+        """Set from COINNLocal's constructor."""
+
+    def set_log_headers(self):
+        """Set from COINNLocal's constructor."""
+
+    def new_metrics(self):
+        if self.cache.get('num_class') == 2:
+            if self.cache.get('monitor_metric') in ['precision', 'recall',
+                                                    'accuracy', 'overlap', 'f1']:
+                return _metrics.Prf1a()
+            if self.cache.get('monitor_metric') == 'auc':
+                return _metrics.AUCROCMetrics()
+        elif (self.cache.get('num_class') or 0) > 2:
+            return _metrics.ConfusionMatrix(num_classes=self.cache['num_class'])
+        return _metrics.Prf1a()
