@@ -1,0 +1,128 @@
+"""RcclCluster: the COINSTAC phase machine on a persistent process group.
+
+One process per GPU "site" (torchrun); rank 0 doubles as the remote
+aggregator. Per round:
+  1. every rank runs COINNLocal.compute with learner_cls=RcclLearner
+     (gradient rounds are RCCL all-reduce over xGMI — no files);
+  2. the tiny control dicts are gathered to rank 0 (gloo object gather);
+  3. rank 0 runs COINNRemote with reducer_cls=RcclReducer (pure control:
+     epoch/mode/fold transitions, score reduction, best-model signal);
+  4. rank 0 relays any cold-path artifacts (pretrained weights.tar,
+     results zip) over the node-local filesystem and broadcasts the
+     remote out-dict, which becomes every rank's next input.
+
+The hot path therefore exchanges ONLY the fused gradient bucket per round;
+control costs one object gather+broadcast of a few KB.
+"""
+import os
+import shutil
+
+import torch.distributed as dist
+
+from ..distrib.nodes.local import COINNLocal
+from ..distrib.nodes.remote import COINNRemote
+from .engine import RcclLearner, RcclReducer, init_distributed
+
+
+class RankSiteState:
+    """Per-rank site directories under one shared (node-local) root."""
+
+    def __init__(self, root, rank):
+        self.clientId = f'rank{rank}'
+        base = os.path.join(root, self.clientId)
+        self.baseDirectory = os.path.join(base, 'input')
+        self.transferDirectory = os.path.join(base, 'transfer')
+        self.outputDirectory = os.path.join(base, 'output')
+        for d in (self.baseDirectory, self.transferDirectory,
+                  self.outputDirectory):
+            os.makedirs(d, exist_ok=True)
+
+    def as_dict(self):
+        return {'clientId': self.clientId,
+                'baseDirectory': self.baseDirectory,
+                'transferDirectory': self.transferDirectory,
+                'outputDirectory': self.outputDirectory}
+
+
+class RcclCluster:
+    def __init__(self, root, local_kw=None, remote_kw=None):
+        self.rank, self.world_size = init_distributed()
+        self.root = root
+        self.site = RankSiteState(root, self.rank)
+        self.remote_state = {
+            'clientId': 'remote',
+            'baseDirectory': os.path.join(root, 'remote', 'input'),
+            'transferDirectory': os.path.join(root, 'remote', 'transfer'),
+            'outputDirectory': os.path.join(root, 'remote', 'output')}
+        if self.rank == 0:
+            for d in self.remote_state.values():
+                if d != 'remote':
+                    os.makedirs(d, exist_ok=True)
+        self.local_kw = local_kw or {}
+        self.remote_kw = remote_kw or {}
+        self.site_cache = {}
+        self.remote_cache = {}
+        self.input = {}
+        self.rounds = 0
+
+    # -- filesystem relay (cold path only) --------------------------------
+    def _site_transfer_to_remote(self, site_outs):
+        for site_id in site_outs:
+            src = os.path.join(self.root, site_id, 'transfer')
+            dst = os.path.join(self.remote_state['baseDirectory'], site_id)
+            os.makedirs(dst, exist_ok=True)
+            for name in os.listdir(src):
+                s, d = os.path.join(src, name), os.path.join(dst, name)
+                if os.path.isdir(d):
+                    shutil.rmtree(d)
+                elif os.path.exists(d):
+                    os.remove(d)
+                shutil.move(s, d)
+
+    def _remote_transfer_to_sites(self, site_outs):
+        src = self.remote_state['transferDirectory']
+        for name in os.listdir(src):
+            for site_id in site_outs:
+                dst_dir = os.path.join(self.root, site_id, 'input')
+                s, d = os.path.join(src, name), os.path.join(dst_dir, name)
+                if os.path.isfile(s):
+                    shutil.copy(s, d)
+            p = os.path.join(src, name)
+            shutil.rmtree(p) if os.path.isdir(p) else os.remove(p)
+
+    # -- one protocol round -------------------------------------------------
+    def run_round(self, trainer_cls, dataset_cls=None, mp_pool=None, **kw):
+        local = COINNLocal(cache=self.site_cache, input=dict(self.input),
+                           state=self.site.as_dict(), **self.local_kw)
+        result = local(mp_pool, trainer_cls, dataset_cls=dataset_cls,
+                       learner_cls=RcclLearner, **kw)
+        my_out = result['output']
+
+        gathered = [None] * self.world_size
+        dist.all_gather_object(gathered, (self.site.clientId, my_out))
+        site_outs = dict(gathered)
+
+        bcast = [None]
+        success = False
+        if self.rank == 0:
+            self._site_transfer_to_remote(site_outs)
+            remote = COINNRemote(cache=self.remote_cache, input=site_outs,
+                                 state=self.remote_state, **self.remote_kw)
+            rres = remote(mp_pool, trainer_cls, reducer_cls=RcclReducer)
+            self._remote_transfer_to_sites(site_outs)
+            bcast = [(rres['output'], rres.get('success', False))]
+        dist.broadcast_object_list(bcast, src=0)
+        remote_out, success = bcast[0]
+        self.input = dict(remote_out)
+        self.rounds += 1
+        return success, remote_out
+
+    def run(self, trainer_cls, dataset_cls=None, mp_pool=None,
+            max_rounds=10000, **kw):
+        success, out = False, {}
+        for _ in range(max_rounds):
+            success, out = self.run_round(trainer_cls, dataset_cls=dataset_cls,
+                                          mp_pool=mp_pool, **kw)
+            if success:
+                break
+        return success, out

@@ -1,0 +1,248 @@
+"""MI355X-native distributed engine: one process per GPU "site", RCCL over xGMI.
+
+This replaces the reference's file/JSON relay (learner.py:49-59 ->
+reducer.py:43-54 -> learner.py:20-30 round trip through the COINSTAC
+simulator) with collectives on the persistent process group:
+
+  - gradient exchange: ONE preallocated flat fp32 bucket per model; every
+    param.grad is a VIEW into it, so autograd itself does the packing
+    (K5/K7 of SURVEY.md §2.9 with zero kernels and zero copies). The
+    buffer is split into size-targeted buckets all-reduced (avg) as their
+    last gradient lands in backward — comm overlaps the rest of backward
+    on a side HIP stream, sized for xGMI's per-link ring bound
+    (~153 GB/s/link => default 50 MB buckets).
+  - control plane: tiny per-round out/input dicts exchanged with
+    all_gather_object / broadcast_object_list on a gloo sub-group (CPU,
+    never blocks the compute stream).
+  - weights relay / results artifacts: the node-local shared filesystem
+    (all 8 "sites" live on one MI355X node).
+
+Backend is "nccl" (=RCCL on ROCm) on GPU, "gloo" on CPU — the same code
+path is exercised by the world_size=2 CPU tests.
+"""
+import datetime as _dt
+import os
+
+import torch
+import torch.distributed as dist
+
+from ..config.keys import Mode
+from ..distrib.learner import COINNLearner
+from ..distrib.reducer import COINNReducer
+
+# xGMI: 7 p2p links x ~153 GB/s; ring all-reduce is per-link bound. Big
+# buckets amortize launch + ring latency; 50 MB keeps >= 4 buckets in
+# flight for typical CNNs so comm still overlaps backward.
+DEFAULT_BUCKET_BYTES = 50 * 1024 * 1024
+
+
+def init_distributed(backend=None, timeout_s=300):
+    """Initialize the process group from torchrun env (idempotent)."""
+    if dist.is_initialized():
+        return dist.get_rank(), dist.get_world_size()
+    if 'RANK' not in os.environ:
+        os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+        os.environ.setdefault('MASTER_PORT', '29671')
+        os.environ.setdefault('RANK', '0')
+        os.environ.setdefault('WORLD_SIZE', '1')
+    if backend is None:
+        backend = 'nccl' if torch.cuda.is_available() else 'gloo'
+    if backend == 'nccl':
+        torch.cuda.set_device(int(os.environ.get('LOCAL_RANK', 0)))
+    dist.init_process_group(backend=backend,
+                            timeout=_dt.timedelta(seconds=timeout_s))
+    return dist.get_rank(), dist.get_world_size()
+
+
+class FlatGradBuffer:
+    """One contiguous fp32 gradient arena; param.grad are views into it.
+
+    Autograd accumulates straight into the arena (packing for free); the
+    arena is carved into ~bucket_bytes buckets in REVERSE parameter order
+    (the order backward produces grads), each all-reduced as soon as its
+    last grad lands.
+    """
+
+    def __init__(self, params, bucket_bytes=DEFAULT_BUCKET_BYTES,
+                 world_size=1, comm_stream=None):
+        self.params = [p for p in params if p.requires_grad]
+        self.world_size = world_size
+        self.comm_stream = comm_stream
+        device = self.params[0].device if self.params else torch.device('cpu')
+        total = sum(p.numel() for p in self.params)
+        self.flat = torch.zeros(total, dtype=torch.float32, device=device)
+
+        # Assign views in reverse order so bucket 0 = last-produced grads.
+        self.views = {}
+        offset = 0
+        rev = list(reversed(self.params))
+        for p in rev:
+            n = p.numel()
+            self.views[p] = self.flat.narrow(0, offset, n).view_as(p)
+            offset += n
+
+        # Build buckets over the same reverse order.
+        self.buckets = []  # (start, numel, [params])
+        start, numel, members = 0, 0, []
+        for p in rev:
+            members.append(p)
+            numel += p.numel()
+            if numel * 4 >= bucket_bytes:
+                self.buckets.append((start, numel, members))
+                start, numel, members = start + numel, 0, []
+        if members:
+            self.buckets.append((start, numel, members))
+        self._bucket_of = {}
+        for bi, (_, _, members) in enumerate(self.buckets):
+            for p in members:
+                self._bucket_of[p] = bi
+
+        self._pending = [0] * len(self.buckets)
+        self._works = [None] * len(self.buckets)
+        self._hooks = []
+        self._install_grads()
+        self._install_hooks()
+
+    # -- wiring ----------------------------------------------------------
+    def _install_grads(self):
+        for p in self.params:
+            p.grad = self.views[p]
+
+    def _install_hooks(self):
+        for p in self.params:
+            h = p.register_post_accumulate_grad_hook(self._on_grad_ready)
+            self._hooks.append(h)
+
+    def remove_hooks(self):
+        for h in self._hooks:
+            h.remove()
+        self._hooks = []
+
+    # -- round lifecycle ---------------------------------------------------
+    def zero_(self):
+        self.flat.zero_()
+        for p in self.params:
+            if p.grad is None or p.grad.data_ptr() != self.views[p].data_ptr():
+                p.grad = self.views[p]  # re-pin if an optimizer detached it
+
+    def begin_round(self, sync=True):
+        """Arm the per-bucket countdowns for one backward pass.
+
+        sync=False (grad-accumulation micro-batches before the last) leaves
+        the hooks dormant so no communication happens.
+        """
+        self._sync_this_round = sync
+        self._finished = False
+        for bi, (_, _, members) in enumerate(self.buckets):
+            self._pending[bi] = len(members)
+            self._works[bi] = None
+
+    def _on_grad_ready(self, p):
+        if not getattr(self, '_sync_this_round', False):
+            return
+        if self.world_size <= 1 or not dist.is_initialized():
+            return
+        bi = self._bucket_of[p]
+        self._pending[bi] -= 1
+        if self._pending[bi] == 0:
+            start, numel, _ = self.buckets[bi]
+            piece = self.flat.narrow(0, start, numel)
+            if self.comm_stream is not None:
+                self.comm_stream.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(self.comm_stream):
+                    self._works[bi] = dist.all_reduce(piece, async_op=True)
+            else:
+                self._works[bi] = dist.all_reduce(piece, async_op=True)
+
+    def finish_round(self):
+        """Wait for every in-flight all-reduce, then divide by world size."""
+        if self.world_size <= 1 or not dist.is_initialized():
+            return
+        if getattr(self, '_finished', True):
+            return
+        self._finished = True
+        launched = False
+        for bi, w in enumerate(self._works):
+            if w is None and self._pending[bi] == len(self.buckets[bi][2]):
+                continue  # bucket never armed (no grads this round)
+            if w is None:
+                # stragglers (params that produced no grad): reduce now
+                start, numel, _ = self.buckets[bi]
+                piece = self.flat.narrow(0, start, numel)
+                w = dist.all_reduce(piece, async_op=True)
+            w.wait()
+            launched = True
+        if self.comm_stream is not None:
+            torch.cuda.current_stream().wait_stream(self.comm_stream)
+        if launched:
+            self.flat.div_(self.world_size)
+
+    def allreduce_now(self):
+        """One-shot whole-arena all-reduce(avg) (no-overlap fallback)."""
+        if self.world_size <= 1 or not dist.is_initialized():
+            return
+        dist.all_reduce(self.flat)
+        self.flat.div_(self.world_size)
+
+
+class RcclLearner(COINNLearner):
+    """dSGD learner whose reduce round is an RCCL all-reduce over xGMI.
+
+    Keeps COINNLearner's (it, out) API so COINNLocal drives it unchanged;
+    to_reduce/step never touch the filesystem.
+    """
+
+    def __init__(self, trainer=None, mp_pool=None, **kw):
+        super().__init__(trainer=trainer, mp_pool=mp_pool, **kw)
+        self.world_size = dist.get_world_size() if dist.is_initialized() else 1
+        model = self.trainer.nn[self.first_model]
+        key = '_rccl_grad_buffer'
+        if key not in self.cache or self.cache[key].params[0] is not \
+                next(iter(model.parameters())):
+            comm_stream = torch.cuda.Stream() \
+                if self.device.type == 'cuda' else None
+            self.cache[key] = FlatGradBuffer(
+                model.parameters(),
+                bucket_bytes=self.cache.get('bucket_bytes',
+                                            DEFAULT_BUCKET_BYTES),
+                world_size=self.world_size, comm_stream=comm_stream)
+        self.grad_buffer = self.cache[key]
+
+    def backward(self):
+        out = {}
+        self.trainer.nn[self.first_model].train()
+        self.grad_buffer.zero_()
+        its = []
+        local_iters = self.cache.get('local_iterations', 1)
+        for li in range(local_iters):
+            self.grad_buffer.begin_round(sync=(li == local_iters - 1))
+            batch, nxt_iter_out = self.trainer.data_handle.next_iter()
+            it = self.trainer.iteration(batch)
+            it['loss'].backward()
+            its.append(it)
+            out.update(**nxt_iter_out)
+        return self.trainer.reduce_iteration(its), out
+
+    def to_reduce(self):
+        it, out = self.backward()
+        # comm was launched bucket-by-bucket inside backward; the round is
+        # completed (waited + averaged) in step(), next round — preserving
+        # the reference's deferred-update semantics (local.py:229-239).
+        out['reduce'] = True
+        return it, out
+
+    def step(self):
+        # Note: like the reference (learner.py:32-47), grads are the SUM over
+        # local_iterations micro-batches, averaged over sites only.
+        out = {}
+        self.grad_buffer.finish_round()
+        self.trainer.optimizer[self.first_optim].step()
+        return out
+
+
+class RcclReducer(COINNReducer):
+    """The all-reduce already produced the average on every rank: the
+    remote role only flips the update flag."""
+
+    def reduce(self):
+        return {'update': True}
