@@ -1,0 +1,143 @@
+"""Flagship benchmark: VBM 3D-CNN dSGD training step (BASELINE.json).
+
+Measures whole-node samples/sec for the decentralized-SGD training round:
+per step = local fwd/bwd on synthetic 64^3 volumes (bf16 autocast) +
+bucketed RCCL all-reduce(avg) of the flat gradient arena overlapped with
+backward + fused-Adam update. One process per GPU "site" (weak scaling:
+per-GPU batch fixed as N grows).
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`;
+launched via torch.distributed.run for N>1. Rank 0 prints ONE JSON line.
+"""
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument('--gpus', type=int, default=1)
+    ap.add_argument('--steps', type=int, default=20)
+    ap.add_argument('--warmup', type=int, default=5)
+    ap.add_argument('--batch', type=int, default=16,
+                    help='per-GPU (per-site) batch size')
+    ap.add_argument('--vol', type=int, default=64, help='volume side length')
+    ap.add_argument('--local-iterations', type=int, default=1)
+    ap.add_argument('--model', type=str, default='vbm',
+                    choices=['vbm', 'mlp', 'resnet18'])
+    return ap.parse_args()
+
+
+def build_model(args, device):
+    from coinstac_dinunet_amd.models import FreeSurferMLP, ResNet18, VBMNet
+    if args.model == 'vbm':
+        net = VBMNet(in_channels=1, num_class=2)
+        data = torch.randn(args.batch, 1, args.vol, args.vol, args.vol)
+    elif args.model == 'mlp':
+        net = FreeSurferMLP(in_features=66, num_class=2)
+        data = torch.randn(args.batch, 66)
+    else:
+        net = ResNet18(in_channels=3, num_class=10)
+        data = torch.randn(args.batch, 3, 224, 224)
+    labels = torch.randint(0, 2, (args.batch,))
+    return net.to(device), data.to(device), labels.to(device)
+
+
+def main():
+    args = parse_args()
+    from coinstac_dinunet_amd import ops
+    from coinstac_dinunet_amd.parallel.engine import (FlatGradBuffer,
+                                                      init_distributed)
+
+    rank, world = init_distributed()
+    on_gpu = torch.cuda.is_available()
+    device = torch.device(f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}"
+                          if on_gpu else 'cpu')
+    if on_gpu:
+        torch.cuda.set_device(device)
+        torch.backends.cudnn.benchmark = True
+
+    net, data, labels = build_model(args, device)
+    comm_stream = torch.cuda.Stream() if on_gpu else None
+    buf = FlatGradBuffer(net.parameters(), world_size=world,
+                         comm_stream=comm_stream)
+    if on_gpu and ops.native_available():
+        opt = ops.FusedAdam(net.parameters(), lr=1e-3)
+    else:
+        opt = torch.optim.Adam(net.parameters(), lr=1e-3)
+
+    import torch.distributed as dist
+
+    def one_step():
+        buf.zero_()
+        for li in range(args.local_iterations):
+            buf.begin_round(sync=(li == args.local_iterations - 1))
+            if on_gpu:
+                with torch.autocast('cuda', dtype=torch.bfloat16):
+                    out = net(data)
+                loss = ops.cross_entropy(out.float(), labels)
+            else:
+                loss = ops.cross_entropy(net(data), labels)
+            loss.backward()
+        buf.finish_round()
+        opt.step()
+
+    # warmup
+    for _ in range(args.warmup):
+        one_step()
+
+    if dist.is_initialized():
+        dist.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    if on_gpu:
+        torch.cuda.synchronize()
+    if dist.is_initialized():
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks (slowest rank defines the lock-step round time)
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if dist.is_initialized() and world > 1:
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    samples_per_sec = args.batch * world * args.steps / elapsed
+
+    if rank == 0:
+        print(json.dumps({
+            'metric': 'samples/sec (whole node), VBM 3D-CNN dSGD'
+                      if args.model == 'vbm' else
+                      f'samples/sec (whole node), {args.model} dSGD',
+            'value': round(samples_per_sec, 3),
+            'unit': 'samples/sec',
+            'n_gpus': world,
+            'steps': args.steps,
+            'warmup': args.warmup,
+            'ms_per_step': round(ms_per_step, 3),
+            'higher_is_better': True,
+            'scaling': 'weak',
+            'vs_baseline': None,
+            'dtype': 'bf16' if on_gpu else 'fp32',
+            'data': 'synthetic',
+            'config': {
+                'model': args.model,
+                'global_batch': args.batch * world,
+                'volume': f'{args.vol}^3' if args.model == 'vbm' else None,
+                'local_iterations': args.local_iterations,
+                'parallelism': f'dsgd-dp{world}',
+            },
+        }))
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == '__main__':
+    main()

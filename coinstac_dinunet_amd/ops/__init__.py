@@ -156,13 +156,16 @@ class _FusedLogSoftmaxNLL(torch.autograd.Function):
         C = require_native()
         loss, logprobs = C.logsoftmax_nll_fwd(logits, target)
         ctx.save_for_backward(logprobs, target)
+        ctx.in_dtype = logits.dtype
         return loss
 
     @staticmethod
     def backward(ctx, grad_out):
         C = require_native()
         logprobs, target = ctx.saved_tensors
-        grad_logits = C.logsoftmax_nll_bwd(logprobs, target, grad_out)
+        grad_logits = C.logsoftmax_nll_bwd(logprobs, target,
+                                           grad_out.contiguous(),
+                                           ctx.in_dtype)
         return grad_logits, None
 
 
@@ -197,26 +200,36 @@ def confusion_matrix(pred, true, num_classes):
 
 # ---- K2: MFMA GEMM (linear) -------------------------------------------------
 class _LinearFn(torch.autograd.Function):
+    """MFMA forward (fused bias/ReLU); backward = rocBLAS GEMMs (plain
+    library GEMMs) + the fused colsum kernel for the bias grad."""
+
     @staticmethod
-    def forward(ctx, x, weight, bias):
+    def forward(ctx, x, weight, bias, relu):
         C = require_native()
-        out = C.linear_fwd(x, weight, bias if bias is not None else
-                           torch.empty(0, device=x.device, dtype=x.dtype))
-        ctx.save_for_backward(x, weight)
+        empty = torch.empty(0, device=x.device, dtype=x.dtype)
+        out = C.linear_fwd(x, weight, bias if bias is not None else empty,
+                           relu)
+        ctx.save_for_backward(x, weight, out if relu else empty)
         ctx.has_bias = bias is not None
+        ctx.relu = relu
         return out
 
     @staticmethod
     def backward(ctx, grad_out):
         C = require_native()
-        x, weight = ctx.saved_tensors
+        x, weight, out = ctx.saved_tensors
         grad_out = grad_out.contiguous()
-        gx, gw, gb = C.linear_bwd(grad_out, x, weight, ctx.has_bias)
-        return gx, gw, (gb if ctx.has_bias else None)
+        if ctx.relu:
+            grad_out = grad_out * (out > 0).to(grad_out.dtype)
+        gx = grad_out @ weight            # rocBLAS
+        gw = grad_out.transpose(0, 1) @ x  # rocBLAS
+        gb = C.colsum(grad_out).to(grad_out.dtype) if ctx.has_bias else None
+        return gx, gw, gb, None
 
 
-def linear(x, weight, bias=None):
-    """MFMA-tiled linear: x[M,K] @ weight[N,K]^T + bias."""
+def linear(x, weight, bias=None, relu=False):
+    """MFMA-tiled linear: x[M,K] @ weight[N,K]^T + bias (+ReLU fused)."""
     if x.is_cuda and native_available():
-        return _LinearFn.apply(x, weight, bias)
-    return torch.nn.functional.linear(x, weight, bias)
+        return _LinearFn.apply(x, weight, bias, relu)
+    out = torch.nn.functional.linear(x, weight, bias)
+    return torch.nn.functional.relu(out) if relu else out

@@ -1,0 +1,52 @@
+// pybind bindings for the gfx950 kernel library (_hip_ops).
+#include <torch/extension.h>
+
+#include <vector>
+
+// adam.hip
+void fused_adam_flat(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                     torch::Tensor v, double lr, double beta1, double beta2,
+                     double eps, double weight_decay, int64_t step);
+void fused_adam(std::vector<torch::Tensor> params,
+                std::vector<torch::Tensor> grads,
+                std::vector<torch::Tensor> exp_avgs,
+                std::vector<torch::Tensor> exp_avg_sqs, double lr,
+                double beta1, double beta2, double eps, double weight_decay,
+                int64_t step);
+void fused_sgd(std::vector<torch::Tensor> params,
+               std::vector<torch::Tensor> grads,
+               std::vector<torch::Tensor> bufs, double lr, double momentum,
+               double weight_decay);
+// pack.hip
+void pack_tensors(std::vector<torch::Tensor> tensors, torch::Tensor flat);
+void unpack_tensors(torch::Tensor flat, std::vector<torch::Tensor> tensors);
+// lsnll.hip
+std::vector<torch::Tensor> logsoftmax_nll_fwd(torch::Tensor logits,
+                                              torch::Tensor target);
+torch::Tensor logsoftmax_nll_bwd(torch::Tensor logprobs, torch::Tensor target,
+                                 torch::Tensor grad_out,
+                                 torch::ScalarType out_dtype);
+torch::Tensor argmax_rows(torch::Tensor x);
+// metrics.hip
+torch::Tensor prf1a_counts(torch::Tensor pred, torch::Tensor true_);
+torch::Tensor confusion_matrix(torch::Tensor pred, torch::Tensor true_,
+                               int64_t num_classes);
+// linear.hip
+torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor weight,
+                         torch::Tensor bias, bool relu);
+torch::Tensor colsum(torch::Tensor g);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fused_adam_flat", &fused_adam_flat, "single-kernel Adam on flat arenas");
+  m.def("fused_adam", &fused_adam, "per-tensor fused Adam");
+  m.def("fused_sgd", &fused_sgd, "per-tensor fused SGD");
+  m.def("pack_tensors", &pack_tensors, "pack tensor list into flat buffer");
+  m.def("unpack_tensors", &unpack_tensors, "scatter flat buffer into tensors");
+  m.def("logsoftmax_nll_fwd", &logsoftmax_nll_fwd);
+  m.def("logsoftmax_nll_bwd", &logsoftmax_nll_bwd);
+  m.def("argmax_rows", &argmax_rows);
+  m.def("prf1a_counts", &prf1a_counts);
+  m.def("confusion_matrix", &confusion_matrix);
+  m.def("linear_fwd", &linear_fwd);
+  m.def("colsum", &colsum);
+}

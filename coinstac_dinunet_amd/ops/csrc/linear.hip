@@ -1,0 +1,184 @@
+// K2 — MFMA-tiled linear with fused bias(+ReLU) epilogue.
+// The MLP family's hidden layers run linear -> bias -> ReLU as one kernel
+// (3 launches + 2 extra HBM round trips in stock eager). GEMM-shaped work
+// belongs on the matrix cores: this uses v_mfma_f32_16x16x4_f32 (exact
+// f32 at the 157 TF vector-peak rate; fragment layout per
+// cdna_hip_programming.md §3: A[l&15][l>>4], B[l>>4][l&15],
+// C col=l&15, row=(l>>4)*4+reg).
+//
+// Geometry: 256-thread block = 4 waves as 2x2, each wave one 32x32 C tile
+// via 2x2 mfma fragments => 64x64 block tile; A/B staged through LDS with
+// +1-float row padding (no bank conflicts); K stepped by 16.
+// Inputs may be fp32 or bf16 (upconverted on load); accumulate is f32.
+#include "common.h"
+
+#include <hip/hip_bf16.h>
+
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+#define BM 64
+#define BN 64
+#define BK 16
+#define LDK (BK + 1)  // +1 pad: column reads hit distinct banks
+
+template <typename T>
+__device__ inline float ldf(const T* p) { return (float)*p; }
+template <>
+__device__ inline float ldf<__hip_bfloat16>(const __hip_bfloat16* p) {
+  return __bfloat162float(*p);
+}
+
+// C[M,N] = A[M,K] (row-major, lda) @ B[N,K]^T (row-major, ldb = weight
+// layout) + bias; optional ReLU. One wave computes 32x32 via 2x2 fragments.
+template <typename TA, typename TB, bool RELU, bool HAS_BIAS>
+__global__ __launch_bounds__(256) void linear_fwd_kernel(
+    const TA* __restrict__ A, const TB* __restrict__ B,
+    const float* __restrict__ bias, float* __restrict__ C, int M, int N,
+    int K, int lda, int ldb, int ldc) {
+  __shared__ float sA[BM][LDK];
+  __shared__ float sB[BN][LDK];
+
+  const int bm = blockIdx.x * BM;
+  const int bn = blockIdx.y * BN;
+  const int tid = threadIdx.x;
+  const int wave = tid / WAVE_SIZE;   // 0..3
+  const int lane = tid % WAVE_SIZE;
+  const int wm = (wave >> 1) * 32;    // wave row offset in block tile
+  const int wn = (wave & 1) * 32;
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // stage A[BM][BK] and B[BN][BK]: 256 threads x 4 elements each
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int idx = tid + r * 256;          // 0..1023 = BM*BK
+      int row = idx / BK, col = idx % BK;
+      int gm = bm + row, gk = k0 + col;
+      sA[row][col] = (gm < M && gk < K) ? ldf(&A[(int64_t)gm * lda + gk]) : 0.f;
+      int gn = bn + row;
+      sB[row][col] = (gn < N && gk < K) ? ldf(&B[(int64_t)gn * ldb + gk]) : 0.f;
+    }
+    __syncthreads();
+
+    // mfma_f32_16x16x4: lane l feeds A[l&15][l>>4], B[l>>4][l&15]
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 4) {
+      const int ar = lane & 15, ak = lane >> 4;  // ak in 0..3
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+          float a = sA[wm + i * 16 + ar][kk + ak];
+          float b = sB[wn + j * 16 + ar][kk + ak];  // B^T: row=n, col=k
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[i][j],
+                                                           0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: C row=(l>>4)*4+reg, col=l&15
+  const int ccol = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int gm = bm + wm + i * 16 + crow0 + r;
+        int gn = bn + wn + j * 16 + ccol;
+        if (gm < M && gn < N) {
+          float v = acc[i][j][r];
+          if (HAS_BIAS) v += bias[gn];
+          if (RELU) v = fmaxf(v, 0.f);
+          C[(int64_t)gm * ldc + gn] = v;
+        }
+      }
+    }
+  }
+}
+
+template <typename TA, typename TB>
+static void launch_linear(const TA* A, const TB* B, const float* bias,
+                          float* C, int M, int N, int K, int lda, int ldb,
+                          int ldc, bool relu, hipStream_t stream) {
+  dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
+  dim3 block(256);
+  if (bias) {
+    if (relu)
+      hipLaunchKernelGGL((linear_fwd_kernel<TA, TB, true, true>), grid, block,
+                         0, stream, A, B, bias, C, M, N, K, lda, ldb, ldc);
+    else
+      hipLaunchKernelGGL((linear_fwd_kernel<TA, TB, false, true>), grid,
+                         block, 0, stream, A, B, bias, C, M, N, K, lda, ldb,
+                         ldc);
+  } else {
+    if (relu)
+      hipLaunchKernelGGL((linear_fwd_kernel<TA, TB, true, false>), grid,
+                         block, 0, stream, A, B, nullptr, C, M, N, K, lda,
+                         ldb, ldc);
+    else
+      hipLaunchKernelGGL((linear_fwd_kernel<TA, TB, false, false>), grid,
+                         block, 0, stream, A, B, nullptr, C, M, N, K, lda,
+                         ldb, ldc);
+  }
+}
+
+torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor weight,
+                         torch::Tensor bias, bool relu) {
+  CHECK_GPU(x);
+  auto x2 = x.contiguous();
+  auto w = weight.contiguous();
+  bool has_bias = bias.numel() > 0;
+  auto b = has_bias ? bias.to(torch::kFloat32).contiguous() : bias;
+  int64_t M = x2.size(0), K = x2.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "weight/input K mismatch");
+  auto out = torch::empty({M, N}, x2.options().dtype(torch::kFloat32));
+  auto stream = current_stream();
+  const float* bp = has_bias ? b.data_ptr<float>() : nullptr;
+
+  if (x2.scalar_type() == torch::kFloat32 &&
+      w.scalar_type() == torch::kFloat32) {
+    launch_linear(x2.data_ptr<float>(), w.data_ptr<float>(), bp,
+                  out.data_ptr<float>(), (int)M, (int)N, (int)K, (int)K,
+                  (int)K, (int)N, relu, stream);
+  } else if (x2.scalar_type() == torch::kBFloat16 &&
+             w.scalar_type() == torch::kBFloat16) {
+    launch_linear(reinterpret_cast<const __hip_bfloat16*>(x2.data_ptr()),
+                  reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()), bp,
+                  out.data_ptr<float>(), (int)M, (int)N, (int)K, (int)K,
+                  (int)K, (int)N, relu, stream);
+  } else {
+    TORCH_CHECK(false, "linear_fwd: unsupported dtype combination");
+  }
+  return out.to(x2.scalar_type());
+}
+
+// bias gradient: column sum of grad_out [M,N] -> [N]
+__global__ void colsum_kernel(const float* __restrict__ g,
+                              float* __restrict__ out, int64_t M, int64_t N) {
+  for (int64_t n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
+       n += (int64_t)gridDim.x * blockDim.x) {
+    float s = 0.f;
+    for (int64_t m = 0; m < M; ++m) s += g[m * N + n];
+    out[n] = s;
+  }
+}
+
+torch::Tensor colsum(torch::Tensor g) {
+  CHECK_GPU(g);
+  auto gc = g.to(torch::kFloat32).contiguous();
+  int64_t M = gc.size(0), N = gc.size(1);
+  auto out = torch::empty({N}, gc.options());
+  hipLaunchKernelGGL(colsum_kernel, dim3(elem_grid(N, 1)), dim3(ELEM_BLOCK),
+                     0, current_stream(), gc.data_ptr<float>(),
+                     out.data_ptr<float>(), M, N);
+  return out;
+}
