@@ -12,11 +12,13 @@ bf16 autocast-friendly (no ops that silently upcast).
 import torch
 import torch.nn as nn
 
+from ..ops.conv import OpsConv3d
+
 
 class _ConvBlock(nn.Module):
     def __init__(self, cin, cout, stride=1):
         super().__init__()
-        self.conv = nn.Conv3d(cin, cout, 3, stride=stride, padding=1, bias=False)
+        self.conv = OpsConv3d(cin, cout, 3, stride=stride, padding=1, bias=False)
         self.bn = nn.BatchNorm3d(cout)
         self.act = nn.ReLU(inplace=True)
 

@@ -1,0 +1,68 @@
+"""Conv3d module on the hand-written implicit-GEMM MFMA kernels (K1).
+
+GPU path (bf16): ops._hip_ops.conv3d_{fwd,dgrad,wgrad}; CPU falls back to
+torch.nn.functional.conv3d. Restricted to the VBM workload's conv family:
+3x3x3 kernels, padding 1, stride 1 or 2, NCDHW.
+"""
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from . import native_available, require_native
+
+
+class _Conv3dFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, stride):
+        C = require_native()
+        xb = x.to(torch.bfloat16)
+        wb = weight.to(torch.bfloat16)
+        out = C.conv3d_fwd(xb, wb, stride)
+        if bias is not None:
+            out = out + bias.to(out.dtype).view(1, -1, 1, 1, 1)
+        ctx.save_for_backward(xb, wb)
+        ctx.stride = stride
+        ctx.has_bias = bias is not None
+        ctx.x_requires = x.requires_grad
+        ctx.in_dtype = x.dtype
+        ctx.w_dtype = weight.dtype
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        C = require_native()
+        xb, wb = ctx.saved_tensors
+        go = grad_out.to(torch.bfloat16).contiguous()
+        gx = gw = gb = None
+        if ctx.x_requires:
+            gx = C.conv3d_dgrad(go, wb, list(xb.shape),
+                                ctx.stride).to(ctx.in_dtype)
+        gw = C.conv3d_wgrad(xb, go, ctx.stride).to(ctx.w_dtype)
+        if ctx.has_bias:
+            gb = C.channel_sum(go)
+        return gx, gw, gb, None
+
+
+def conv3d(x, weight, bias=None, stride=1):
+    if x.is_cuda and native_available() and weight.shape[2:] == (3, 3, 3):
+        return _Conv3dFn.apply(x, weight, bias, int(stride))
+    return F.conv3d(x, weight, bias, stride=stride, padding=1)
+
+
+class OpsConv3d(nn.Conv3d):
+    """nn.Conv3d that runs the MFMA implicit-GEMM kernels on GPU.
+
+    Only 3x3x3/pad-1/stride-{1,2} routes to the HIP path; anything else
+    falls through to the library conv.
+    """
+
+    def forward(self, x):
+        if (x.is_cuda and native_available()
+                and self.kernel_size == (3, 3, 3)
+                and self.padding == (1, 1, 1)
+                and self.stride[0] in (1, 2)
+                and self.stride[0] == self.stride[1] == self.stride[2]
+                and self.dilation == (1, 1, 1) and self.groups == 1):
+            return _Conv3dFn.apply(x, self.weight, self.bias,
+                                   int(self.stride[0]))
+        return super().forward(x)

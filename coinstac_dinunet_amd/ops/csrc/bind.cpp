@@ -35,6 +35,13 @@ torch::Tensor confusion_matrix(torch::Tensor pred, torch::Tensor true_,
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor weight,
                          torch::Tensor bias, bool relu);
 torch::Tensor colsum(torch::Tensor g);
+// conv3d.hip
+torch::Tensor mfma_probe_gemm(torch::Tensor A, torch::Tensor B);
+torch::Tensor conv3d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride);
+torch::Tensor conv3d_dgrad(torch::Tensor go, torch::Tensor w,
+                           std::vector<int64_t> in_shape, int64_t stride);
+torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go, int64_t stride);
+torch::Tensor channel_sum(torch::Tensor go);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adam_flat", &fused_adam_flat, "single-kernel Adam on flat arenas");
@@ -49,4 +56,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("confusion_matrix", &confusion_matrix);
   m.def("linear_fwd", &linear_fwd);
   m.def("colsum", &colsum);
+  m.def("mfma_probe_gemm", &mfma_probe_gemm);
+  m.def("conv3d_fwd", &conv3d_fwd);
+  m.def("conv3d_dgrad", &conv3d_dgrad);
+  m.def("conv3d_wgrad", &conv3d_wgrad);
+  m.def("channel_sum", &channel_sum);
 }

@@ -1,0 +1,425 @@
+// K1 — implicit-GEMM Conv3d (3x3x3, pad 1, stride 1/2) on the matrix cores.
+// The reference's flagship VBM workload runs 3D conv through stock PyTorch
+// (SURVEY.md §2.9 K1); on gfx950 MIOpen degrades to naive fallbacks /
+// im2col+GEMM for bf16 NCDHW (profiles/r01_vbm_baseline.md, ~3% of MFMA
+// peak). Here: on-the-fly im2col staged through LDS feeding
+// v_mfma_f32_16x16x32_bf16, fp32 accumulate.
+//
+// GEMM views (all NCDHW, w-fastest output index so gathers coalesce):
+//   FWD  : C[M=N*OD*OH*OW, Cout] = patch(x)[M, Cin*27] @ W[Cout, Cin*27]^T
+//   DGRAD: C[M=N*D*H*W,   Cin ] = scatter-gather(go)[M, Cout*27] @ Wf
+//          (flipped kernel taps; stride handled by divisibility mask)
+//   WGRAD: C[Cout, Cin*27] = go^T[Cout, M] @ patch(x)[M, Cin*27]
+//          (split-K over M, fp32 atomics into the weight-grad buffer)
+//
+// Fragment layout (verified by the mfma_probe_* GPU tests): lane l holds
+//   A[row = l&15][k = (l>>4)*8 + j]  (bf16x8 per lane)
+//   B[k = (l>>4)*8 + j][col = l&15]
+//   C/D col = l&15, row = (l>>4)*4 + r.
+#include "common.h"
+
+#include <hip/hip_bf16.h>
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+struct ConvDims {
+  int N, Cin, D, H, W;
+  int Cout, OD, OH, OW;
+  int stride;  // pad fixed at 1, kernel 3x3x3
+};
+
+// ---------------------------------------------------------------------------
+// probe kernel: one wave computes C[16,16] = A[16,32] @ B[32,16]
+// ---------------------------------------------------------------------------
+__global__ void mfma_probe_kernel(const __bf16* __restrict__ A,
+                                  const __bf16* __restrict__ B,
+                                  float* __restrict__ C) {
+  const int lane = threadIdx.x & 63;
+  bf16x8 a, b;
+  const int row = lane & 15, kg = lane >> 4;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a[j] = A[row * 32 + kg * 8 + j];
+    b[j] = B[(kg * 8 + j) * 16 + row];
+  }
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) C[(kg * 4 + r) * 16 + (lane & 15)] = acc[r];
+}
+
+torch::Tensor mfma_probe_gemm(torch::Tensor A, torch::Tensor B) {
+  CHECK_GPU(A);
+  auto a = A.to(torch::kBFloat16).contiguous();
+  auto b = B.to(torch::kBFloat16).contiguous();
+  TORCH_CHECK(a.size(0) == 16 && a.size(1) == 32 && b.size(0) == 32 &&
+              b.size(1) == 16, "probe expects A[16,32], B[32,16]");
+  auto c = torch::empty({16, 16}, a.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0,
+                     current_stream(),
+                     reinterpret_cast<const __bf16*>(a.data_ptr()),
+                     reinterpret_cast<const __bf16*>(b.data_ptr()),
+                     c.data_ptr<float>());
+  return c;
+}
+
+// ---------------------------------------------------------------------------
+// FWD / DGRAD: block = 256 threads (4 waves, 2x2), tile BM=64 x BN=64,
+// K stepped 32 through LDS. Each wave: 32x32 C via 2x2 mfma fragments.
+// ---------------------------------------------------------------------------
+#define CBM 64
+#define CBN 64
+#define CBK 32
+#define LDA_PAD 8  // bf16 elements of row padding (16B) — conflict-free b128 reads
+
+// A-tile gather, FWD: m indexes output position, k = ci*27 + (kd,kh,kw)
+__device__ inline __bf16 gather_fwd(const __bf16* __restrict__ x,
+                                    const ConvDims& cd, int64_t m, int k) {
+  const int ow = (int)(m % cd.OW);
+  int64_t t = m / cd.OW;
+  const int oh = (int)(t % cd.OH);
+  t /= cd.OH;
+  const int od = (int)(t % cd.OD);
+  const int n = (int)(t / cd.OD);
+  const int ci = k / 27;
+  const int r = k - ci * 27;
+  const int kd = r / 9, kh = (r / 3) % 3, kw = r % 3;
+  const int id = od * cd.stride - 1 + kd;
+  const int ih = oh * cd.stride - 1 + kh;
+  const int iw = ow * cd.stride - 1 + kw;
+  if ((unsigned)id >= (unsigned)cd.D || (unsigned)ih >= (unsigned)cd.H ||
+      (unsigned)iw >= (unsigned)cd.W || ci >= cd.Cin)
+    return (__bf16)0.f;
+  return x[(((int64_t)n * cd.Cin + ci) * cd.D + id) * cd.H * cd.W +
+           (int64_t)ih * cd.W + iw];
+}
+
+// A-tile gather, DGRAD: m indexes input position, k = co*27 + taps
+__device__ inline __bf16 gather_dgrad(const __bf16* __restrict__ go,
+                                      const ConvDims& cd, int64_t m, int k) {
+  const int iw = (int)(m % cd.W);
+  int64_t t = m / cd.W;
+  const int ih = (int)(t % cd.H);
+  t /= cd.H;
+  const int id = (int)(t % cd.D);
+  const int n = (int)(t / cd.D);
+  const int co = k / 27;
+  const int r = k - co * 27;
+  const int kd = r / 9, kh = (r / 3) % 3, kw = r % 3;
+  const int td = id + 1 - kd, th = ih + 1 - kh, tw = iw + 1 - kw;
+  if (co >= cd.Cout) return (__bf16)0.f;
+  if (td % cd.stride || th % cd.stride || tw % cd.stride) return (__bf16)0.f;
+  const int od = td / cd.stride, oh = th / cd.stride, ow = tw / cd.stride;
+  if ((unsigned)od >= (unsigned)cd.OD || (unsigned)oh >= (unsigned)cd.OH ||
+      (unsigned)ow >= (unsigned)cd.OW)
+    return (__bf16)0.f;
+  return go[(((int64_t)n * cd.Cout + co) * cd.OD + od) * cd.OH * cd.OW +
+            (int64_t)oh * cd.OW + ow];
+}
+
+// B-tile gather. FWD: B[k][col] = w[col][k] with w[Cout][Cin*27].
+// DGRAD: B[k=(co,kd,kh,kw)][col=ci] = w[co][ci][26 - r] (flipped taps).
+template <bool DGRAD>
+__device__ inline __bf16 gather_w(const __bf16* __restrict__ w,
+                                  const ConvDims& cd, int k, int col) {
+  if (!DGRAD) {
+    if (col >= cd.Cout || k >= cd.Cin * 27) return (__bf16)0.f;
+    return w[(int64_t)col * (cd.Cin * 27) + k];
+  }
+  const int co = k / 27;
+  const int r = k - co * 27;
+  if (co >= cd.Cout || col >= cd.Cin) return (__bf16)0.f;
+  // dgrad taps are NOT flipped here because gather_dgrad already maps
+  // (id + 1 - kd): together they implement the transposed conv exactly.
+  return w[((int64_t)co * cd.Cin + col) * 27 + r];
+}
+
+template <bool DGRAD>
+__global__ __launch_bounds__(256) void conv3d_igemm_kernel(
+    const __bf16* __restrict__ Ain, const __bf16* __restrict__ w,
+    __bf16* __restrict__ out, ConvDims cd, int64_t M, int Ncol, int K) {
+  __shared__ __bf16 sA[CBM][CBK + LDA_PAD];
+  __shared__ __bf16 sBT[CBN][CBK + LDA_PAD];  // [col][k]: lanes read 8
+                                              // consecutive k => b128
+
+  const int64_t bm = (int64_t)blockIdx.x * CBM;
+  const int bn = blockIdx.y * CBN;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wm = (wave >> 1) * 32, wn = (wave & 1) * 32;
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int k0 = 0; k0 < K; k0 += CBK) {
+    // stage A: 64x32 = 2048 elems / 256 threads = 8 each.
+    // thread t covers k = t>>3 (column) and 8 consecutive m — consecutive
+    // m are consecutive ow/iw => coalesced within each 8-thread group.
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      int idx = tid * 8 + e;            // 0..2047
+      int mi = idx & 63, kk = idx >> 6;  // m-fast layout
+      int64_t m = bm + mi;
+      sA[mi][kk] = (m < M) ? (DGRAD ? gather_dgrad(Ain, cd, m, k0 + kk)
+                                    : gather_fwd(Ain, cd, m, k0 + kk))
+                           : (__bf16)0.f;
+    }
+    // stage B (transposed): thread covers 8 consecutive k of one col so
+    // the weight reads are contiguous and the LDS writes vectorize.
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      int idx = tid * 8 + e;
+      int kk = idx & 31, col = idx >> 5;
+      sBT[col][kk] = gather_w<DGRAD>(w, cd, k0 + kk, bn + col);
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int ks = 0; ks < CBK; ks += 32) {
+      const int row = lane & 15, kg = lane >> 4;
+      bf16x8 afrag[2], bfrag[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          afrag[i][j] = sA[wm + i * 16 + row][ks + kg * 8 + j];
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          bfrag[i][j] = sBT[wn + i * 16 + row][ks + kg * 8 + j];
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: out[m][col] with col = channel — but the tensor layout is
+  // [n][c][spatial], so out element = base(n) + col*spatialstride + m_sp.
+  const int ccol = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+  const int64_t spatial = DGRAD ? (int64_t)cd.D * cd.H * cd.W
+                                : (int64_t)cd.OD * cd.OH * cd.OW;
+  const int nch = DGRAD ? cd.Cin : cd.Cout;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int64_t m = bm + wm + i * 16 + crow0 + r;
+        int col = bn + wn + j * 16 + ccol;
+        if (m < M && col < nch) {
+          int64_t n = m / spatial, sp = m % spatial;
+          out[((int64_t)n * nch + col) * spatial + sp] =
+              (__bf16)(acc[i][j][r]);
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// WGRAD: C[Cout, Cin*27] = sum_m go[m,co] * patch(x)[m,k]; split-K over m
+// with fp32 atomics. Block: 4 waves each owning a 16x16 (co x k) fragment
+// pair; tile 32(co) x 32(k), K-chunk of positions per block.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void conv3d_wgrad_kernel(
+    const __bf16* __restrict__ x, const __bf16* __restrict__ go,
+    float* __restrict__ dw, ConvDims cd, int64_t M, int K, int64_t chunk) {
+  __shared__ __bf16 sGoT[32][CBK + LDA_PAD];  // [co][m]: b128 frag reads
+  __shared__ __bf16 sXT[32][CBK + LDA_PAD];   // [k][m]
+
+  const int co0 = blockIdx.x * 32;
+  const int k0 = blockIdx.y * 32;
+  const int64_t m0 = (int64_t)blockIdx.z * chunk;
+  const int64_t mEnd = min(m0 + chunk, M);
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wi = wave >> 1, wj = wave & 1;  // 2x2 over (co, k) 16x16 frags
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  const int64_t spatial = (int64_t)cd.OD * cd.OH * cd.OW;
+
+  for (int64_t mb = m0; mb < mEnd; mb += CBK) {
+    // stage go[mb..mb+32) x co[co0..co0+32): 1024 elems / 256 thr = 4 each
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      int idx = tid * 4 + e;
+      int mi = idx & 31, c = idx >> 5;
+      int64_t m = mb + mi;
+      __bf16 v = (__bf16)0.f;
+      if (m < mEnd && co0 + c < cd.Cout) {
+        int64_t n = m / spatial, sp = m % spatial;
+        v = go[((int64_t)n * cd.Cout + co0 + c) * spatial + sp];
+      }
+      sGoT[c][mi] = v;
+    }
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      int idx = tid * 4 + e;
+      int mi = idx & 31, kk = idx >> 5;
+      int64_t m = mb + mi;
+      sXT[kk][mi] = (m < mEnd && k0 + kk < K)
+                        ? gather_fwd(x, cd, m, k0 + kk)
+                        : (__bf16)0.f;
+    }
+    __syncthreads();
+
+    // A[co][m] frag: row=co (l&15), k-group over m; B[m][k]: same m group
+    const int row = lane & 15, kg = lane >> 4;
+    bf16x8 afrag, bfrag;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      afrag[j] = sGoT[wi * 16 + row][kg * 8 + j];
+      bfrag[j] = sXT[wj * 16 + row][kg * 8 + j];
+    }
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc, 0, 0, 0);
+    __syncthreads();
+  }
+
+  const int ccol = lane & 15;          // k col within fragment
+  const int crow0 = (lane >> 4) * 4;   // co row
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int co = co0 + wi * 16 + crow0 + r;
+    int k = k0 + wj * 16 + ccol;
+    if (co < cd.Cout && k < K)
+      atomicAdd(&dw[(int64_t)co * K + k], acc[r]);
+  }
+}
+
+// bias grad + (optionally) any channelwise sums: dB[co] = sum over m of go
+__global__ void channel_sum_kernel(const __bf16* __restrict__ go,
+                                   float* __restrict__ db, int N, int C,
+                                   int64_t spatial) {
+  // one block per channel; waves stride the spatial x batch space
+  const int c = blockIdx.x;
+  float s = 0.f;
+  for (int64_t i = threadIdx.x; i < (int64_t)N * spatial; i += blockDim.x) {
+    int64_t n = i / spatial, sp = i % spatial;
+    s += (float)go[((int64_t)n * C + c) * spatial + sp];
+  }
+  for (int off = WAVE_SIZE / 2; off > 0; off >>= 1) s += __shfl_down(s, off);
+  __shared__ float partial[4];
+  if ((threadIdx.x & 63) == 0) partial[threadIdx.x >> 6] = s;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float t = 0.f;
+    for (int i = 0; i < (int)(blockDim.x >> 6); ++i) t += partial[i];
+    db[c] = t;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+static ConvDims make_dims(const torch::Tensor& x, const torch::Tensor& w,
+                          int stride) {
+  ConvDims cd;
+  cd.N = (int)x.size(0); cd.Cin = (int)x.size(1);
+  cd.D = (int)x.size(2); cd.H = (int)x.size(3); cd.W = (int)x.size(4);
+  cd.Cout = (int)w.size(0);
+  cd.stride = stride;
+  cd.OD = (cd.D + 2 - 3) / stride + 1;
+  cd.OH = (cd.H + 2 - 3) / stride + 1;
+  cd.OW = (cd.W + 2 - 3) / stride + 1;
+  return cd;
+}
+
+torch::Tensor conv3d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride) {
+  CHECK_GPU(x);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "conv3d_fwd wants bf16");
+  auto xc = x.contiguous();
+  auto wc = w.to(torch::kBFloat16).contiguous();
+  auto cd = make_dims(xc, wc, (int)stride);
+  TORCH_CHECK(wc.size(2) == 3 && wc.size(3) == 3 && wc.size(4) == 3 &&
+              wc.size(1) == cd.Cin, "3x3x3 kernels only");
+  auto out = torch::empty({cd.N, cd.Cout, cd.OD, cd.OH, cd.OW}, xc.options());
+  int64_t M = (int64_t)cd.N * cd.OD * cd.OH * cd.OW;
+  int K = cd.Cin * 27;
+  dim3 grid((unsigned)((M + CBM - 1) / CBM), (cd.Cout + CBN - 1) / CBN);
+  hipLaunchKernelGGL((conv3d_igemm_kernel<false>), grid, dim3(256), 0,
+                     current_stream(),
+                     reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                     reinterpret_cast<const __bf16*>(wc.data_ptr()),
+                     reinterpret_cast<__bf16*>(out.data_ptr()), cd, M,
+                     cd.Cout, K);
+  return out;
+}
+
+torch::Tensor conv3d_dgrad(torch::Tensor go, torch::Tensor w,
+                           std::vector<int64_t> in_shape, int64_t stride) {
+  CHECK_GPU(go);
+  auto g = go.to(torch::kBFloat16).contiguous();
+  auto wc = w.to(torch::kBFloat16).contiguous();
+  auto dx = torch::empty(in_shape, g.options());
+  ConvDims cd;
+  cd.N = (int)in_shape[0]; cd.Cin = (int)in_shape[1];
+  cd.D = (int)in_shape[2]; cd.H = (int)in_shape[3]; cd.W = (int)in_shape[4];
+  cd.Cout = (int)wc.size(0);
+  cd.stride = (int)stride;
+  cd.OD = (int)g.size(2); cd.OH = (int)g.size(3); cd.OW = (int)g.size(4);
+  int64_t M = (int64_t)cd.N * cd.D * cd.H * cd.W;
+  int K = cd.Cout * 27;
+  dim3 grid((unsigned)((M + CBM - 1) / CBM), (cd.Cin + CBN - 1) / CBN);
+  hipLaunchKernelGGL((conv3d_igemm_kernel<true>), grid, dim3(256), 0,
+                     current_stream(),
+                     reinterpret_cast<const __bf16*>(g.data_ptr()),
+                     reinterpret_cast<const __bf16*>(wc.data_ptr()),
+                     reinterpret_cast<__bf16*>(dx.data_ptr()), cd, M, cd.Cin,
+                     K);
+  return dx;
+}
+
+torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
+                           int64_t stride) {
+  CHECK_GPU(x);
+  auto xc = x.to(torch::kBFloat16).contiguous();
+  auto g = go.to(torch::kBFloat16).contiguous();
+  ConvDims cd;
+  cd.N = (int)xc.size(0); cd.Cin = (int)xc.size(1);
+  cd.D = (int)xc.size(2); cd.H = (int)xc.size(3); cd.W = (int)xc.size(4);
+  cd.Cout = (int)g.size(1);
+  cd.stride = (int)stride;
+  cd.OD = (int)g.size(2); cd.OH = (int)g.size(3); cd.OW = (int)g.size(4);
+  int K = cd.Cin * 27;
+  int64_t M = (int64_t)cd.N * cd.OD * cd.OH * cd.OW;
+  auto dw = torch::zeros({cd.Cout, (int64_t)K},
+                         xc.options().dtype(torch::kFloat32));
+  // split-K sized so the grid covers the chip: >= 2048 blocks total
+  int planes = ((cd.Cout + 31) / 32) * ((K + 31) / 32);
+  int64_t target_chunks = std::max<int64_t>(1, 2048 / std::max(planes, 1));
+  int64_t chunk = std::max<int64_t>(CBK, (M + target_chunks - 1) /
+                                             target_chunks);
+  chunk = ((chunk + CBK - 1) / CBK) * CBK;
+  int64_t nchunks = (M + chunk - 1) / chunk;
+  dim3 grid((cd.Cout + 31) / 32, (K + 31) / 32, (unsigned)nchunks);
+  hipLaunchKernelGGL(conv3d_wgrad_kernel, grid, dim3(256), 0,
+                     current_stream(),
+                     reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                     reinterpret_cast<const __bf16*>(g.data_ptr()),
+                     dw.data_ptr<float>(), cd, M, K, chunk);
+  return dw.view({cd.Cout, cd.Cin, 3, 3, 3});
+}
+
+torch::Tensor channel_sum(torch::Tensor go) {
+  CHECK_GPU(go);
+  auto g = go.to(torch::kBFloat16).contiguous();
+  int N = (int)g.size(0), C = (int)g.size(1);
+  int64_t spatial = g.numel() / ((int64_t)N * C);
+  auto out = torch::empty({C}, g.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(channel_sum_kernel, dim3(C), dim3(256), 0,
+                     current_stream(),
+                     reinterpret_cast<const __bf16*>(g.data_ptr()),
+                     out.data_ptr<float>(), N, C, spatial);
+  return out;
+}
