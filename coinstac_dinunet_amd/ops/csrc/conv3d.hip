@@ -135,7 +135,7 @@ __device__ inline __bf16 gather_w(const __bf16* __restrict__ w,
   return w[((int64_t)co * cd.Cin + col) * 27 + r];
 }
 
-template <bool DGRAD>
+template <bool DGRAD, int STRIDE>
 __global__ __launch_bounds__(256) void conv3d_igemm_kernel(
     const __bf16* __restrict__ Ain, const __bf16* __restrict__ w,
     __bf16* __restrict__ out, ConvDims cd, int64_t M, int Ncol, int K) {
@@ -149,24 +149,86 @@ __global__ __launch_bounds__(256) void conv3d_igemm_kernel(
   const int wave = tid >> 6, lane = tid & 63;
   const int wm = (wave >> 1) * 32, wn = (wave & 1) * 32;
 
+  // This thread stages ONE k column (kk_t) for 8 consecutive m positions.
+  // Decode the m's once — the K loop then runs division-free (k-tap decode
+  // uses compile-time-constant divisors the compiler strength-reduces).
+  const int kk_t = tid >> 3;
+  const int mbase = (tid * 8) & 63;
+  const int SD = DGRAD ? cd.D : cd.OD;
+  const int SH = DGRAD ? cd.H : cd.OH;
+  const int SW = DGRAD ? cd.W : cd.OW;
+  int pn[8], pd[8], ph[8], pw[8];
+  {
+    int64_t m = bm + mbase;
+    int ww = (int)(m % SW);
+    int64_t t = m / SW;
+    int hh = (int)(t % SH);
+    t /= SH;
+    int dd = (int)(t % SD);
+    int nn = (int)(t / SD);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      pn[j] = nn; pd[j] = dd; ph[j] = hh; pw[j] = ww;
+      if (++ww == SW) { ww = 0; if (++hh == SH) { hh = 0;
+          if (++dd == SD) { dd = 0; ++nn; } } }
+    }
+  }
+  const bool m_ok = (bm + mbase + 7) < M;  // fast path: whole octet valid
+
   f32x4 acc[2][2];
 #pragma unroll
   for (int i = 0; i < 2; ++i)
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
+  const int64_t HW = (int64_t)cd.H * cd.W;
+  const int64_t OHW = (int64_t)cd.OH * cd.OW;
+
   for (int k0 = 0; k0 < K; k0 += CBK) {
-    // stage A: 64x32 = 2048 elems / 256 threads = 8 each.
-    // thread t covers k = t>>3 (column) and 8 consecutive m — consecutive
-    // m are consecutive ow/iw => coalesced within each 8-thread group.
+    const int k = k0 + kk_t;
+    if (!DGRAD) {
+      const int ci = k / 27;               // constant divisors: mul+shift
+      const int r = k - ci * 27;
+      const int kd = r / 9, kh = (r / 3) % 3, kw = r % 3;
+      const bool k_ok = ci < cd.Cin;
 #pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      int idx = tid * 8 + e;            // 0..2047
-      int mi = idx & 63, kk = idx >> 6;  // m-fast layout
-      int64_t m = bm + mi;
-      sA[mi][kk] = (m < M) ? (DGRAD ? gather_dgrad(Ain, cd, m, k0 + kk)
-                                    : gather_fwd(Ain, cd, m, k0 + kk))
-                           : (__bf16)0.f;
+      for (int j = 0; j < 8; ++j) {
+        __bf16 v = (__bf16)0.f;
+        if (k_ok && (m_ok || (bm + mbase + j) < M)) {
+          const int id = pd[j] * STRIDE - 1 + kd;
+          const int ih = ph[j] * STRIDE - 1 + kh;
+          const int iw = pw[j] * STRIDE - 1 + kw;
+          if ((unsigned)id < (unsigned)cd.D && (unsigned)ih < (unsigned)cd.H &&
+              (unsigned)iw < (unsigned)cd.W)
+            v = Ain[(((int64_t)pn[j] * cd.Cin + ci) * cd.D + id) * HW +
+                    (int64_t)ih * cd.W + iw];
+        }
+        sA[mbase + j][kk_t] = v;
+      }
+    } else {
+      const int co = k / 27;
+      const int r = k - co * 27;
+      const int kd = r / 9, kh = (r / 3) % 3, kw = r % 3;
+      const bool k_ok = co < cd.Cout;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        __bf16 v = (__bf16)0.f;
+        if (k_ok && (m_ok || (bm + mbase + j) < M)) {
+          const int td = pd[j] + 1 - kd, th = ph[j] + 1 - kh,
+                    tw = pw[j] + 1 - kw;
+          if (STRIDE == 1 ||
+              (!(td & (STRIDE - 1)) && !(th & (STRIDE - 1)) &&
+               !(tw & (STRIDE - 1)))) {
+            const int od = td / STRIDE, oh = th / STRIDE, ow = tw / STRIDE;
+            if ((unsigned)od < (unsigned)cd.OD &&
+                (unsigned)oh < (unsigned)cd.OH &&
+                (unsigned)ow < (unsigned)cd.OW)
+              v = Ain[(((int64_t)pn[j] * cd.Cout + co) * cd.OD + od) * OHW +
+                      (int64_t)oh * cd.OW + ow];
+          }
+        }
+        sA[mbase + j][kk_t] = v;
+      }
     }
     // stage B (transposed): thread covers 8 consecutive k of one col so
     // the weight reads are contiguous and the LDS writes vectorize.
@@ -232,11 +294,13 @@ __global__ __launch_bounds__(256) void conv3d_igemm_kernel(
 // with fp32 atomics. Block: 4 waves each owning a 16x16 (co x k) fragment
 // pair; tile 32(co) x 32(k), K-chunk of positions per block.
 // ---------------------------------------------------------------------------
+#define WMB 128  // m positions staged per iteration (4 MFMA/wave/barrier)
+
 __global__ __launch_bounds__(256) void conv3d_wgrad_kernel(
     const __bf16* __restrict__ x, const __bf16* __restrict__ go,
     float* __restrict__ dw, ConvDims cd, int64_t M, int K, int64_t chunk) {
-  __shared__ __bf16 sGoT[32][CBK + LDA_PAD];  // [co][m]: b128 frag reads
-  __shared__ __bf16 sXT[32][CBK + LDA_PAD];   // [k][m]
+  __shared__ __bf16 sGoT[32][WMB + LDA_PAD];  // [co][m]: b128 frag reads
+  __shared__ __bf16 sXT[32][WMB + LDA_PAD];   // [k][m]
 
   const int co0 = blockIdx.x * 32;
   const int k0 = blockIdx.y * 32;
@@ -246,43 +310,76 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_kernel(
   const int wave = tid >> 6, lane = tid & 63;
   const int wi = wave >> 1, wj = wave & 1;  // 2x2 over (co, k) 16x16 frags
 
+  // this thread stages 16 consecutive m for one co column AND one k column
+  const int mi0 = (tid * 16) & 127;
+  const int ct = tid >> 3;          // co/k column (0..31)
+  const int kq = k0 + ct;
+  const int ci = kq / 27;
+  const int rr = kq - ci * 27;
+  const int kd = rr / 9, kh = (rr / 3) % 3, kw = rr % 3;
+  const bool k_ok = ci < cd.Cin && kq < K;
+  const bool c_ok = (co0 + ct) < cd.Cout;
+
+  // incremental output-position decode for m = mb + mi0 (advances by WMB)
+  int nn, od, oh, ow;
+  {
+    int64_t m = m0 + mi0;
+    ow = (int)(m % cd.OW);
+    int64_t t = m / cd.OW;
+    oh = (int)(t % cd.OH);
+    t /= cd.OH;
+    od = (int)(t % cd.OD);
+    nn = (int)(t / cd.OD);
+  }
+
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   const int64_t spatial = (int64_t)cd.OD * cd.OH * cd.OW;
+  const int64_t HW = (int64_t)cd.H * cd.W;
 
-  for (int64_t mb = m0; mb < mEnd; mb += CBK) {
-    // stage go[mb..mb+32) x co[co0..co0+32): 1024 elems / 256 thr = 4 each
+  for (int64_t mb = m0; mb < mEnd; mb += WMB) {
+    int jn = nn, jd = od, jh = oh, jw = ow;
 #pragma unroll
-    for (int e = 0; e < 4; ++e) {
-      int idx = tid * 4 + e;
-      int mi = idx & 31, c = idx >> 5;
-      int64_t m = mb + mi;
-      __bf16 v = (__bf16)0.f;
-      if (m < mEnd && co0 + c < cd.Cout) {
-        int64_t n = m / spatial, sp = m % spatial;
-        v = go[((int64_t)n * cd.Cout + co0 + c) * spatial + sp];
+    for (int j = 0; j < 16; ++j) {
+      const int64_t m = mb + mi0 + j;
+      const bool ok = m < mEnd;
+      __bf16 gv = (__bf16)0.f;
+      if (ok && c_ok)
+        gv = go[((int64_t)jn * cd.Cout + (co0 + ct)) * spatial +
+                ((int64_t)jd * cd.OH + jh) * cd.OW + jw];
+      sGoT[ct][mi0 + j] = gv;
+      __bf16 xv = (__bf16)0.f;
+      if (ok && k_ok) {
+        const int id = jd * cd.stride - 1 + kd;
+        const int ih = jh * cd.stride - 1 + kh;
+        const int iw = jw * cd.stride - 1 + kw;
+        if ((unsigned)id < (unsigned)cd.D && (unsigned)ih < (unsigned)cd.H &&
+            (unsigned)iw < (unsigned)cd.W)
+          xv = x[(((int64_t)jn * cd.Cin + ci) * cd.D + id) * HW +
+                 (int64_t)ih * cd.W + iw];
       }
-      sGoT[c][mi] = v;
+      sXT[ct][mi0 + j] = xv;
+      if (++jw == cd.OW) { jw = 0; if (++jh == cd.OH) { jh = 0;
+          if (++jd == cd.OD) { jd = 0; ++jn; } } }
     }
-#pragma unroll
-    for (int e = 0; e < 4; ++e) {
-      int idx = tid * 4 + e;
-      int mi = idx & 31, kk = idx >> 5;
-      int64_t m = mb + mi;
-      sXT[kk][mi] = (m < mEnd && k0 + kk < K)
-                        ? gather_fwd(x, cd, m, k0 + kk)
-                        : (__bf16)0.f;
+    ow += WMB;
+    while (ow >= cd.OW) {
+      ow -= cd.OW;
+      if (++oh == cd.OH) { oh = 0; if (++od == cd.OD) { od = 0; ++nn; } }
     }
     __syncthreads();
 
-    // A[co][m] frag: row=co (l&15), k-group over m; B[m][k]: same m group
     const int row = lane & 15, kg = lane >> 4;
-    bf16x8 afrag, bfrag;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      afrag[j] = sGoT[wi * 16 + row][kg * 8 + j];
-      bfrag[j] = sXT[wj * 16 + row][kg * 8 + j];
+    for (int ks = 0; ks < WMB; ks += 32) {
+      bf16x8 afrag, bfrag;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        afrag[j] = sGoT[wi * 16 + row][ks + kg * 8 + j];
+        bfrag[j] = sXT[wj * 16 + row][ks + kg * 8 + j];
+      }
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc,
+                                                    0, 0, 0);
     }
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc, 0, 0, 0);
     __syncthreads();
   }
 
@@ -295,6 +392,179 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_kernel(
     if (co < cd.Cout && k < K)
       atomicAdd(&dw[(int64_t)co * K + k], acc[r]);
   }
+}
+
+// ---------------------------------------------------------------------------
+// Stride-2 DGRAD, parity-decomposed: the dense mask formulation wastes 7/8
+// of the MFMA work (only taps with (id+1-kd) even contribute). Decompose
+// dx by (id,ih,iw) mod 2 into 8 classes; each class is a DENSE implicit
+// GEMM with K = Cout * (1 or 2)^3 taps — total FLOPs equal to the forward
+// pass. blockIdx.z = class.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void conv3d_dgrad_s2_kernel(
+    const __bf16* __restrict__ go, const __bf16* __restrict__ w,
+    __bf16* __restrict__ dx, ConvDims cd) {
+  __shared__ __bf16 sA[CBM][CBK + LDA_PAD];
+  __shared__ __bf16 sBT[CBN][CBK + LDA_PAD];
+
+  const int cls = blockIdx.z;
+  const int a = (cls >> 2) & 1, b = (cls >> 1) & 1, c = cls & 1;
+  const int Da = (cd.D - a + 1) >> 1;
+  const int Hb = (cd.H - b + 1) >> 1;
+  const int Wc = (cd.W - c + 1) >> 1;
+  const int nd = a ? 2 : 1, nh = b ? 2 : 1, nw = c ? 2 : 1;
+  const int l2w = c, l2h = b;              // log2 tap counts
+  const int T = nd * nh * nw;
+  const int l2T = a + b + c;
+  const int K = cd.Cout << l2T;
+  const int64_t M = (int64_t)cd.N * Da * Hb * Wc;
+
+  const int64_t bm = (int64_t)blockIdx.x * CBM;
+  if (bm >= M) return;
+  const int bn = blockIdx.y * CBN;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wm = (wave >> 1) * 32, wn = (wave & 1) * 32;
+
+  const int kk_t = tid >> 3;
+  const int mbase = (tid * 8) & 63;
+  int pn[8], pd[8], ph[8], pw[8];
+  {
+    int64_t m = bm + mbase;
+    int ww = (int)(m % Wc);
+    int64_t t = m / Wc;
+    int hh = (int)(t % Hb);
+    t /= Hb;
+    int dd = (int)(t % Da);
+    int nn = (int)(t / Da);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      pn[j] = nn; pd[j] = dd; ph[j] = hh; pw[j] = ww;
+      if (++ww == Wc) { ww = 0; if (++hh == Hb) { hh = 0;
+          if (++dd == Da) { dd = 0; ++nn; } } }
+    }
+  }
+  const bool m_ok = (bm + mbase + 7) < M;
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int64_t OHW = (int64_t)cd.OH * cd.OW;
+
+  for (int k0 = 0; k0 < K; k0 += CBK) {
+    {
+      const int k = k0 + kk_t;
+      const int co = k >> l2T;
+      const int r = k & (T - 1);
+      const int tw_i = r & (nw - 1);
+      const int th_i = (r >> l2w) & (nh - 1);
+      const int td_i = r >> (l2w + l2h);
+      // tap: a==0 -> kd=1 (od=id'); a==1 -> kd=2*td_i (od=id'+1-td_i)
+      const int dod = a ? (1 - td_i) : 0;   // od = id' + dod
+      const int doh = b ? (1 - th_i) : 0;
+      const int dow = c ? (1 - tw_i) : 0;
+      const bool k_ok = co < cd.Cout;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        __bf16 v = (__bf16)0.f;
+        if (k_ok && (m_ok || (bm + mbase + j) < M)) {
+          const int od = pd[j] + dod, oh = ph[j] + doh, ow = pw[j] + dow;
+          if ((unsigned)od < (unsigned)cd.OD &&
+              (unsigned)oh < (unsigned)cd.OH &&
+              (unsigned)ow < (unsigned)cd.OW)
+            v = go[(((int64_t)pn[j] * cd.Cout + co) * cd.OD + od) * OHW +
+                   (int64_t)oh * cd.OW + ow];
+        }
+        sA[mbase + j][kk_t] = v;
+      }
+    }
+    // B: w[co][ci=col][kd*9+kh*3+kw]
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      int idx = tid * 8 + e;
+      int kk = idx & 31, col = idx >> 5;
+      const int k = k0 + kk;
+      const int co = k >> l2T;
+      const int r = k & (T - 1);
+      const int tw_i = r & (nw - 1);
+      const int th_i = (r >> l2w) & (nh - 1);
+      const int td_i = r >> (l2w + l2h);
+      const int kd = a ? (td_i * 2) : 1;
+      const int kh = b ? (th_i * 2) : 1;
+      const int kw = c ? (tw_i * 2) : 1;
+      __bf16 v = (__bf16)0.f;
+      if (co < cd.Cout && (bn + col) < cd.Cin)
+        v = w[((int64_t)co * cd.Cin + (bn + col)) * 27 + kd * 9 + kh * 3 +
+              kw];
+      sBT[col][kk] = v;
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int ks = 0; ks < CBK; ks += 32) {
+      const int row = lane & 15, kg = lane >> 4;
+      bf16x8 afrag[2], bfrag[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          afrag[i][j] = sA[wm + i * 16 + row][ks + kg * 8 + j];
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          bfrag[i][j] = sBT[wn + i * 16 + row][ks + kg * 8 + j];
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int ccol = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+  const int64_t HW = (int64_t)cd.H * cd.W;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int64_t m = bm + wm + i * 16 + crow0 + r;
+        int ci = bn + wn + j * 16 + ccol;
+        if (m < M && ci < cd.Cin) {
+          const int iw = (int)(m % Wc);
+          int64_t t = m / Wc;
+          const int ih = (int)(t % Hb);
+          t /= Hb;
+          const int id = (int)(t % Da);
+          const int n = (int)(t / Da);
+          dx[(((int64_t)n * cd.Cin + ci) * cd.D + (2 * id + a)) * HW +
+             (int64_t)(2 * ih + b) * cd.W + (2 * iw + c)] =
+              (__bf16)(acc[i][j][r]);
+        }
+      }
+    }
+  }
+}
+
+static torch::Tensor conv3d_dgrad_s2(torch::Tensor g, torch::Tensor wc,
+                                     torch::Tensor dx, ConvDims cd) {
+  int Da = (cd.D + 1) >> 1, Hb = (cd.H + 1) >> 1, Wc = (cd.W + 1) >> 1;
+  int64_t Mmax = (int64_t)cd.N * Da * Hb * Wc;
+  dim3 grid((unsigned)((Mmax + CBM - 1) / CBM), (cd.Cin + CBN - 1) / CBN, 8);
+  hipLaunchKernelGGL(conv3d_dgrad_s2_kernel, grid, dim3(256), 0,
+                     current_stream(),
+                     reinterpret_cast<const __bf16*>(g.data_ptr()),
+                     reinterpret_cast<const __bf16*>(wc.data_ptr()),
+                     reinterpret_cast<__bf16*>(dx.data_ptr()), cd);
+  return dx;
 }
 
 // bias grad + (optionally) any channelwise sums: dB[co] = sum over m of go
@@ -347,12 +617,16 @@ torch::Tensor conv3d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride) {
   int64_t M = (int64_t)cd.N * cd.OD * cd.OH * cd.OW;
   int K = cd.Cin * 27;
   dim3 grid((unsigned)((M + CBM - 1) / CBM), (cd.Cout + CBN - 1) / CBN);
-  hipLaunchKernelGGL((conv3d_igemm_kernel<false>), grid, dim3(256), 0,
-                     current_stream(),
-                     reinterpret_cast<const __bf16*>(xc.data_ptr()),
-                     reinterpret_cast<const __bf16*>(wc.data_ptr()),
-                     reinterpret_cast<__bf16*>(out.data_ptr()), cd, M,
-                     cd.Cout, K);
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, current_stream(),
+                       reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                       reinterpret_cast<const __bf16*>(wc.data_ptr()),
+                       reinterpret_cast<__bf16*>(out.data_ptr()), cd, M,
+                       cd.Cout, K);
+  };
+  TORCH_CHECK(stride == 1 || stride == 2, "stride must be 1 or 2");
+  if (stride == 1) launch(conv3d_igemm_kernel<false, 1>);
+  else launch(conv3d_igemm_kernel<false, 2>);
   return out;
 }
 
@@ -370,8 +644,10 @@ torch::Tensor conv3d_dgrad(torch::Tensor go, torch::Tensor w,
   cd.OD = (int)g.size(2); cd.OH = (int)g.size(3); cd.OW = (int)g.size(4);
   int64_t M = (int64_t)cd.N * cd.D * cd.H * cd.W;
   int K = cd.Cout * 27;
+  TORCH_CHECK(stride == 1 || stride == 2, "stride must be 1 or 2");
+  if (stride == 2) return conv3d_dgrad_s2(g, wc, dx, cd);
   dim3 grid((unsigned)((M + CBM - 1) / CBM), (cd.Cin + CBN - 1) / CBN);
-  hipLaunchKernelGGL((conv3d_igemm_kernel<true>), grid, dim3(256), 0,
+  hipLaunchKernelGGL((conv3d_igemm_kernel<true, 1>), grid, dim3(256), 0,
                      current_stream(),
                      reinterpret_cast<const __bf16*>(g.data_ptr()),
                      reinterpret_cast<const __bf16*>(wc.data_ptr()),
@@ -398,9 +674,9 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
   // split-K sized so the grid covers the chip: >= 2048 blocks total
   int planes = ((cd.Cout + 31) / 32) * ((K + 31) / 32);
   int64_t target_chunks = std::max<int64_t>(1, 2048 / std::max(planes, 1));
-  int64_t chunk = std::max<int64_t>(CBK, (M + target_chunks - 1) /
-                                             target_chunks);
-  chunk = ((chunk + CBK - 1) / CBK) * CBK;
+  int64_t chunk = std::max<int64_t>(128, (M + target_chunks - 1) /
+                                              target_chunks);
+  chunk = ((chunk + 127) / 128) * 128;
   int64_t nchunks = (M + chunk - 1) / chunk;
   dim3 grid((cd.Cout + 31) / 32, (K + 31) / 32, (unsigned)nchunks);
   hipLaunchKernelGGL(conv3d_wgrad_kernel, grid, dim3(256), 0,
