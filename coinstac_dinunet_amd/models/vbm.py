@@ -12,6 +12,7 @@ bf16 autocast-friendly (no ops that silently upcast).
 import torch
 import torch.nn as nn
 
+from ..ops.bnorm import OpsBatchNorm3d
 from ..ops.conv import OpsConv3d
 
 
@@ -19,11 +20,10 @@ class _ConvBlock(nn.Module):
     def __init__(self, cin, cout, stride=1):
         super().__init__()
         self.conv = OpsConv3d(cin, cout, 3, stride=stride, padding=1, bias=False)
-        self.bn = nn.BatchNorm3d(cout)
-        self.act = nn.ReLU(inplace=True)
+        self.bn = OpsBatchNorm3d(cout, relu=True)  # ReLU fused into BN pass
 
     def forward(self, x):
-        return self.act(self.bn(self.conv(x)))
+        return self.bn(self.conv(x))
 
 
 class VBMNet(nn.Module):

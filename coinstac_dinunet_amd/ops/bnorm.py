@@ -1,0 +1,64 @@
+"""BatchNorm3d(+fused ReLU) on the HIP streaming kernels.
+
+Training fwd: one reduce + one normalize pass (HBM-roofline); the ReLU
+folds into the normalize so conv->bn->relu is 2 passes instead of
+MIOpen's BN + a separate clamp kernel. Running-stat semantics match
+nn.BatchNorm3d (biased var normalizes, unbiased updates running_var).
+"""
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from . import native_available, require_native
+
+
+class _BN3dFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var, momentum,
+                eps, relu):
+        C = require_native()
+        xb = x.to(torch.bfloat16)
+        y, mean, var, mean_rstd = C.bn3d_fwd(xb, gamma, beta, eps, relu)
+        if running_mean is not None:
+            with torch.no_grad():
+                n = xb.numel() // xb.size(1)
+                unbiased = var * (n / max(n - 1, 1))
+                running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
+        ctx.save_for_backward(xb, mean_rstd, gamma, beta)
+        ctx.relu = relu
+        ctx.in_dtype = x.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        C = require_native()
+        xb, mean_rstd, gamma, beta = ctx.saved_tensors
+        dx, dgamma, dbeta = C.bn3d_bwd(dy.to(torch.bfloat16), xb, mean_rstd,
+                                       gamma, beta, ctx.relu)
+        return (dx.to(ctx.in_dtype), dgamma.to(gamma.dtype),
+                dbeta.to(beta.dtype), None, None, None, None, None)
+
+
+class OpsBatchNorm3d(nn.BatchNorm3d):
+    """BatchNorm3d with optional fused ReLU; HIP kernels on GPU."""
+
+    def __init__(self, num_features, eps=1e-5, momentum=0.1, relu=False,
+                 **kw):
+        super().__init__(num_features, eps=eps, momentum=momentum, **kw)
+        self.relu = relu
+
+    def forward(self, x):
+        if not (x.is_cuda and native_available()):
+            y = super().forward(x)
+            return F.relu(y, inplace=True) if self.relu else y
+        if self.training:
+            if self.num_batches_tracked is not None:
+                self.num_batches_tracked.add_(1)
+            return _BN3dFn.apply(x, self.weight, self.bias,
+                                 self.running_mean, self.running_var,
+                                 self.momentum, self.eps, self.relu)
+        C = require_native()
+        return C.bn3d_infer(x.to(torch.bfloat16), self.weight, self.bias,
+                            self.running_mean, self.running_var, self.eps,
+                            self.relu)

@@ -42,6 +42,16 @@ torch::Tensor conv3d_dgrad(torch::Tensor go, torch::Tensor w,
                            std::vector<int64_t> in_shape, int64_t stride);
 torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go, int64_t stride);
 torch::Tensor channel_sum(torch::Tensor go);
+// bnorm.hip
+std::vector<torch::Tensor> bn3d_fwd(torch::Tensor x, torch::Tensor gamma,
+                                    torch::Tensor beta, double eps, bool relu);
+torch::Tensor bn3d_infer(torch::Tensor x, torch::Tensor gamma,
+                         torch::Tensor beta, torch::Tensor running_mean,
+                         torch::Tensor running_var, double eps, bool relu);
+std::vector<torch::Tensor> bn3d_bwd(torch::Tensor dy, torch::Tensor x,
+                                    torch::Tensor mean_rstd,
+                                    torch::Tensor gamma, torch::Tensor beta,
+                                    bool relu);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adam_flat", &fused_adam_flat, "single-kernel Adam on flat arenas");
@@ -61,4 +71,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3d_dgrad", &conv3d_dgrad);
   m.def("conv3d_wgrad", &conv3d_wgrad);
   m.def("channel_sum", &channel_sum);
+  m.def("bn3d_fwd", &bn3d_fwd);
+  m.def("bn3d_infer", &bn3d_infer);
+  m.def("bn3d_bwd", &bn3d_bwd);
 }

@@ -310,8 +310,10 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_kernel(
   const int wave = tid >> 6, lane = tid & 63;
   const int wi = wave >> 1, wj = wave & 1;  // 2x2 over (co, k) 16x16 frags
 
-  // this thread stages 16 consecutive m for one co column AND one k column
-  const int mi0 = (tid * 16) & 127;
+  // this thread stages 4x4 m (four 32-m quads) for one co column AND one
+  // k column — short 4-element runs keep lane addresses 8 B apart
+  // (coalesced) while each barrier pair still feeds 4 MFMAs.
+  const int mi0 = (tid & 7) * 4;
   const int ct = tid >> 3;          // co/k column (0..31)
   const int kq = k0 + ct;
   const int ci = kq / 27;
@@ -320,7 +322,7 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_kernel(
   const bool k_ok = ci < cd.Cin && kq < K;
   const bool c_ok = (co0 + ct) < cd.Cout;
 
-  // incremental output-position decode for m = mb + mi0 (advances by WMB)
+  // incremental output-position decode for m = mb + mi0 (advances by 32)
   int nn, od, oh, ow;
   {
     int64_t m = m0 + mi0;
@@ -337,34 +339,40 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_kernel(
   const int64_t HW = (int64_t)cd.H * cd.W;
 
   for (int64_t mb = m0; mb < mEnd; mb += WMB) {
-    int jn = nn, jd = od, jh = oh, jw = ow;
 #pragma unroll
-    for (int j = 0; j < 16; ++j) {
-      const int64_t m = mb + mi0 + j;
-      const bool ok = m < mEnd;
-      __bf16 gv = (__bf16)0.f;
-      if (ok && c_ok)
-        gv = go[((int64_t)jn * cd.Cout + (co0 + ct)) * spatial +
-                ((int64_t)jd * cd.OH + jh) * cd.OW + jw];
-      sGoT[ct][mi0 + j] = gv;
-      __bf16 xv = (__bf16)0.f;
-      if (ok && k_ok) {
-        const int id = jd * cd.stride - 1 + kd;
-        const int ih = jh * cd.stride - 1 + kh;
-        const int iw = jw * cd.stride - 1 + kw;
-        if ((unsigned)id < (unsigned)cd.D && (unsigned)ih < (unsigned)cd.H &&
-            (unsigned)iw < (unsigned)cd.W)
-          xv = x[(((int64_t)jn * cd.Cin + ci) * cd.D + id) * HW +
-                 (int64_t)ih * cd.W + iw];
+    for (int q = 0; q < 4; ++q) {
+      int jn = nn, jd = od, jh = oh, jw = ow;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int mloc = q * 32 + mi0 + j;
+        const int64_t m = mb + mloc;
+        const bool ok = m < mEnd;
+        __bf16 gv = (__bf16)0.f;
+        if (ok && c_ok)
+          gv = go[((int64_t)jn * cd.Cout + (co0 + ct)) * spatial +
+                  ((int64_t)jd * cd.OH + jh) * cd.OW + jw];
+        sGoT[ct][mloc] = gv;
+        __bf16 xv = (__bf16)0.f;
+        if (ok && k_ok) {
+          const int id = jd * cd.stride - 1 + kd;
+          const int ih = jh * cd.stride - 1 + kh;
+          const int iw = jw * cd.stride - 1 + kw;
+          if ((unsigned)id < (unsigned)cd.D &&
+              (unsigned)ih < (unsigned)cd.H &&
+              (unsigned)iw < (unsigned)cd.W)
+            xv = x[(((int64_t)jn * cd.Cin + ci) * cd.D + id) * HW +
+                   (int64_t)ih * cd.W + iw];
+        }
+        sXT[ct][mloc] = xv;
+        if (++jw == cd.OW) { jw = 0; if (++jh == cd.OH) { jh = 0;
+            if (++jd == cd.OD) { jd = 0; ++jn; } } }
       }
-      sXT[ct][mi0 + j] = xv;
-      if (++jw == cd.OW) { jw = 0; if (++jh == cd.OH) { jh = 0;
-          if (++jd == cd.OD) { jd = 0; ++jn; } } }
-    }
-    ow += WMB;
-    while (ow >= cd.OW) {
-      ow -= cd.OW;
-      if (++oh == cd.OH) { oh = 0; if (++od == cd.OD) { od = 0; ++nn; } }
+      // advance the walker by one 32-m quad
+      ow += 32;
+      while (ow >= cd.OW) {
+        ow -= cd.OW;
+        if (++oh == cd.OH) { oh = 0; if (++od == cd.OD) { od = 0; ++nn; } }
+      }
     }
     __syncthreads();
 
