@@ -17,7 +17,10 @@ class _Conv3dFn(torch.autograd.Function):
         C = require_native()
         xb = x.to(torch.bfloat16)
         wb = weight.to(torch.bfloat16)
-        out = C.conv3d_fwd(xb, wb, stride)
+        if stride == 1 and xb.size(4) % 8 == 0:
+            out = C.conv3d_fwd_spatial(xb, wb)
+        else:
+            out = C.conv3d_fwd(xb, wb, stride)
         if bias is not None:
             out = out + bias.to(out.dtype).view(1, -1, 1, 1, 1)
         ctx.save_for_backward(xb, wb)
@@ -35,8 +38,12 @@ class _Conv3dFn(torch.autograd.Function):
         go = grad_out.to(torch.bfloat16).contiguous()
         gx = gw = gb = None
         if ctx.x_requires:
-            gx = C.conv3d_dgrad(go, wb, list(xb.shape),
-                                ctx.stride).to(ctx.in_dtype)
+            if ctx.stride == 1 and xb.size(4) % 8 == 0:
+                gx = C.conv3d_dgrad_spatial(go, wb,
+                                            list(xb.shape)).to(ctx.in_dtype)
+            else:
+                gx = C.conv3d_dgrad(go, wb, list(xb.shape),
+                                    ctx.stride).to(ctx.in_dtype)
         gw = C.conv3d_wgrad(xb, go, ctx.stride).to(ctx.w_dtype)
         if ctx.has_bias:
             gb = C.channel_sum(go)

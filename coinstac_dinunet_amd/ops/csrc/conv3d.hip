@@ -575,6 +575,138 @@ static torch::Tensor conv3d_dgrad_s2(torch::Tensor g, torch::Tensor wc,
   return dx;
 }
 
+// ---------------------------------------------------------------------------
+// Stride-1 WGRAD with tap reuse: the split-K implicit-GEMM form re-reads
+// x and go once PER (co,k)-plane (M*K element traffic ~ 27x the tensor).
+// Here a block stages one x spatial slab [ci32][3d][OHT+2][OWT+2] and one
+// go tile [co32][256m] in LDS ONCE, then computes ALL 27 tap GEMMs from
+// it: per 32-m sub-chunk the go fragment loads once and feeds 27 MFMAs
+// whose x fragments read the same slab at shifted offsets. Waves 2x2 over
+// (co32, ci32); acc = 27 f32x4 per wave. Blocks grid-stride over spatial
+// chunks and atomically fold their partials into dw once at the end.
+// ---------------------------------------------------------------------------
+template <int OWT>
+__global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
+    const __bf16* __restrict__ x, const __bf16* __restrict__ go,
+    float* __restrict__ dw, ConvDims cd, int64_t nchunks, int64_t zstride) {
+  constexpr int OHT = 256 / OWT;
+  constexpr int W2 = OWT + 4;   // staged row length (2 halo + pad)
+  constexpr int H2 = OHT + 2;
+  __shared__ __bf16 sX[32][3][H2][W2];
+  __shared__ __bf16 sGo[32][256 + LDA_PAD];
+
+  const int co0 = blockIdx.x * 32;
+  const int ci0 = blockIdx.y * 32;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wi = wave >> 1, wj = wave & 1;   // (co, ci) 16x16 fragment
+  const int row = lane & 15, kg = lane >> 4;
+
+  const int wtiles = (cd.OW + OWT - 1) / OWT;
+  const int htiles = (cd.OH + OHT - 1) / OHT;
+
+  f32x4 acc[27];
+#pragma unroll
+  for (int t = 0; t < 27; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
+
+  const int64_t HW = (int64_t)cd.H * cd.W;
+  const int64_t OHW = (int64_t)cd.OH * cd.OW;
+
+  for (int64_t z = blockIdx.z; z < nchunks; z += zstride) {
+    // chunk -> (n, od, oh_tile, ow_tile)
+    int64_t t = z;
+    const int wt = (int)(t % wtiles);
+    t /= wtiles;
+    const int ht = (int)(t % htiles);
+    t /= htiles;
+    const int od = (int)(t % cd.OD);
+    const int n = (int)(t / cd.OD);
+    const int oh0 = ht * OHT, ow0 = wt * OWT;
+
+    // ---- stage x slab: [ci32][id=od-1+kd][ih=oh0-1..][iw=ow0-1..] ------
+    constexpr int XELEMS = 32 * 3 * H2 * W2;
+    const __bf16* xn = x + (int64_t)n * cd.Cin * cd.D * HW;
+    for (int idx = tid; idx < XELEMS; idx += 256) {
+      const int col = idx % W2;
+      int r = idx / W2;
+      const int hrow = r % H2;
+      r /= H2;
+      const int kd = r % 3;
+      const int ci = r / 3;
+      const int id = od - 1 + kd;
+      const int ih = oh0 - 1 + hrow;
+      const int iw = ow0 - 1 + col;
+      __bf16 v = (__bf16)0.f;
+      if ((unsigned)id < (unsigned)cd.D && (unsigned)ih < (unsigned)cd.H &&
+          (unsigned)iw < (unsigned)cd.W && (ci0 + ci) < cd.Cin)
+        v = xn[((int64_t)(ci0 + ci) * cd.D + id) * HW + (int64_t)ih * cd.W +
+               iw];
+      sX[ci][kd][hrow][col] = v;
+    }
+    // ---- stage go tile: [co32][m=256 over (OHT x OWT)] -----------------
+    const __bf16* gon = go + (int64_t)n * cd.Cout * cd.OD * OHW;
+    for (int idx = tid; idx < 32 * 256; idx += 256) {
+      const int m = idx & 255;
+      const int co = idx >> 8;
+      const int oh = oh0 + m / OWT;
+      const int ow = ow0 + m % OWT;
+      __bf16 v = (__bf16)0.f;
+      if ((co0 + co) < cd.Cout && oh < cd.OH && ow < cd.OW)
+        v = gon[((int64_t)(co0 + co) * cd.OD + od) * OHW +
+                (int64_t)oh * cd.OW + ow];
+      sGo[co][m] = v;
+    }
+    __syncthreads();
+
+    // ---- 8 m-subchunks x 27 taps ---------------------------------------
+#pragma unroll 1
+    for (int ms = 0; ms < 8; ++ms) {
+      bf16x8 afrag;   // go[co16][m32]
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        afrag[j] = sGo[wi * 16 + row][ms * 32 + kg * 8 + j];
+      // element m = ms*32 + kg*8 + j -> oh_off = m/OWT, ow = m%OWT; the
+      // 8-run stays inside one w-row for OWT in {8,16,32}
+      const int mbase = ms * 32 + kg * 8;
+      const int oh_off = mbase / OWT;
+      const int ow_off = mbase % OWT;
+#pragma unroll
+      for (int kd = 0; kd < 3; ++kd) {
+#pragma unroll
+        for (int kh = 0; kh < 3; ++kh) {
+#pragma unroll
+          for (int kw = 0; kw < 3; ++kw) {
+            bf16x8 bfrag;
+            const __bf16* src =
+                &sX[wj * 16 + row][kd][oh_off + kh][ow_off + kw];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) bfrag[j] = src[j];
+            acc[(kd * 3 + kh) * 3 + kw] =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    afrag, bfrag, acc[(kd * 3 + kh) * 3 + kw], 0, 0, 0);
+          }
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- fold partials into dw[co][ci*27 + tap] -------------------------
+  const int K = cd.Cin * 27;
+  const int ccol = lane & 15;          // ci col within fragment
+  const int crow0 = (lane >> 4) * 4;   // co row
+#pragma unroll 1
+  for (int tp = 0; tp < 27; ++tp) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int co = co0 + wi * 16 + crow0 + r;
+      const int ci = ci0 + wj * 16 + ccol;
+      if (co < cd.Cout && ci < cd.Cin)
+        atomicAdd(&dw[(int64_t)co * K + ci * 27 + tp], acc[tp][r]);
+    }
+  }
+}
+
 // bias grad + (optionally) any channelwise sums: dB[co] = sum over m of go
 __global__ void channel_sum_kernel(const __bf16* __restrict__ go,
                                    float* __restrict__ db, int N, int C,
@@ -679,7 +811,40 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
   int64_t M = (int64_t)cd.N * cd.OD * cd.OH * cd.OW;
   auto dw = torch::zeros({cd.Cout, (int64_t)K},
                          xc.options().dtype(torch::kFloat32));
-  // split-K sized so the grid covers the chip: >= 2048 blocks total
+
+  if (stride == 1 && (cd.OW % 8) == 0) {
+    // tap-reuse path
+    int OWT = cd.OW % 32 == 0 ? 32 : (cd.OW % 16 == 0 ? 16 : 8);
+    int wtiles = (cd.OW + OWT - 1) / OWT;
+    int OHT = 256 / OWT;
+    int htiles = (cd.OH + OHT - 1) / OHT;
+    int co_t = (cd.Cout + 31) / 32, ci_t = (cd.Cin + 31) / 32;
+    int64_t nchunks = (int64_t)cd.N * cd.OD * htiles * wtiles;
+    int64_t zstride = std::max<int64_t>(
+        1, std::min<int64_t>(nchunks, 768 / std::max(co_t * ci_t, 1)));
+    dim3 grid(co_t, ci_t, (unsigned)zstride);
+    if (OWT == 32)
+      hipLaunchKernelGGL(conv3d_wgrad_s1_kernel<32>, grid, dim3(256), 0,
+                         current_stream(),
+                         reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                         reinterpret_cast<const __bf16*>(g.data_ptr()),
+                         dw.data_ptr<float>(), cd, nchunks, zstride);
+    else if (OWT == 16)
+      hipLaunchKernelGGL(conv3d_wgrad_s1_kernel<16>, grid, dim3(256), 0,
+                         current_stream(),
+                         reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                         reinterpret_cast<const __bf16*>(g.data_ptr()),
+                         dw.data_ptr<float>(), cd, nchunks, zstride);
+    else
+      hipLaunchKernelGGL(conv3d_wgrad_s1_kernel<8>, grid, dim3(256), 0,
+                         current_stream(),
+                         reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                         reinterpret_cast<const __bf16*>(g.data_ptr()),
+                         dw.data_ptr<float>(), cd, nchunks, zstride);
+    return dw.view({cd.Cout, cd.Cin, 3, 3, 3});
+  }
+
+  // fallback: split-K implicit GEMM (any shape / stride)
   int planes = ((cd.Cout + 31) / 32) * ((K + 31) / 32);
   int64_t target_chunks = std::max<int64_t>(1, 2048 / std::max(planes, 1));
   int64_t chunk = std::max<int64_t>(128, (M + target_chunks - 1) /

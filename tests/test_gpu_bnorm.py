@@ -29,19 +29,23 @@ def test_bn3d_forward_backward(dev, relu):
     from coinstac_dinunet_amd.ops.bnorm import _BN3dFn
     y = _BN3dFn.apply(x, gamma, beta, None, None, 0.1, 1e-5, relu)
 
-    xr = x.detach().clone().requires_grad_(True)
+    # reference on the SAME bf16-quantized input so the ReLU mask agrees
+    xr = x.detach().to(torch.bfloat16).float().requires_grad_(True)
     gr = gamma.detach().clone().requires_grad_(True)
     br = beta.detach().clone().requires_grad_(True)
-    ref = torch.nn.functional.batch_norm(
-        xr.float(), None, None, gr, br, training=True, eps=1e-5)
-    if relu:
-        ref = torch.relu(ref)
-    torch.testing.assert_close(y.float(), ref, rtol=5e-2, atol=5e-2)
+    z = torch.nn.functional.batch_norm(
+        xr, None, None, gr, br, training=True, eps=1e-5)
+    ref = torch.relu(z) if relu else z
+    # elements with z ~ 0 can flip the mask under bf16 rounding: exclude
+    keep = (z.abs() > 1e-2).detach()
+    torch.testing.assert_close(y.float()[keep], ref[keep],
+                               rtol=5e-2, atol=5e-2)
 
     g = torch.randn_like(ref)
     y.backward(g.to(y.dtype))
     ref.backward(g)
-    torch.testing.assert_close(x.grad, xr.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(x.grad[keep], xr.grad[keep],
+                               rtol=5e-2, atol=5e-2)
     torch.testing.assert_close(gamma.grad, gr.grad, rtol=5e-2, atol=0.5)
     torch.testing.assert_close(beta.grad, br.grad, rtol=5e-2, atol=0.5)
 

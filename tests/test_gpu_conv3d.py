@@ -53,6 +53,10 @@ def test_conv3d_fwd(dev, case):
     ref = _ref_conv(x, w, s)
     torch.testing.assert_close(out.float(), ref, rtol=5e-2,
                                atol=5e-2 * (Cin * 27) ** 0.5 * 0.2)
+    if s == 1 and W % 8 == 0:
+        out2 = C.conv3d_fwd_spatial(x, w)
+        torch.testing.assert_close(out2.float(), ref, rtol=5e-2,
+                                   atol=5e-2 * (Cin * 27) ** 0.5 * 0.2)
 
 
 @pytest.mark.parametrize('case', CASES)
@@ -68,6 +72,11 @@ def test_conv3d_dgrad(dev, case):
                         list(x.shape), s)
     torch.testing.assert_close(dx.float(), x.grad, rtol=5e-2,
                                atol=5e-2 * (Cout * 27) ** 0.5 * 0.2)
+    if s == 1 and W % 8 == 0:
+        dx2 = C.conv3d_dgrad_spatial(go.to(torch.bfloat16),
+                                     w.to(torch.bfloat16), list(x.shape))
+        torch.testing.assert_close(dx2.float(), x.grad, rtol=5e-2,
+                                   atol=5e-2 * (Cout * 27) ** 0.5 * 0.2)
 
 
 @pytest.mark.parametrize('case', CASES)

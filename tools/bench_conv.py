@@ -41,8 +41,12 @@ def main():
         OD = (D + 2 - 3) // s + 1
         go = torch.randn(B, Cout, OD, OD, OD, device=dev,
                          dtype=torch.bfloat16)
-        ms_f = t(lambda: C.conv3d_fwd(x, w, s))
-        ms_d = t(lambda: C.conv3d_dgrad(go, w, list(x.shape), s))
+        if s == 1 and D % 8 == 0:
+            ms_f = t(lambda: C.conv3d_fwd_spatial(x, w))
+            ms_d = t(lambda: C.conv3d_dgrad_spatial(go, w, list(x.shape)))
+        else:
+            ms_f = t(lambda: C.conv3d_fwd(x, w, s))
+            ms_d = t(lambda: C.conv3d_dgrad(go, w, list(x.shape), s))
         ms_w = t(lambda: C.conv3d_wgrad(x, go, s))
         xr = x.clone().requires_grad_(True)
         ms_lf = t(lambda: torch.nn.functional.conv3d(x, w, stride=s, padding=1))
