@@ -28,13 +28,35 @@ def parse_args():
     ap.add_argument('--local-iterations', type=int, default=1)
     ap.add_argument('--model', type=str, default='vbm',
                     choices=['vbm', 'mlp', 'resnet18'])
+    ap.add_argument('--stock', action='store_true',
+                    help='stock torch ops (MIOpen conv/BN) for comparison')
     return ap.parse_args()
 
 
 def build_model(args, device):
     from coinstac_dinunet_amd.models import FreeSurferMLP, ResNet18, VBMNet
     if args.model == 'vbm':
-        net = VBMNet(in_channels=1, num_class=2)
+        if args.stock:
+            import coinstac_dinunet_amd.models.vbm as _vbm
+            import torch.nn as _nn
+
+            class _StockBlock(_nn.Module):
+                def __init__(self, cin, cout, stride=1):
+                    super().__init__()
+                    self.conv = _nn.Conv3d(cin, cout, 3, stride=stride,
+                                           padding=1, bias=False)
+                    self.bn = _nn.BatchNorm3d(cout)
+                    self.act = _nn.ReLU(inplace=True)
+
+                def forward(self, x):
+                    return self.act(self.bn(self.conv(x)))
+
+            _orig = _vbm._ConvBlock
+            _vbm._ConvBlock = _StockBlock
+            net = VBMNet(in_channels=1, num_class=2)
+            _vbm._ConvBlock = _orig
+        else:
+            net = VBMNet(in_channels=1, num_class=2)
         data = torch.randn(args.batch, 1, args.vol, args.vol, args.vol)
     elif args.model == 'mlp':
         net = FreeSurferMLP(in_features=66, num_class=2)

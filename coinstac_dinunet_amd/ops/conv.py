@@ -17,7 +17,10 @@ class _Conv3dFn(torch.autograd.Function):
         C = require_native()
         xb = x.to(torch.bfloat16)
         wb = weight.to(torch.bfloat16)
-        if stride == 1 and xb.size(4) % 8 == 0:
+        # spatial tap-reuse pays when the channel tile fills and chunks
+        # are dense; else the igemm kernel wins (profiled in tools/bench_conv)
+        if (stride == 1 and xb.size(4) % 8 == 0 and xb.size(1) >= 16
+                and xb.size(3) * xb.size(4) >= 256):
             out = C.conv3d_fwd_spatial(xb, wb)
         else:
             out = C.conv3d_fwd(xb, wb, stride)
@@ -38,7 +41,9 @@ class _Conv3dFn(torch.autograd.Function):
         go = grad_out.to(torch.bfloat16).contiguous()
         gx = gw = gb = None
         if ctx.x_requires:
-            if ctx.stride == 1 and xb.size(4) % 8 == 0:
+            if (ctx.stride == 1 and xb.size(4) % 8 == 0
+                    and go.size(1) >= 16
+                    and xb.size(3) * xb.size(4) >= 256):
                 gx = C.conv3d_dgrad_spatial(go, wb,
                                             list(xb.shape)).to(ctx.in_dtype)
             else:

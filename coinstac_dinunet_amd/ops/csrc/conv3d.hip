@@ -812,7 +812,8 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
   auto dw = torch::zeros({cd.Cout, (int64_t)K},
                          xc.options().dtype(torch::kFloat32));
 
-  if (stride == 1 && (cd.OW % 8) == 0) {
+  if (stride == 1 && (cd.OW % 8) == 0 && cd.Cin >= 16 &&
+      cd.OH * cd.OW >= 256) {
     // tap-reuse path
     int OWT = cd.OW % 32 == 0 ? 32 : (cd.OW % 16 == 0 ? 16 : 8);
     int wtiles = (cd.OW + OWT - 1) / OWT;

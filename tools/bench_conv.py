@@ -41,7 +41,7 @@ def main():
         OD = (D + 2 - 3) // s + 1
         go = torch.randn(B, Cout, OD, OD, OD, device=dev,
                          dtype=torch.bfloat16)
-        if s == 1 and D % 8 == 0:
+        if s == 1 and D % 8 == 0 and Cin >= 16 and D * D >= 256:
             ms_f = t(lambda: C.conv3d_fwd_spatial(x, w))
             ms_d = t(lambda: C.conv3d_dgrad_spatial(go, w, list(x.shape)))
         else:
