@@ -589,11 +589,11 @@ template <int OWT>
 __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
     const __bf16* __restrict__ x, const __bf16* __restrict__ go,
     float* __restrict__ dw, ConvDims cd, int64_t nchunks, int64_t zstride) {
-  constexpr int OHT = 256 / OWT;
+  constexpr int OHT = 128 / OWT;  // 128-m chunks: ~36 KB LDS => 4 blocks/CU
   constexpr int W2 = OWT + 4;   // staged row length (2 halo + pad)
   constexpr int H2 = OHT + 2;
   __shared__ __bf16 sX[32][3][H2][W2];
-  __shared__ __bf16 sGo[32][256 + LDA_PAD];
+  __shared__ __bf16 sGo[32][128 + LDA_PAD];
 
   const int co0 = blockIdx.x * 32;
   const int ci0 = blockIdx.y * 32;
@@ -645,9 +645,9 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
     }
     // ---- stage go tile: [co32][m=256 over (OHT x OWT)] -----------------
     const __bf16* gon = go + (int64_t)n * cd.Cout * cd.OD * OHW;
-    for (int idx = tid; idx < 32 * 256; idx += 256) {
-      const int m = idx & 255;
-      const int co = idx >> 8;
+    for (int idx = tid; idx < 32 * 128; idx += 256) {
+      const int m = idx & 127;
+      const int co = idx >> 7;
       const int oh = oh0 + m / OWT;
       const int ow = ow0 + m % OWT;
       __bf16 v = (__bf16)0.f;
@@ -660,7 +660,7 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
 
     // ---- 8 m-subchunks x 27 taps ---------------------------------------
 #pragma unroll 1
-    for (int ms = 0; ms < 8; ++ms) {
+    for (int ms = 0; ms < 4; ++ms) {
       bf16x8 afrag;   // go[co16][m32]
 #pragma unroll
       for (int j = 0; j < 8; ++j)
@@ -817,7 +817,7 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
     // tap-reuse path
     int OWT = cd.OW % 32 == 0 ? 32 : (cd.OW % 16 == 0 ? 16 : 8);
     int wtiles = (cd.OW + OWT - 1) / OWT;
-    int OHT = 256 / OWT;
+    int OHT = 128 / OWT;
     int htiles = (cd.OH + OHT - 1) / OHT;
     int co_t = (cd.Cout + 31) / 32, ci_t = (cd.Cin + 31) / 32;
     int64_t nchunks = (int64_t)cd.N * cd.OD * htiles * wtiles;
