@@ -1,0 +1,330 @@
+"""rankDAD engine: distributed activation-gradient decomposition.
+
+Protocol parity: /root/reference/coinstac_dinunet/distrib/rankdad/
+(__init__.py:12-96, spi.py:9-250). Semantics preserved:
+  - DADParallel wraps the model; forward/backward hooks on every trainable
+    LEAF module except norm layers capture input activations and output
+    grads;
+  - dad_backward(): per layer, flatten leading dims, deflated power
+    iteration extracts the top-`dad_reduction_rank` structure of
+    grad^T @ act without forming it; ships (grad-factor [out,r,1],
+    act-factor [in,r]) per layer, layers in REVERSE child order;
+  - aggregation is CONCATENATION of per-site factors along the rank axis
+    (the sum of site outer products — NOT a mean) with optional
+    re-compression on the aggregator;
+  - synced_param_update(): weight.grad = (act_tall @ grad_tall^T)^T,
+    bias.grad = grad_tall.sum(0), walking children in reverse;
+  - gradient accumulation (local_iterations>1) is unsupported, as in the
+    reference (rankdad/__init__.py:48-49).
+"""
+import os as _os
+
+import numpy as _np
+import torch as _torch
+
+from ..utils import tensorutils as _tu
+from .learner import COINNLearner
+from .reducer import COINNReducer
+
+_SKIP_NORM_LAYERS = (_torch.nn.BatchNorm1d, _torch.nn.BatchNorm2d,
+                     _torch.nn.BatchNorm3d, _torch.nn.LayerNorm,
+                     _torch.nn.GroupNorm)
+
+DAD_FILE = 'dad_data.npy'
+DAD_AGG_FILE = 'reduced_dad_data.npy'
+
+
+def power_iteration_BC(B, C, rank=10, numiterations=5, tol=1e-3):
+    """Top-`rank` structure of G = B @ C^T without forming G.
+
+    B [n, k], C [m, k] (k = flattened batch). Returns (Bf [n, r], Cf [m, r])
+    with G ~= Bf @ Cf^T; Bf columns carry the singular values. Deflated
+    power iteration: each new direction is orthogonalized against the
+    already-extracted ones via the accumulated sigma^2-weighted projector.
+    """
+    device = B.device
+    n, k = B.shape
+    m = C.shape[0]
+    small_k = k <= m  # work through the k x k Gram when cheaper
+    if small_k:
+        BCT = B @ C.t()          # [n, m] — small side
+        G2 = BCT @ BCT.t()       # [n, n]
+    else:
+        CC = C.t() @ C           # [k, k]
+
+    bs, cs, sigmas = [], [], []
+    lam1 = None  # first (largest) eigenvalue of G G^T, sets the zero scale
+    for _ in range(rank):
+        b = _torch.rand(n, device=device)
+        degenerate = False
+        for _ in range(numiterations):
+            if small_k:
+                v = G2 @ b
+            else:
+                v = B @ (CC @ (B.t() @ b))
+            # deflate: project out the already-extracted directions
+            for bb in bs:
+                v = v - bb * (bb @ v)
+            norm = _torch.norm(v)
+            if lam1 is None:
+                pass
+            elif norm < tol * lam1 or norm < 1e-12:
+                degenerate = True
+                break
+            if norm < 1e-12:
+                degenerate = True
+                break
+            b = v / norm
+        if degenerate:
+            break
+        # keep b orthogonal to previous components before reading sigma
+        for bb in bs:
+            b = b - bb * (bb @ b)
+        bn = _torch.norm(b)
+        if bn < 1e-12:
+            break
+        b = b / bn
+        if lam1 is None:
+            lam1 = norm  # ||G G^T b|| at convergence ~ sigma_1^2
+        if small_k:
+            Bv = BCT.t() @ b
+            sigma = _torch.sqrt((Bv * Bv).sum())
+        else:
+            Bv = B.t() @ b
+            sigma = _torch.sqrt(_torch.clamp(Bv @ (CC @ Bv), min=0.0))
+        if _torch.isnan(sigma) or sigma < 1e-12 or \
+                (lam1 is not None and sigma * sigma < tol * lam1):
+            break
+        c = (C @ (B.t() @ b)) / sigma
+        bs.append(b)
+        cs.append(c)
+        sigmas.append(sigma)
+
+    if not bs:  # degenerate input: one zero component
+        return (_torch.zeros(n, 1, device=device),
+                _torch.zeros(m, 1, device=device))
+    Bf = _torch.stack([s * b for b, s in zip(bs, sigmas)], 1)
+    Cf = _torch.stack(cs, 1)
+    return Bf, Cf
+
+
+def _trainable_leaf(module):
+    if isinstance(module, _SKIP_NORM_LAYERS):
+        return False
+    return len(list(module.parameters(recurse=False))) > 0 or \
+        (len(list(module.children())) == 0 and
+         len(list(module.parameters())) > 0)
+
+
+def _mm_flatten(*tensors):
+    if tensors[0].dim() > 2:
+        return [t.flatten(0, t.dim() - 2) for t in tensors]
+    return list(tensors)
+
+
+class DADParallel(_torch.nn.Module):
+    """Model wrapper capturing per-leaf activations/output-grads."""
+
+    def __init__(self, module, cache=None, input=None, state=None,
+                 device=None, dtype='float32', **kw):
+        super().__init__()
+        self.module = module.module if isinstance(module, DADParallel) else module
+        self.cache = cache if cache is not None else {}
+        self.input = input if input is not None else {}
+        self.state = state if state is not None else {}
+        self.device = device
+        self.dtype = dtype
+        self.rank = self.cache.setdefault('dad_reduction_rank', 10)
+        self.num_pow_iters = self.cache.setdefault('dad_num_pow_iters', 5)
+        self.dad_tol = self.cache.setdefault('dad_tol', 1e-3)
+        self._reset()
+
+    def _reset(self):
+        self._fw_handles = []
+        self._bk_handles = []
+        self._activations = {}
+        self._grads = {}
+
+    # ---- leaf discovery (stable reverse order) --------------------------
+    def _leaves(self):
+        out = []
+
+        def walk(name, mod):
+            children = list(mod.named_children())
+            if children:
+                for cn, child in children:
+                    walk(f'{name}.{cn}' if name else cn, child)
+            elif _trainable_leaf(mod):
+                out.append((name, mod))
+
+        for cn, child in self.module.named_children():
+            walk(cn, child)
+        return out
+
+    # ---- hooks -----------------------------------------------------------
+    def _hook(self):
+        def fw(key):
+            def fn(mod, inputs, output):
+                if inputs and inputs[0] is not None:
+                    self._activations[key] = inputs[0]
+            return fn
+
+        def bk(key):
+            def fn(mod, grad_in, grad_out):
+                for g in grad_out:
+                    if g is not None:
+                        self._grads[key] = g
+                        break
+            return fn
+
+        for name, mod in self._leaves():
+            self._fw_handles.append(mod.register_forward_hook(fw(name)))
+            self._bk_handles.append(mod.register_full_backward_hook(bk(name)))
+
+    def _unhook(self):
+        for h in self._fw_handles + self._bk_handles:
+            h.remove()
+        self._fw_handles, self._bk_handles = [], []
+
+    def train(self, mode=True):
+        self.module.train(mode)
+        if mode:
+            if not self._fw_handles:
+                self._hook()
+        else:
+            self._unhook()
+        return self
+
+    def eval(self):
+        return self.train(False)
+
+    def forward(self, *args, **kw):
+        if self.training:
+            self._activations = {}
+            self._grads = {}
+        return self.module(*args, **kw)
+
+    # ---- DAD rounds -------------------------------------------------------
+    def dad_backward(self):
+        """Compress every leaf's (grad, act) and ship (reverse leaf order)."""
+        out = {'dad_data': DAD_FILE}
+        data = []
+        for name, mod in self._leaves()[::-1]:
+            if name not in self._grads or name not in self._activations:
+                continue
+            grad, act = _mm_flatten(self._grads[name].detach().float(),
+                                    self._activations[name].detach().float())
+            gf, af = power_iteration_BC(grad.t(), act.t(), self.rank,
+                                        self.num_pow_iters, self.dad_tol)
+            data.append([gf.unsqueeze(-1).cpu().numpy().astype(self.dtype),
+                        af.cpu().numpy().astype(self.dtype)])
+        _tu.save_arrays(self.state['transferDirectory'] + _os.sep + DAD_FILE,
+                        data)
+        return out
+
+    def synced_param_update(self):
+        """Rebuild grads from the aggregator's concatenated factors."""
+        path = self.state['baseDirectory'] + _os.sep + self.input['reduced_dad_data']
+        data = list(_tu.load_arrays(path))
+        for (name, mod), (gf, af) in zip(self._leaves()[::-1], data):
+            params = dict(mod.named_parameters(recurse=False))
+            grad_tall = _torch.from_numpy(_np.asarray(gf)).float() \
+                .to(self.device).squeeze(-1).t()            # [r, out]
+            act_tall = _torch.from_numpy(_np.asarray(af)).float() \
+                .to(self.device)                            # [in, r]
+            if 'weight' in params and params['weight'].grad is not None:
+                params['weight'].grad.data = \
+                    act_tall.mm(grad_tall).t().contiguous()
+            elif 'weight' in params:
+                params['weight'].grad = \
+                    act_tall.mm(grad_tall).t().contiguous()
+            if params.get('bias') is not None:
+                params['bias'].grad = grad_tall.sum(0)
+
+    # passthroughs
+    def state_dict(self, *a, **kw):
+        return self.module.state_dict(*a, **kw)
+
+    def load_state_dict(self, *a, **kw):
+        return self.module.load_state_dict(*a, **kw)
+
+    def parameters(self, recurse=True):
+        return self.module.parameters(recurse)
+
+    def named_parameters(self, *a, **kw):
+        return self.module.named_parameters(*a, **kw)
+
+
+class DADLearner(COINNLearner):
+    def __init__(self, trainer=None, mp_pool=None, **kw):
+        super().__init__(trainer=trainer, mp_pool=mp_pool, **kw)
+        for fk in self.trainer.nn:
+            if not isinstance(self.trainer.nn[fk], DADParallel):
+                self.trainer.nn[fk] = DADParallel(
+                    self.trainer.nn[fk], cache=self.cache, input=self.input,
+                    state=self.state, device=self.device, dtype=self.dtype)
+            else:  # refresh per-round frozen input
+                self.trainer.nn[fk].input = self.input
+
+    def step(self):
+        out = {}
+        self.trainer.nn[self.first_model].synced_param_update()
+        self.trainer.optimizer[self.first_optim].step()
+        return out
+
+    def forward(self):
+        out = {}
+        model = self.trainer.nn[self.first_model]
+        model.train()
+        self.trainer.optimizer[self.first_optim].zero_grad()
+        its = []
+        for _ in range(self.cache.get('local_iterations', 1)):
+            batch, nxt_iter_out = self.trainer.data_handle.next_iter()
+            it = self.trainer.iteration(batch)
+            it['loss'].backward()
+            its.append(it)
+            out.update(**nxt_iter_out)
+            break  # DAD does not support gradient accumulation
+        return self.trainer.reduce_iteration(its), out
+
+    def to_reduce(self):
+        model = self.trainer.nn[self.first_model]
+        model.train()
+        it, out = self.forward()
+        out.update(**model.dad_backward())
+        out['reduce'] = True
+        return it, out
+
+
+class DADReducer(COINNReducer):
+    """Concatenate per-site factors along rank; optionally re-compress."""
+
+    def __init__(self, trainer=None, mp_pool=None, **kw):
+        super().__init__(trainer=trainer, mp_pool=mp_pool, **kw)
+        self.rank = self.cache.setdefault('dad_reduction_rank', 10)
+        self.num_pow_iters = self.cache.setdefault('dad_num_pow_iters', 5)
+        self.dad_tol = self.cache.setdefault('dad_tol', 1e-3)
+
+    def reduce(self):
+        out = {'reduced_dad_data': DAD_AGG_FILE}
+        site_data = self._load('dad_data')
+        reduced = []
+        for layer_parts in zip(*site_data):
+            grads, acts = zip(*layer_parts)
+            grad = _torch.cat([_torch.from_numpy(_np.asarray(g)).float()
+                               .to(self.device) for g in grads], 1).squeeze(-1)
+            act = _torch.cat([_torch.from_numpy(_np.asarray(a)).float()
+                              .to(self.device) for a in acts], 1)
+            if grad.shape[1] > self.rank:
+                grad, act = self._recompress(grad, act)
+            reduced.append([grad.unsqueeze(-1).cpu().numpy().astype(self.dtype),
+                            act.cpu().numpy().astype(self.dtype)])
+        _tu.save_arrays(self.state['transferDirectory'] + _os.sep +
+                        DAD_AGG_FILE, reduced)
+        out['update'] = True
+        return out
+
+    def _recompress(self, grad, act):
+        """grad [out, R], act [in, R] -> rank-r of grad @ act^T."""
+        return power_iteration_BC(grad, act, self.rank, self.num_pow_iters,
+                                  self.dad_tol)

@@ -55,11 +55,20 @@ def assign_grads(model, grads, device=None):
         p.grad = t
 
 
+def _to_obj(a):
+    if isinstance(a, (list, tuple)):
+        inner = np.empty(len(a), dtype=object)
+        for i, x in enumerate(a):
+            inner[i] = _to_obj(x)
+        return inner
+    return np.asarray(a)
+
+
 def save_arrays(path, arrays):
-    """np.save of an object array of per-parameter arrays (wire format)."""
+    """np.save of an object array of (possibly nested) arrays (wire format)."""
     obj = np.empty(len(arrays), dtype=object)
     for i, a in enumerate(arrays):
-        obj[i] = np.asarray(a)
+        obj[i] = _to_obj(a)
     np.save(path, obj)
 
 

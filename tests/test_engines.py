@@ -1,0 +1,117 @@
+"""PowerSGD and rankDAD engines: unit math + full loopback protocol runs."""
+import numpy as np
+import pytest
+import torch
+
+from coinstac_dinunet_amd import COINNLocal, COINNRemote
+from coinstac_dinunet_amd.config.keys import Key, Mode
+from coinstac_dinunet_amd.distrib.powersgd import orthogonalize
+from coinstac_dinunet_amd.distrib.rankdad import (DADParallel,
+                                                  power_iteration_BC)
+from coinstac_dinunet_amd.simulator import LoopbackCluster
+
+from computations import TabularDataset, TabularTrainer, make_site_data
+
+
+def test_orthogonalize_columns():
+    torch.manual_seed(0)
+    m = torch.randn(20, 4)
+    orthogonalize(m)
+    gram = m.t() @ m
+    torch.testing.assert_close(gram, torch.eye(4), rtol=1e-4, atol=1e-4)
+
+
+def test_power_iteration_low_rank_recovery():
+    torch.manual_seed(1)
+    # construct an exactly rank-3 G = B @ C^T
+    U = torch.randn(30, 3)
+    V = torch.randn(20, 3)
+    G = U @ V.t()
+    # feed as B [n,k], C [m,k] with k=20... use B=G, C=I
+    B, C = G, torch.eye(20)
+    Bf, Cf = power_iteration_BC(B, C, rank=6, numiterations=30, tol=1e-6)
+    recon = Bf @ Cf.t()
+    assert torch.norm(recon - G) / torch.norm(G) < 0.05
+
+
+def test_dad_parallel_hooks_capture():
+    torch.manual_seed(2)
+    net = torch.nn.Sequential(torch.nn.Linear(8, 4), torch.nn.ReLU(),
+                              torch.nn.Linear(4, 2))
+    wrapped = DADParallel(net, cache={}, input={}, state={},
+                          device=torch.device('cpu'))
+    wrapped.train()
+    x = torch.randn(5, 8)
+    out = wrapped(x)
+    out.sum().backward()
+    leaves = [n for n, _ in wrapped._leaves()]
+    assert len(leaves) == 2  # two Linear leaves, ReLU skipped
+    for n in leaves:
+        assert n in wrapped._activations
+        assert n in wrapped._grads
+    # grad reconstruction sanity: rank-full factors reproduce W grads
+    g = wrapped._grads[leaves[0]]
+    a = wrapped._activations[leaves[0]]
+    gf, af = power_iteration_BC(g.t(), a.t(), rank=8, numiterations=50,
+                                tol=1e-9)
+    recon = (af @ gf.t()).t()
+    ref = g.t() @ a
+    assert torch.norm(recon - ref) / torch.norm(ref) < 0.05
+
+
+def _run_cluster(tmp_path, agg_engine, extra_kw=None, epochs=1):
+    cluster = LoopbackCluster(
+        str(tmp_path), n_sites=2,
+        site_data=lambda s: make_site_data(s.as_dict(), n_samples=16,
+                                           seed=int(s.clientId[-1])))
+    kw = dict(task_id='tab', mode=Mode.TRAIN, batch_size=4, epochs=epochs,
+              validation_epochs=1, local_iterations=1,
+              split_ratio=(0.6, 0.2, 0.2), data_dir='data', num_class=2,
+              seed_all=True, patience=epochs, verbose=False,
+              agg_engine=agg_engine)
+    kw.update(extra_kw or {})
+
+    def make_local(cache, input, state):
+        return COINNLocal(cache=cache, input=input, state=state, **kw)
+
+    def make_remote(cache, input, state):
+        return COINNRemote(cache=cache, input=input, state=state)
+
+    success, out = cluster.run(make_local, make_remote, TabularTrainer,
+                               dataset_cls=TabularDataset, max_rounds=500)
+    return cluster, success
+
+
+def test_powersgd_full_protocol(tmp_path):
+    cluster, success = _run_cluster(
+        tmp_path, 'powerSGD',
+        extra_kw=dict(matrix_approximation_rank=2, start_powerSGD_iter=2,
+                      seed=7))
+    assert success
+    assert cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
+    # compression actually engaged (iter counter passed warmup)
+    assert cluster.site_caches[0]['powerSGD_iter'] > 2
+    # sites stay weight-synchronized under compression
+    m0 = cluster.site_caches[0]['nn']['net']
+    m1 = cluster.site_caches[1]['nn']['net']
+    for p0, p1 in zip(m0.parameters(), m1.parameters()):
+        assert torch.allclose(p0, p1, atol=1e-5)
+
+
+def test_rankdad_full_protocol(tmp_path):
+    cluster, success = _run_cluster(
+        tmp_path, 'rankDAD', extra_kw=dict(dad_reduction_rank=4))
+    assert success
+    assert cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
+    # DAD layers (non-norm) stay weight-synchronized across sites
+    m0 = cluster.site_caches[0]['nn']['net']
+    m1 = cluster.site_caches[1]['nn']['net']
+    net0 = m0.module if hasattr(m0, 'module') else m0
+    net1 = m1.module if hasattr(m1, 'module') else m1
+    norm_prefixes = {n for n, m in net0.named_modules()
+                     if isinstance(m, torch.nn.BatchNorm1d)}
+    for (n0, p0), (n1, p1) in zip(net0.named_parameters(),
+                                  net1.named_parameters()):
+        if n0.rsplit('.', 1)[0] in norm_prefixes:
+            continue  # norm layers stay local under DAD (by design)
+        assert torch.allclose(p0, p1, atol=1e-4), n0
