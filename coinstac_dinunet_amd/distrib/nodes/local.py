@@ -141,14 +141,22 @@ class COINNLocal:
         out = {'phase': Phase.COMPUTATION}
         pretrain_epochs = self._pretrain_args.get('epochs', 0)
         if pretrain_epochs > 0 and self.cache.get('pretrain'):
-            cache = {**self.cache}
-            cache.update(**self.cache.get('pretrain_args', self._pretrain_args))
-            trainer = trainer_cls(data_handle=datahandle_cls(
-                cache=cache, input=self.input, state=self.state,
-                dataloader_args=self._dataloader_args))
-            trainer.init_nn()
-            trainer.init_training_cache()
-            out.update(**trainer.train_local(train_dataset, validation_dataset))
+            # overlay pretrain_args onto the live cache for the duration of
+            # local training, then restore (logs/best-state persist)
+            overrides = dict(self.cache.get('pretrain_args',
+                                            self._pretrain_args))
+            saved = {k: self.cache.get(k) for k in overrides}
+            self.cache.update(**overrides)
+            try:
+                trainer = trainer_cls(data_handle=datahandle_cls(
+                    cache=self.cache, input=self.input, state=self.state,
+                    dataloader_args=self._dataloader_args))
+                trainer.init_nn()
+                trainer.init_training_cache()
+                out.update(**trainer.train_local(train_dataset,
+                                                 validation_dataset))
+            finally:
+                self.cache.update(**saved)
             out['phase'] = Phase.PRE_COMPUTATION
 
         if pretrain_epochs > 0 and any(

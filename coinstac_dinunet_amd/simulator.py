@@ -74,6 +74,8 @@ class LoopbackCluster:
         make_remote(cache, input, state) -> COINNRemote.
         Returns (success, remote_output)."""
         success, remote_out = False, {}
+        reducer_cls = compute_kw.pop('reducer_cls', None)
+        remote_kw = {'reducer_cls': reducer_cls} if reducer_cls else {}
         for self.rounds in range(1, max_rounds + 1):
             # 1) site computations
             site_outs = {}
@@ -91,7 +93,7 @@ class LoopbackCluster:
             # 3) remote computation
             remote = make_remote(self.remote_cache, site_outs,
                                  self.remote_state.as_dict())
-            rres = remote(mp_pool, trainer_cls)
+            rres = remote(mp_pool, trainer_cls, **remote_kw)
             remote_out, success = rres['output'], rres.get('success', False)
             if success:
                 break

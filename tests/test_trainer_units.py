@@ -1,0 +1,119 @@
+"""Unit tests: checkpoint format, train_local, site runner, FedAvg engine."""
+import json
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from coinstac_dinunet_amd import COINNLocal, COINNRemote
+from coinstac_dinunet_amd.config.keys import Key, Mode
+from coinstac_dinunet_amd.simulator import LoopbackCluster
+
+from computations import TabularDataset, TabularTrainer, make_site_data
+
+
+def _mk_trainer(tmp_path, n=16):
+    from coinstac_dinunet_amd.data import COINNDataHandle
+    state = {'baseDirectory': str(tmp_path), 'outputDirectory': str(tmp_path),
+             'transferDirectory': str(tmp_path), 'clientId': 'local0'}
+    make_site_data(state, n_samples=n, seed=0)
+    cache = {'task_id': 'tab', 'batch_size': 4, 'epochs': 2,
+             'validation_epochs': 1, 'learning_rate': 0.01, 'num_class': 2,
+             'monitor_metric': 'f1', 'metric_direction': 'maximize',
+             'log_dir': str(tmp_path / 'logs'), 'data_dir': 'data',
+             'load_limit': 10 ** 9, 'verbose': False, 'num_folds': None,
+             'split_ratio': (0.7, 0.15, 0.15), 'local_iterations': 1}
+    os.makedirs(cache['log_dir'], exist_ok=True)
+    handle = COINNDataHandle(cache=cache, input={}, state=state)
+    trainer = TabularTrainer(data_handle=handle)
+    return trainer, cache, state
+
+
+def test_checkpoint_roundtrip_multi_model(tmp_path):
+    """Checkpoint keeps EVERY model/optimizer entry (the reference's
+    last-entry-only bug is fixed — basetrainer.py:104-114)."""
+    trainer, cache, state = _mk_trainer(tmp_path)
+    trainer.init_nn(init_weights=True)
+    trainer.nn['aux'] = torch.nn.Linear(3, 2)
+    trainer.optimizer['aux_opt'] = torch.optim.SGD(
+        trainer.nn['aux'].parameters(), lr=0.1)
+    path = str(tmp_path / 'chk.pt')
+    trainer.save_checkpoint(path)
+    chk = torch.load(path, weights_only=False)
+    assert chk['source'] == 'coinstac'
+    assert set(chk['models']) == {'net', 'aux'}
+    assert set(chk['optimizers']) == {'adam', 'aux_opt'}
+
+    # mutate then load back
+    with torch.no_grad():
+        for p in trainer.nn['net'].parameters():
+            p.add_(1.0)
+    trainer.load_checkpoint(path)
+    chk2 = chk['models']['net']
+    for (k, v) in trainer.nn['net'].state_dict().items():
+        torch.testing.assert_close(v, chk2[k])
+
+
+def test_train_local_runs_and_logs(tmp_path):
+    trainer, cache, state = _mk_trainer(tmp_path, n=20)
+    trainer.data_handle.prepare_data()
+    cache['split_file'] = cache['splits']['0']
+    trainer.init_nn(init_weights=True)
+    trainer.init_training_cache()
+    train_d = trainer.data_handle.get_train_dataset(TabularDataset)
+    val_d = trainer.data_handle.get_validation_dataset(TabularDataset)
+    out = trainer.train_local(train_d, val_d)
+    assert len(cache[Key.TRAIN_LOG]) > 0
+    assert len(cache[Key.VALIDATION_LOG]) > 0
+    # improvement at least once => weights.tar shipped
+    if out.get('weights_file'):
+        assert os.path.exists(os.path.join(state['transferDirectory'],
+                                           out['weights_file']))
+
+
+def test_site_runner(tmp_path):
+    from coinstac_dinunet_amd.site_runner import SiteRunner
+    base = tmp_path / 'input' / 'local0' / 'simulatorRun'
+    os.makedirs(base)
+    make_site_data({'baseDirectory': str(base)}, n_samples=16, seed=1)
+    runner = SiteRunner(task_id='tab', data_path=str(tmp_path), site_index=0,
+                        mode=Mode.TRAIN, batch_size=4, epochs=1,
+                        split_ratio=(0.7, 0.15, 0.15), data_dir='data',
+                        num_class=2, patience=1)
+    cache = runner.run(TabularTrainer, TabularDataset,
+                       pretrain_args={'epochs': 1})
+    assert cache.get(Key.TRAIN_LOG) is not None
+
+
+def test_fedavg_custom_engine(tmp_path):
+    """Custom engine injection via compute(learner_cls=..., reducer_cls=...)"""
+    from coinstac_dinunet_amd.distrib.fedavg import (FedAvgLearner,
+                                                     FedAvgReducer)
+    cluster = LoopbackCluster(
+        str(tmp_path), n_sites=2,
+        site_data=lambda s: make_site_data(s.as_dict(), n_samples=16,
+                                           seed=int(s.clientId[-1])))
+    kw = dict(task_id='tab', mode=Mode.TRAIN, batch_size=4, epochs=1,
+              validation_epochs=1, local_iterations=2,
+              split_ratio=(0.6, 0.2, 0.2), data_dir='data', num_class=2,
+              patience=1, verbose=False, agg_engine='fedAvg')
+
+    def make_local(cache, input, state):
+        return COINNLocal(cache=cache, input=input, state=state, **kw)
+
+    def make_remote(cache, input, state):
+        return COINNRemote(cache=cache, input=input, state=state)
+
+    success, _ = cluster.run(make_local, make_remote, TabularTrainer,
+                             dataset_cls=TabularDataset, max_rounds=300,
+                             learner_cls=FedAvgLearner,
+                             reducer_cls=FedAvgReducer)
+    assert success
+    m0 = cluster.site_caches[0]['nn']['net']
+    m1 = cluster.site_caches[1]['nn']['net']
+    diverged = any(not torch.allclose(p0, p1, atol=1e-6)
+                   for p0, p1 in zip(m0.parameters(), m1.parameters()))
+    # after the last averaging round sites may have trained locally again
+    # (weights equal only right after step); just assert the run finished
+    assert cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
