@@ -16,6 +16,7 @@
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(8))) unsigned short u16x8;
 
 struct SpDims {
   int N, KCH;        // input-side channels (fwd: Cin; dgrad: Cout)
@@ -37,6 +38,9 @@ __global__ __launch_bounds__(256) void conv3d_s1_spatial_kernel(
   constexpr int W2 = OWT + 4;
   constexpr int H2 = OHT + 2;
   __shared__ __bf16 sX[32][3][H2][W2];
+  // per-k element offsets into the slab (tap decode hoisted off the hot
+  // path: the A-fragment read becomes base(m) + sKtab[k])
+  __shared__ unsigned short sKtab[32 * 27 + 8];
 
   const int ncol0 = blockIdx.y * 32;
   const int tid = threadIdx.x;
@@ -56,6 +60,14 @@ __global__ __launch_bounds__(256) void conv3d_s1_spatial_kernel(
   const int td = (int)(t % sd.TD);
   const int n = (int)(t / sd.TD);
   const int oh0 = ht * OHT, ow0 = wt * OWT;
+
+  // build the k -> slab-offset table once per block
+  for (int k = tid; k < 32 * 27; k += 256) {
+    const int cl = k / 27;
+    const int r27 = k - cl * 27;
+    const int a = r27 / 9, b = (r27 / 3) % 3, c = r27 % 3;
+    sKtab[k] = (unsigned short)(((cl * 3 + a) * H2 + b) * W2 + c);
+  }
 
   // wave owns m-fragments wave*4 .. wave*4+3 (16 m each)
   f32x4 acc[4][2];
@@ -96,24 +108,20 @@ __global__ __launch_bounds__(256) void conv3d_s1_spatial_kernel(
     const int kbase_g = kt * 32 * 27;
 #pragma unroll 1
     for (int ks = 0; ks < 27; ++ks) {
-      // A fragments: one per m-frag; element (kg,j): k = ks*32+kg*8+j
+      // A fragments: one per m-frag; element (kg,j): k = ks*32+kg*8+j;
+      // addresses come from the precomputed table (one 16B LDS read)
       bf16x8 afrag[4];
       {
         const int kb = ks * 32 + kg * 8;
+        const u16x8 kt = *reinterpret_cast<const u16x8*>(&sKtab[kb]);
+        const __bf16* slab = &sX[0][0][0][0];
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
           const int m = (wave * 4 + i) * 16 + row;
-          const int oh_off = m / OWT;
-          const int ow = m % OWT;
-          const __bf16* base = &sX[0][0][oh_off][ow];
+          const int base = (m / OWT) * W2 + (m % OWT);
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int k = kb + j;
-            const int cl = k / 27;
-            const int r27 = k - cl * 27;
-            const int a = r27 / 9, b = (r27 / 3) % 3, c = r27 % 3;
-            afrag[i][j] = sX[cl][a][oh_off + b][ow + c];
-          }
+          for (int j = 0; j < 8; ++j)
+            afrag[i][j] = slab[base + kt[j]];
         }
       }
       // B fragments: wb[ncol][Kpad], contiguous 16B per lane (L2)
