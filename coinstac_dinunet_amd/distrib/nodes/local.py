@@ -258,8 +258,13 @@ class COINNLocal:
         return learner_cls
 
     def __call__(self, *args, **kwargs):
+        t0 = _time.time()
         try:
             self.compute(*args, **kwargs)
+            timings = self.cache.setdefault('round_timings', [])
+            timings.append([self.out.get('phase'), self.out.get('mode'),
+                            round(_time.time() - t0, 4)])
+            del timings[:-1000]  # bounded trace
             return {'output': self.out}
         except Exception:
             _tback.print_exc()

@@ -7,6 +7,7 @@ reduction, best-model signaling and the results zip.
 """
 import datetime as _datetime
 import os as _os
+import time as _time
 import shutil as _shutil
 import traceback as _tback
 
@@ -279,8 +280,13 @@ class COINNRemote:
         return reducer_cls
 
     def __call__(self, *args, **kwargs):
+        t0 = _time.time()
         try:
             self.compute(*args, **kwargs)
+            timings = self.cache.setdefault('round_timings', [])
+            timings.append([self.out.get('phase'),
+                            round(_time.time() - t0, 4)])
+            del timings[:-1000]
             return {'output': self.out,
                     'success': check(all, 'phase', Phase.SUCCESS, self.input)}
         except Exception:

@@ -45,8 +45,11 @@ class RankSiteState:
 
 
 class RcclCluster:
-    def __init__(self, root, local_kw=None, remote_kw=None):
-        self.rank, self.world_size = init_distributed()
+    def __init__(self, root, local_kw=None, remote_kw=None, timeout_s=300):
+        # unlike the reference (a crashed site stalls the all-site quorum
+        # forever — SURVEY §5.3), the process group carries a timeout: a
+        # dead rank aborts the collective with an error on every survivor
+        self.rank, self.world_size = init_distributed(timeout_s=timeout_s)
         self.root = root
         self.site = RankSiteState(root, self.rank)
         self.remote_state = {
