@@ -82,25 +82,45 @@ __global__ __launch_bounds__(256) void conv3d_s1_spatial_kernel(
 
   for (int kt = 0; kt < kts; ++kt) {
     // ---- stage the input slab for channels [kt*32, kt*32+32) ----------
-    constexpr int XELEMS = 32 * 3 * H2 * W2;
+    // row-wise: each thread owns whole (ch, kd, h) rows; the OWT interior
+    // columns are IN-BOUNDS by construction (OWT divides W, stride 1), so
+    // they load as OWT/8 16-byte vectors; only the 4 halo columns are
+    // bounds-checked scalars. (The elementwise form was 135 scalar
+    // loads/thread and made the kernel VALU-bound.)
+    constexpr int NROWS = 32 * 3 * H2;
     if (kt) __syncthreads();
-    for (int idx = tid; idx < XELEMS; idx += 256) {
-      const int col = idx % W2;
-      int r = idx / W2;
+    for (int r = tid; r < NROWS; r += 256) {
       const int hrow = r % H2;
-      r /= H2;
-      const int a = r % 3;
-      const int c = r / 3;
+      const int a = (r / H2) % 3;
+      const int c = r / (3 * H2);
       const int id = td - 1 + a;
       const int ih = oh0 - 1 + hrow;
-      const int iw = ow0 - 1 + col;
       const int ch = kt * 32 + c;
-      __bf16 v = (__bf16)0.f;
-      if ((unsigned)id < (unsigned)sd.D && (unsigned)ih < (unsigned)sd.H &&
-          (unsigned)iw < (unsigned)sd.W && ch < sd.KCH)
-        v = in[in_n + ((int64_t)ch * sd.D + id) * HW + (int64_t)ih * sd.W +
-               iw];
-      sX[c][a][hrow][col] = v;
+      __bf16* dst = &sX[c][a][hrow][0];
+      const bool row_ok = (unsigned)id < (unsigned)sd.D &&
+                          (unsigned)ih < (unsigned)sd.H && ch < sd.KCH;
+      if (!row_ok) {
+#pragma unroll
+        for (int col = 0; col < W2; ++col) dst[col] = (__bf16)0.f;
+        continue;
+      }
+      const __bf16* src = in + in_n + ((int64_t)ch * sd.D + id) * HW +
+                          (int64_t)ih * sd.W;
+      // left halo (iw = ow0-1)
+      dst[0] = (ow0 > 0) ? src[ow0 - 1] : (__bf16)0.f;
+      // interior: iw = ow0 .. ow0+OWT-1 (aligned 16B when ow0%8==0)
+#pragma unroll
+      for (int v = 0; v < OWT / 8; ++v) {
+        bf16x8 vec = *reinterpret_cast<const bf16x8*>(src + ow0 + v * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) dst[1 + v * 8 + j] = vec[j];
+      }
+      // right halo (iw = ow0+OWT .. ow0+OWT+2)
+#pragma unroll
+      for (int e = 0; e < 3; ++e) {
+        const int iw = ow0 + OWT + e;
+        dst[1 + OWT + e] = (iw < sd.W) ? src[iw] : (__bf16)0.f;
+      }
     }
     __syncthreads();
 
