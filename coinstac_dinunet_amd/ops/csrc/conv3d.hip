@@ -623,38 +623,57 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
     const int n = (int)(t / cd.OD);
     const int oh0 = ht * OHT, ow0 = wt * OWT;
 
-    // ---- stage x slab: [ci32][id=od-1+kd][ih=oh0-1..][iw=ow0-1..] ------
-    constexpr int XELEMS = 32 * 3 * H2 * W2;
+    // ---- stage x slab row-wise (16B interior vectors, scalar halo) -----
+    constexpr int NXROWS = 32 * 3 * H2;
     const __bf16* xn = x + (int64_t)n * cd.Cin * cd.D * HW;
-    for (int idx = tid; idx < XELEMS; idx += 256) {
-      const int col = idx % W2;
-      int r = idx / W2;
+    for (int r = tid; r < NXROWS; r += 256) {
       const int hrow = r % H2;
-      r /= H2;
-      const int kd = r % 3;
-      const int ci = r / 3;
-      const int id = od - 1 + kd;
+      const int a = (r / H2) % 3;
+      const int ci = r / (3 * H2);
+      const int id = od - 1 + a;
       const int ih = oh0 - 1 + hrow;
-      const int iw = ow0 - 1 + col;
-      __bf16 v = (__bf16)0.f;
-      if ((unsigned)id < (unsigned)cd.D && (unsigned)ih < (unsigned)cd.H &&
-          (unsigned)iw < (unsigned)cd.W && (ci0 + ci) < cd.Cin)
-        v = xn[((int64_t)(ci0 + ci) * cd.D + id) * HW + (int64_t)ih * cd.W +
-               iw];
-      sX[ci][kd][hrow][col] = v;
+      __bf16* dst = &sX[ci][a][hrow][0];
+      const bool row_ok = (unsigned)id < (unsigned)cd.D &&
+                          (unsigned)ih < (unsigned)cd.H &&
+                          (ci0 + ci) < cd.Cin;
+      if (!row_ok) {
+#pragma unroll
+        for (int col = 0; col < W2; ++col) dst[col] = (__bf16)0.f;
+        continue;
+      }
+      const __bf16* src = xn + ((int64_t)(ci0 + ci) * cd.D + id) * HW +
+                          (int64_t)ih * cd.W;
+      dst[0] = (ow0 > 0) ? src[ow0 - 1] : (__bf16)0.f;
+#pragma unroll
+      for (int v = 0; v < OWT / 8; ++v) {
+        bf16x8 vec = *reinterpret_cast<const bf16x8*>(src + ow0 + v * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) dst[1 + v * 8 + j] = vec[j];
+      }
+#pragma unroll
+      for (int e = 0; e < 3; ++e) {
+        const int iw = ow0 + OWT + e;
+        dst[1 + OWT + e] = (iw < cd.W) ? src[iw] : (__bf16)0.f;
+      }
     }
-    // ---- stage go tile: [co32][m=256 over (OHT x OWT)] -----------------
+    // ---- stage go tile row-wise: [co32][m over (OHT x OWT)] ------------
     const __bf16* gon = go + (int64_t)n * cd.Cout * cd.OD * OHW;
-    for (int idx = tid; idx < 32 * 128; idx += 256) {
-      const int m = idx & 127;
-      const int co = idx >> 7;
-      const int oh = oh0 + m / OWT;
-      const int ow = ow0 + m % OWT;
-      __bf16 v = (__bf16)0.f;
-      if ((co0 + co) < cd.Cout && oh < cd.OH && ow < cd.OW)
-        v = gon[((int64_t)(co0 + co) * cd.OD + od) * OHW +
-                (int64_t)oh * cd.OW + ow];
-      sGo[co][m] = v;
+    for (int r = tid; r < 32 * OHT; r += 256) {
+      const int oh_off = r % OHT;
+      const int co = r / OHT;
+      __bf16* dst = &sGo[co][oh_off * OWT];
+      const int oh = oh0 + oh_off;
+      if ((co0 + co) >= cd.Cout || oh >= cd.OH) {
+#pragma unroll
+        for (int j = 0; j < OWT; ++j) dst[j] = (__bf16)0.f;
+        continue;
+      }
+      const __bf16* src = gon + ((int64_t)(co0 + co) * cd.OD + od) * OHW +
+                          (int64_t)oh * cd.OW + ow0;
+#pragma unroll
+      for (int v = 0; v < OWT / 8; ++v)
+        *reinterpret_cast<bf16x8*>(dst + v * 8) =
+            *reinterpret_cast<const bf16x8*>(src + v * 8);
     }
     __syncthreads();
 
