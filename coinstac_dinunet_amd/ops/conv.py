@@ -19,9 +19,11 @@ class _Conv3dFn(torch.autograd.Function):
         wb = weight.to(torch.bfloat16)
         # spatial tap-reuse pays when the channel tile fills and chunks
         # are dense; else the igemm kernel wins (profiled in tools/bench_conv)
-        if (stride == 1 and xb.size(4) % 8 == 0 and xb.size(1) >= 16
-                and xb.size(3) * xb.size(4) >= 256):
-            out = C.conv3d_fwd_spatial(xb, wb)
+        ow = (xb.size(4) + 2 - 3) // stride + 1
+        oh = (xb.size(3) + 2 - 3) // stride + 1
+        min_chunk = 256 if stride == 1 else 128
+        if (ow % 8 == 0 and xb.size(1) >= 16 and oh * ow >= min_chunk):
+            out = C.conv3d_fwd_spatial(xb, wb, stride)
         else:
             out = C.conv3d_fwd(xb, wb, stride)
         if bias is not None:

@@ -43,7 +43,8 @@ torch::Tensor conv3d_dgrad(torch::Tensor go, torch::Tensor w,
 torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go, int64_t stride);
 torch::Tensor channel_sum(torch::Tensor go);
 // conv3d_spatial.hip
-torch::Tensor conv3d_fwd_spatial(torch::Tensor x, torch::Tensor w);
+torch::Tensor conv3d_fwd_spatial(torch::Tensor x, torch::Tensor w,
+                                 int64_t stride);
 torch::Tensor conv3d_dgrad_spatial(torch::Tensor go, torch::Tensor w,
                                    std::vector<int64_t> in_shape);
 // bnorm.hip

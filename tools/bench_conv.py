@@ -41,11 +41,14 @@ def main():
         OD = (D + 2 - 3) // s + 1
         go = torch.randn(B, Cout, OD, OD, OD, device=dev,
                          dtype=torch.bfloat16)
-        if s == 1 and D % 8 == 0 and Cin >= 16 and D * D >= 256:
-            ms_f = t(lambda: C.conv3d_fwd_spatial(x, w))
-            ms_d = t(lambda: C.conv3d_dgrad_spatial(go, w, list(x.shape)))
+        ow = (D + 2 - 3) // s + 1
+        if ow % 8 == 0 and Cin >= 16 and ow * ow >= (256 if s == 1 else 128):
+            ms_f = t(lambda: C.conv3d_fwd_spatial(x, w, s))
         else:
             ms_f = t(lambda: C.conv3d_fwd(x, w, s))
+        if s == 1 and D % 8 == 0 and Cout >= 16 and D * D >= 256:
+            ms_d = t(lambda: C.conv3d_dgrad_spatial(go, w, list(x.shape)))
+        else:
             ms_d = t(lambda: C.conv3d_dgrad(go, w, list(x.shape), s))
         ms_w = t(lambda: C.conv3d_wgrad(x, go, s))
         xr = x.clone().requires_grad_(True)
