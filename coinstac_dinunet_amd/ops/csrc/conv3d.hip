@@ -576,30 +576,37 @@ static torch::Tensor conv3d_dgrad_s2(torch::Tensor g, torch::Tensor wc,
 }
 
 // ---------------------------------------------------------------------------
-// Stride-1 WGRAD with tap reuse: the split-K implicit-GEMM form re-reads
-// x and go once PER (co,k)-plane (M*K element traffic ~ 27x the tensor).
-// Here a block stages one x spatial slab [ci32][3d][OHT+2][OWT+2] and one
-// go tile [co32][256m] in LDS ONCE, then computes ALL 27 tap GEMMs from
-// it: per 32-m sub-chunk the go fragment loads once and feeds 27 MFMAs
-// whose x fragments read the same slab at shifted offsets. Waves 2x2 over
-// (co32, ci32); acc = 27 f32x4 per wave. Blocks grid-stride over spatial
-// chunks and atomically fold their partials into dw once at the end.
+// WGRAD with tap reuse (stride 1 and 2): the split-K implicit-GEMM form
+// re-reads x and go once PER (co,k)-plane (M*K element traffic ~ 27x the
+// tensor). Here a block stages one x spatial slab [CT][3][H2][W2] (row-
+// vectorized 16B loads) and one go tile [co][m] in LDS ONCE, then computes
+// ALL 27 tap GEMMs from it: per 32-m sub-chunk the go fragment loads once
+// and feeds 27 MFMAs whose x fragments read the same slab at (stride-
+// scaled) tap-shifted offsets. STRIDE=1: waves 2x2 over (co32, ci32);
+// STRIDE=2: waves 4x1 over (co64, ci16) with the strided window staged.
+// Blocks grid-stride over spatial chunks; partials fold into dw by fp32
+// atomics once at the end.
 // ---------------------------------------------------------------------------
-template <int OWT>
+template <int OWT, int STRIDE>
 __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
     const __bf16* __restrict__ x, const __bf16* __restrict__ go,
     float* __restrict__ dw, ConvDims cd, int64_t nchunks, int64_t zstride) {
-  constexpr int OHT = 128 / OWT;  // 128-m chunks: ~36 KB LDS => 4 blocks/CU
-  constexpr int W2 = OWT + 4;   // staged row length (2 halo + pad)
-  constexpr int H2 = OHT + 2;
-  __shared__ __bf16 sX[32][3][H2][W2];
-  __shared__ __bf16 sGo[32][128 + LDA_PAD];
+  constexpr int OHT = 128 / OWT;  // 128-m chunks
+  constexpr int IW = STRIDE * OWT;
+  constexpr int W2 = IW + (STRIDE == 1 ? 4 : 2);
+  constexpr int H2 = STRIDE * (OHT - 1) + 3;
+  constexpr int CT = STRIDE == 1 ? 32 : 16;     // ci tile
+  constexpr int COT = STRIDE == 1 ? 32 : 64;    // co tile
+  __shared__ __bf16 sX[CT][3][H2][W2];
+  __shared__ __bf16 sGo[COT][128 + LDA_PAD];
 
-  const int co0 = blockIdx.x * 32;
-  const int ci0 = blockIdx.y * 32;
+  const int co0 = blockIdx.x * COT;
+  const int ci0 = blockIdx.y * CT;
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
-  const int wi = wave >> 1, wj = wave & 1;   // (co, ci) 16x16 fragment
+  // fragment ownership: (co, ci) 16x16 per wave
+  const int wi = (STRIDE == 1) ? (wave >> 1) : wave;
+  const int wj = (STRIDE == 1) ? (wave & 1) : 0;
   const int row = lane & 15, kg = lane >> 4;
 
   const int wtiles = (cd.OW + OWT - 1) / OWT;
@@ -613,7 +620,6 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
   const int64_t OHW = (int64_t)cd.OH * cd.OW;
 
   for (int64_t z = blockIdx.z; z < nchunks; z += zstride) {
-    // chunk -> (n, od, oh_tile, ow_tile)
     int64_t t = z;
     const int wt = (int)(t % wtiles);
     t /= wtiles;
@@ -624,14 +630,14 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
     const int oh0 = ht * OHT, ow0 = wt * OWT;
 
     // ---- stage x slab row-wise (16B interior vectors, scalar halo) -----
-    constexpr int NXROWS = 32 * 3 * H2;
+    constexpr int NXROWS = CT * 3 * H2;
     const __bf16* xn = x + (int64_t)n * cd.Cin * cd.D * HW;
     for (int r = tid; r < NXROWS; r += 256) {
       const int hrow = r % H2;
       const int a = (r / H2) % 3;
       const int ci = r / (3 * H2);
-      const int id = od - 1 + a;
-      const int ih = oh0 - 1 + hrow;
+      const int id = STRIDE * od - 1 + a;
+      const int ih = STRIDE * oh0 - 1 + hrow;
       __bf16* dst = &sX[ci][a][hrow][0];
       const bool row_ok = (unsigned)id < (unsigned)cd.D &&
                           (unsigned)ih < (unsigned)cd.H &&
@@ -643,22 +649,23 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
       }
       const __bf16* src = xn + ((int64_t)(ci0 + ci) * cd.D + id) * HW +
                           (int64_t)ih * cd.W;
-      dst[0] = (ow0 > 0) ? src[ow0 - 1] : (__bf16)0.f;
+      const int iw0 = STRIDE * ow0;
+      dst[0] = (iw0 > 0) ? src[iw0 - 1] : (__bf16)0.f;
 #pragma unroll
-      for (int v = 0; v < OWT / 8; ++v) {
-        bf16x8 vec = *reinterpret_cast<const bf16x8*>(src + ow0 + v * 8);
+      for (int v = 0; v < IW / 8; ++v) {
+        bf16x8 vec = *reinterpret_cast<const bf16x8*>(src + iw0 + v * 8);
 #pragma unroll
         for (int j = 0; j < 8; ++j) dst[1 + v * 8 + j] = vec[j];
       }
 #pragma unroll
-      for (int e = 0; e < 3; ++e) {
-        const int iw = ow0 + OWT + e;
-        dst[1 + OWT + e] = (iw < cd.W) ? src[iw] : (__bf16)0.f;
+      for (int e = 0; e < W2 - IW - 1; ++e) {
+        const int iw = iw0 + IW + e;
+        dst[1 + IW + e] = (iw < cd.W) ? src[iw] : (__bf16)0.f;
       }
     }
-    // ---- stage go tile row-wise: [co32][m over (OHT x OWT)] ------------
+    // ---- stage go tile row-wise: [co][m over (OHT x OWT)] --------------
     const __bf16* gon = go + (int64_t)n * cd.Cout * cd.OD * OHW;
-    for (int r = tid; r < 32 * OHT; r += 256) {
+    for (int r = tid; r < COT * OHT; r += 256) {
       const int oh_off = r % OHT;
       const int co = r / OHT;
       __bf16* dst = &sGo[co][oh_off * OWT];
@@ -677,15 +684,13 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
     }
     __syncthreads();
 
-    // ---- 8 m-subchunks x 27 taps ---------------------------------------
+    // ---- 4 m-subchunks x 27 taps ---------------------------------------
 #pragma unroll 1
     for (int ms = 0; ms < 4; ++ms) {
       bf16x8 afrag;   // go[co16][m32]
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         afrag[j] = sGo[wi * 16 + row][ms * 32 + kg * 8 + j];
-      // element m = ms*32 + kg*8 + j -> oh_off = m/OWT, ow = m%OWT; the
-      // 8-run stays inside one w-row for OWT in {8,16,32}
       const int mbase = ms * 32 + kg * 8;
       const int oh_off = mbase / OWT;
       const int ow_off = mbase % OWT;
@@ -696,10 +701,11 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
 #pragma unroll
           for (int kw = 0; kw < 3; ++kw) {
             bf16x8 bfrag;
-            const __bf16* src =
-                &sX[wj * 16 + row][kd][oh_off + kh][ow_off + kw];
+            const __bf16* src = &sX[wj * 16 + row][kd]
+                                   [STRIDE * oh_off + kh]
+                                   [STRIDE * ow_off + kw];
 #pragma unroll
-            for (int j = 0; j < 8; ++j) bfrag[j] = src[j];
+            for (int j = 0; j < 8; ++j) bfrag[j] = src[STRIDE * j];
             acc[(kd * 3 + kh) * 3 + kw] =
                 __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                     afrag, bfrag, acc[(kd * 3 + kh) * 3 + kw], 0, 0, 0);
@@ -831,36 +837,34 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
   auto dw = torch::zeros({cd.Cout, (int64_t)K},
                          xc.options().dtype(torch::kFloat32));
 
-  if (stride == 1 && (cd.OW % 8) == 0 && cd.Cin >= 16 &&
-      cd.OH * cd.OW >= 256) {
-    // tap-reuse path
+  if ((cd.OW % 8) == 0 && cd.Cin >= 16 && cd.OH * cd.OW >= 128 &&
+      (stride == 1 || stride == 2)) {
+    // tap-reuse path (stride-templated)
     int OWT = cd.OW % 32 == 0 ? 32 : (cd.OW % 16 == 0 ? 16 : 8);
     int wtiles = (cd.OW + OWT - 1) / OWT;
     int OHT = 128 / OWT;
     int htiles = (cd.OH + OHT - 1) / OHT;
-    int co_t = (cd.Cout + 31) / 32, ci_t = (cd.Cin + 31) / 32;
+    int COT = stride == 1 ? 32 : 64, CT = stride == 1 ? 32 : 16;
+    int co_t = (cd.Cout + COT - 1) / COT, ci_t = (cd.Cin + CT - 1) / CT;
     int64_t nchunks = (int64_t)cd.N * cd.OD * htiles * wtiles;
     int64_t zstride = std::max<int64_t>(
         1, std::min<int64_t>(nchunks, 768 / std::max(co_t * ci_t, 1)));
     dim3 grid(co_t, ci_t, (unsigned)zstride);
-    if (OWT == 32)
-      hipLaunchKernelGGL(conv3d_wgrad_s1_kernel<32>, grid, dim3(256), 0,
-                         current_stream(),
+    auto L = [&](auto kern) {
+      hipLaunchKernelGGL(kern, grid, dim3(256), 0, current_stream(),
                          reinterpret_cast<const __bf16*>(xc.data_ptr()),
                          reinterpret_cast<const __bf16*>(g.data_ptr()),
                          dw.data_ptr<float>(), cd, nchunks, zstride);
-    else if (OWT == 16)
-      hipLaunchKernelGGL(conv3d_wgrad_s1_kernel<16>, grid, dim3(256), 0,
-                         current_stream(),
-                         reinterpret_cast<const __bf16*>(xc.data_ptr()),
-                         reinterpret_cast<const __bf16*>(g.data_ptr()),
-                         dw.data_ptr<float>(), cd, nchunks, zstride);
-    else
-      hipLaunchKernelGGL(conv3d_wgrad_s1_kernel<8>, grid, dim3(256), 0,
-                         current_stream(),
-                         reinterpret_cast<const __bf16*>(xc.data_ptr()),
-                         reinterpret_cast<const __bf16*>(g.data_ptr()),
-                         dw.data_ptr<float>(), cd, nchunks, zstride);
+    };
+    if (stride == 1) {
+      if (OWT == 32) L(conv3d_wgrad_s1_kernel<32, 1>);
+      else if (OWT == 16) L(conv3d_wgrad_s1_kernel<16, 1>);
+      else L(conv3d_wgrad_s1_kernel<8, 1>);
+    } else {
+      if (OWT == 32) L(conv3d_wgrad_s1_kernel<32, 2>);
+      else if (OWT == 16) L(conv3d_wgrad_s1_kernel<16, 2>);
+      else L(conv3d_wgrad_s1_kernel<8, 2>);
+    }
     return dw.view({cd.Cout, cd.Cin, 3, 3, 3});
   }
 
