@@ -22,7 +22,7 @@ def parse_args():
     ap.add_argument('--gpus', type=int, default=1)
     ap.add_argument('--steps', type=int, default=20)
     ap.add_argument('--warmup', type=int, default=5)
-    ap.add_argument('--batch', type=int, default=16,
+    ap.add_argument('--batch', type=int, default=64,
                     help='per-GPU (per-site) batch size')
     ap.add_argument('--vol', type=int, default=64, help='volume side length')
     ap.add_argument('--local-iterations', type=int, default=1)

@@ -43,11 +43,17 @@ class _Conv3dFn(torch.autograd.Function):
         go = grad_out.to(torch.bfloat16).contiguous()
         gx = gw = gb = None
         if ctx.x_requires:
+            wsub = (xb.size(4) + 1) // 2
+            hsub = (xb.size(3) + 1) // 2
             if (ctx.stride == 1 and xb.size(4) % 8 == 0
                     and go.size(1) >= 16
                     and xb.size(3) * xb.size(4) >= 256):
                 gx = C.conv3d_dgrad_spatial(go, wb,
                                             list(xb.shape)).to(ctx.in_dtype)
+            elif (ctx.stride == 2 and wsub % 8 == 0 and go.size(1) >= 32
+                    and hsub * wsub >= 128):
+                gx = C.conv3d_dgrad_s2_spatial(
+                    go, wb, list(xb.shape)).to(ctx.in_dtype)
             else:
                 gx = C.conv3d_dgrad(go, wb, list(xb.shape),
                                     ctx.stride).to(ctx.in_dtype)

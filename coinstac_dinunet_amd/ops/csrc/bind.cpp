@@ -47,6 +47,8 @@ torch::Tensor conv3d_fwd_spatial(torch::Tensor x, torch::Tensor w,
                                  int64_t stride);
 torch::Tensor conv3d_dgrad_spatial(torch::Tensor go, torch::Tensor w,
                                    std::vector<int64_t> in_shape);
+torch::Tensor conv3d_dgrad_s2_spatial(torch::Tensor go, torch::Tensor w,
+                                      std::vector<int64_t> in_shape);
 // bnorm.hip
 std::vector<torch::Tensor> bn3d_fwd(torch::Tensor x, torch::Tensor gamma,
                                     torch::Tensor beta, double eps, bool relu);
@@ -78,6 +80,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("channel_sum", &channel_sum);
   m.def("conv3d_fwd_spatial", &conv3d_fwd_spatial);
   m.def("conv3d_dgrad_spatial", &conv3d_dgrad_spatial);
+  m.def("conv3d_dgrad_s2_spatial", &conv3d_dgrad_s2_spatial);
   m.def("bn3d_fwd", &bn3d_fwd);
   m.def("bn3d_infer", &bn3d_infer);
   m.def("bn3d_bwd", &bn3d_bwd);

@@ -271,3 +271,247 @@ torch::Tensor conv3d_dgrad_spatial(torch::Tensor go, torch::Tensor w,
   launch_spatial(g, wb, dx, sd, 1);
   return dx;
 }
+
+// ---------------------------------------------------------------------------
+// Stride-2 DGRAD, parity classes through the spatial-slab structure.
+// dx decomposes by (id,ih,iw) mod 2 into 8 dense sub-problems; within a
+// class the taps shift the go window by only {0,+1} per axis, so a block
+// stages a TINY go slab [co32][2][OHT+1][OWT+1] once per 32-co tile and
+// computes K = 32*T tap-products from it (T = 1..8 taps per class). The
+// per-class weight matrix WBc[ci][co*T + r] is prepared host-side
+// (tap-gathered), contiguous 16B per lane. blockIdx.z = class.
+// ---------------------------------------------------------------------------
+template <int OWT>
+__global__ __launch_bounds__(256) void conv3d_dgrad_s2_sp_kernel(
+    const __bf16* __restrict__ go, const __bf16* __restrict__ wb,
+    __bf16* __restrict__ dx, SpDims sd, int cls_off0, int cls_off1,
+    int cls_off2, int cls_off3, int cls_off4, int cls_off5, int cls_off6,
+    int cls_off7) {
+  constexpr int OHT = 128 / OWT;
+  constexpr int W2 = OWT + 4;        // OWT+1 used; padded
+  constexpr int H2 = OHT + 1;
+  constexpr int MPW = (OWT * OHT) / 64;  // = 2
+  __shared__ __bf16 sG[32][2][H2][W2];
+  __shared__ unsigned short sKtab[32 * 8 + 8];
+
+  const int cls = blockIdx.z;
+  const int a = (cls >> 2) & 1, b = (cls >> 1) & 1, c = cls & 1;
+  const int l2w = c, l2h = b;
+  const int T = 1 << (a + b + c);
+  const int cls_offs[8] = {cls_off0, cls_off1, cls_off2, cls_off3,
+                           cls_off4, cls_off5, cls_off6, cls_off7};
+  const int wb_cls = cls_offs[cls];
+
+  // dx sub-grid dims for this class
+  const int Da = (sd.TD - a + 1) >> 1;
+  const int Hb = (sd.TH - b + 1) >> 1;
+  const int Wc = (sd.TW - c + 1) >> 1;
+
+  const int ncol0 = blockIdx.y * 32;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int row = lane & 15, kg = lane >> 4;
+
+  const int wtiles = (Wc + OWT - 1) / OWT;
+  const int htiles = (Hb + OHT - 1) / OHT;
+  const int64_t nchunks = (int64_t)sd.N * Da * htiles * wtiles;
+  if (blockIdx.x >= nchunks) return;
+
+  int64_t t = blockIdx.x;
+  const int wt = (int)(t % wtiles);
+  t /= wtiles;
+  const int ht = (int)(t % htiles);
+  t /= htiles;
+  const int tdp = (int)(t % Da);     // id' (sub-grid d)
+  const int n = (int)(t / Da);
+  const int oh0 = ht * OHT, ow0 = wt * OWT;
+
+  // k -> slab offset: k = co_l*T + r; r -> (dod, doh, dow) in {0,1}
+  for (int k = tid; k < 32 * T; k += 256) {
+    const int co_l = k >> (a + b + c);
+    const int r = k & (T - 1);
+    const int tw_i = r & ((1 << l2w) - 1);
+    const int th_i = (r >> l2w) & ((1 << l2h) - 1);
+    const int td_i = r >> (l2w + l2h);
+    const int dod = a ? (1 - td_i) : 0;
+    const int doh = b ? (1 - th_i) : 0;
+    const int dow = c ? (1 - tw_i) : 0;
+    sKtab[k] = (unsigned short)(((co_l * 2 + dod) * H2 + doh) * W2 + dow);
+  }
+
+  f32x4 acc[MPW][2];
+#pragma unroll
+  for (int i = 0; i < MPW; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int64_t OHW = (int64_t)sd.H * sd.W;  // go plane (input-side dims)
+  const int64_t go_n = (int64_t)n * sd.KCH * sd.D * OHW;
+  const int kts = (sd.KCH + 31) / 32;        // co tiles
+  const int KT = 32 * T;                     // k per co tile (mult of 32)
+
+  for (int kt = 0; kt < kts; ++kt) {
+    constexpr int NRMAX = 32 * 2 * H2;
+    if (kt) __syncthreads();
+    else __syncthreads();  // ktab ready
+    for (int r = tid; r < NRMAX; r += 256) {
+      const int hrow = r % H2;
+      const int p = (r / H2) & 1;
+      const int co = r / (2 * H2);
+      const int od = tdp + p;                // dod in {0,1}
+      const int oh = oh0 + hrow;
+      const int ch = kt * 32 + co;
+      __bf16* dst = &sG[co][p][hrow][0];
+      const bool row_ok = od < sd.D && oh < sd.H && ch < sd.KCH &&
+                          (a || p == 0);     // a==0 uses only plane 0
+      if (!row_ok) {
+#pragma unroll
+        for (int col = 0; col < W2; ++col) dst[col] = (__bf16)0.f;
+        continue;
+      }
+      const __bf16* src = go + go_n + ((int64_t)ch * sd.D + od) * OHW +
+                          (int64_t)oh * sd.W + ow0;
+#pragma unroll
+      for (int v = 0; v < OWT / 8; ++v) {
+        if (ow0 + v * 8 + 7 < sd.W) {
+          bf16x8 vec = *reinterpret_cast<const bf16x8*>(src + v * 8);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) dst[v * 8 + j] = vec[j];
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int ow = ow0 + v * 8 + j;
+            dst[v * 8 + j] = (ow < sd.W) ? src[v * 8 + j] : (__bf16)0.f;
+          }
+        }
+      }
+      {  // +1 tail column
+        const int ow = ow0 + OWT;
+        dst[OWT] = (ow < sd.W) ? src[OWT] : (__bf16)0.f;
+      }
+    }
+    __syncthreads();
+
+#pragma unroll 1
+    for (int ks = 0; ks < KT / 32; ++ks) {
+      bf16x8 afrag[MPW];
+      {
+        const int kb = ks * 32 + kg * 8;
+        const u16x8 kt8 = *reinterpret_cast<const u16x8*>(&sKtab[kb]);
+        const __bf16* slab = &sG[0][0][0][0];
+#pragma unroll
+        for (int i = 0; i < MPW; ++i) {
+          const int m = (wave * MPW + i) * 16 + row;
+          const int base = (m / OWT) * W2 + (m % OWT);
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            afrag[i][j] = slab[base + kt8[j]];
+        }
+      }
+      bf16x8 bfrag[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        const int col = ncol0 + i * 16 + row;
+        const int64_t off = (int64_t)col * sd.Kpad + wb_cls + kt * KT +
+                            ks * 32 + kg * 8;
+        bfrag[i] = (col < sd.NCOL)
+                       ? *reinterpret_cast<const bf16x8*>(wb + off)
+                       : bf16x8{};
+      }
+#pragma unroll
+      for (int i = 0; i < MPW; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    }
+  }
+
+  // epilogue: dx[n][ci][2*id'+a][2*ih'+b][2*iw'+c]
+  const int64_t THW = (int64_t)sd.TH * sd.TW;
+  const int64_t dx_n = (int64_t)n * sd.NCOL * sd.TD * THW;
+  const int ccol = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < MPW; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int col = ncol0 + j * 16 + ccol;
+      if (col >= sd.NCOL) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = (wave * MPW + i) * 16 + crow0 + r;
+        const int ihp = oh0 + m / OWT;
+        const int iwp = ow0 + m % OWT;
+        if (ihp < Hb && iwp < Wc)
+          dx[dx_n + ((int64_t)col * sd.TD + (2 * tdp + a)) * THW +
+             (int64_t)(2 * ihp + b) * sd.TW + (2 * iwp + c)] =
+              (__bf16)(acc[i][j][r]);
+      }
+    }
+  }
+}
+
+torch::Tensor conv3d_dgrad_s2_spatial(torch::Tensor go, torch::Tensor w,
+                                      std::vector<int64_t> in_shape) {
+  CHECK_GPU(go);
+  auto g = go.to(torch::kBFloat16).contiguous();
+  auto wc = w.to(torch::kBFloat16).contiguous();
+  int Cout = (int)wc.size(0), Cin = (int)wc.size(1);
+  SpDims sd;
+  sd.N = (int)g.size(0); sd.KCH = Cout;
+  sd.D = (int)g.size(2); sd.H = (int)g.size(3); sd.W = (int)g.size(4);
+  sd.NCOL = Cin;
+  sd.TD = (int)in_shape[2]; sd.TH = (int)in_shape[3];
+  sd.TW = (int)in_shape[4];
+
+  // per-class tap-gathered weights: WBc[ci][co*T + r]
+  auto w3 = wc.reshape({Cout, Cin, 27}).permute({1, 0, 2}).contiguous();
+  std::vector<torch::Tensor> blocks;
+  std::vector<int64_t> offs(8, 0);
+  int64_t cur = 0;
+  for (int cls = 0; cls < 8; ++cls) {
+    int a = (cls >> 2) & 1, b = (cls >> 1) & 1, c = cls & 1;
+    std::vector<int64_t> taps;
+    for (int td_i = 0; td_i < (a ? 2 : 1); ++td_i)
+      for (int th_i = 0; th_i < (b ? 2 : 1); ++th_i)
+        for (int tw_i = 0; tw_i < (c ? 2 : 1); ++tw_i) {
+          int kd = a ? td_i * 2 : 1;
+          int kh = b ? th_i * 2 : 1;
+          int kw = c ? tw_i * 2 : 1;
+          taps.push_back(kd * 9 + kh * 3 + kw);
+        }
+    auto idx = torch::tensor(taps, torch::TensorOptions()
+                                       .dtype(torch::kLong)
+                                       .device(wc.device()));
+    auto blk = w3.index_select(2, idx).reshape({Cin, -1});  // [Cin][Co*T]
+    offs[cls] = cur;
+    cur += blk.size(1);
+    blocks.push_back(blk);
+  }
+  auto wb = torch::cat(blocks, 1).contiguous();  // [Cin][Cout*27]
+  sd.Kpad = (int)wb.size(1);
+
+  auto dx = torch::empty(in_shape, g.options());
+  int Wc = (sd.TW + 1) >> 1, Hb = (sd.TH + 1) >> 1, Da = (sd.TD + 1) >> 1;
+  int OWT = Wc % 32 == 0 ? 32 : (Wc % 16 == 0 ? 16 : 8);
+  int OHT = 128 / OWT;
+  int wtiles = (Wc + OWT - 1) / OWT;
+  int htiles = (Hb + OHT - 1) / OHT;
+  int64_t nchunks = (int64_t)sd.N * Da * htiles * wtiles;
+  dim3 grid((unsigned)nchunks, (Cin + 31) / 32, 8);
+  auto s = current_stream();
+  auto L = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, s,
+                       reinterpret_cast<const __bf16*>(g.data_ptr()),
+                       reinterpret_cast<const __bf16*>(wb.data_ptr()),
+                       reinterpret_cast<__bf16*>(dx.data_ptr()), sd,
+                       (int)offs[0], (int)offs[1], (int)offs[2],
+                       (int)offs[3], (int)offs[4], (int)offs[5],
+                       (int)offs[6], (int)offs[7]);
+  };
+  if (OWT == 32) L(conv3d_dgrad_s2_sp_kernel<32>);
+  else if (OWT == 16) L(conv3d_dgrad_s2_sp_kernel<16>);
+  else L(conv3d_dgrad_s2_sp_kernel<8>);
+  return dx;
+}

@@ -46,8 +46,11 @@ def main():
             ms_f = t(lambda: C.conv3d_fwd_spatial(x, w, s))
         else:
             ms_f = t(lambda: C.conv3d_fwd(x, w, s))
+        wsub = (D + 1) // 2
         if s == 1 and D % 8 == 0 and Cout >= 16 and D * D >= 256:
             ms_d = t(lambda: C.conv3d_dgrad_spatial(go, w, list(x.shape)))
+        elif s == 2 and wsub % 8 == 0 and Cout >= 32 and wsub * wsub >= 128:
+            ms_d = t(lambda: C.conv3d_dgrad_s2_spatial(go, w, list(x.shape)))
         else:
             ms_d = t(lambda: C.conv3d_dgrad(go, w, list(x.shape), s))
         ms_w = t(lambda: C.conv3d_wgrad(x, go, s))
