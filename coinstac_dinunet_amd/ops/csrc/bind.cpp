@@ -35,6 +35,7 @@ torch::Tensor confusion_matrix(torch::Tensor pred, torch::Tensor true_,
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor weight,
                          torch::Tensor bias, bool relu);
 torch::Tensor colsum(torch::Tensor g);
+void gram_schmidt(torch::Tensor m, double eps);
 // conv3d.hip
 torch::Tensor mfma_probe_gemm(torch::Tensor A, torch::Tensor B);
 torch::Tensor conv3d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride);
@@ -73,6 +74,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("confusion_matrix", &confusion_matrix);
   m.def("linear_fwd", &linear_fwd);
   m.def("colsum", &colsum);
+  m.def("gram_schmidt", &gram_schmidt);
   m.def("mfma_probe_gemm", &mfma_probe_gemm);
   m.def("conv3d_fwd", &conv3d_fwd);
   m.def("conv3d_dgrad", &conv3d_dgrad);

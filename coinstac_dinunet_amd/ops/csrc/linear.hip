@@ -182,3 +182,66 @@ torch::Tensor colsum(torch::Tensor g) {
                      out.data_ptr<float>(), M, N);
   return out;
 }
+
+// K8 — fused Gram-Schmidt column orthogonalization (PowerSGD).
+// One workgroup per matrix [n, r] (r small, typically 1-4): each column
+// pass computes the norm (wave-reduce over rows), scales, and projects the
+// remaining columns — a single launch instead of 3*r elementwise kernels.
+__global__ void gram_schmidt_kernel(float* __restrict__ m, int n, int r,
+                                    float eps) {
+  extern __shared__ __attribute__((aligned(16))) char smem_gs[];
+  float* red = reinterpret_cast<float*>(smem_gs);  // [blockDim/64]
+  const int tid = threadIdx.x;
+  const int nw = blockDim.x / WAVE_SIZE;
+  for (int i = 0; i < r; ++i) {
+    // norm of column i
+    float s = 0.f;
+    for (int row = tid; row < n; row += blockDim.x) {
+      float v = m[(int64_t)row * r + i];
+      s += v * v;
+    }
+    for (int off = WAVE_SIZE / 2; off > 0; off >>= 1)
+      s += __shfl_down(s, off);
+    if ((tid & 63) == 0) red[tid >> 6] = s;
+    __syncthreads();
+    if (tid == 0) {
+      float t = 0.f;
+      for (int w = 0; w < nw; ++w) t += red[w];
+      red[0] = sqrtf(t) + eps;
+    }
+    __syncthreads();
+    const float inv = 1.0f / red[0];
+    for (int row = tid; row < n; row += blockDim.x)
+      m[(int64_t)row * r + i] *= inv;
+    __syncthreads();
+    // project out of the remaining columns
+    for (int j = i + 1; j < r; ++j) {
+      float d = 0.f;
+      for (int row = tid; row < n; row += blockDim.x)
+        d += m[(int64_t)row * r + i] * m[(int64_t)row * r + j];
+      for (int off = WAVE_SIZE / 2; off > 0; off >>= 1)
+        d += __shfl_down(d, off);
+      if ((tid & 63) == 0) red[tid >> 6] = d;
+      __syncthreads();
+      if (tid == 0) {
+        float t = 0.f;
+        for (int w = 0; w < nw; ++w) t += red[w];
+        red[0] = t;
+      }
+      __syncthreads();
+      const float dot = red[0];
+      for (int row = tid; row < n; row += blockDim.x)
+        m[(int64_t)row * r + j] -= dot * m[(int64_t)row * r + i];
+      __syncthreads();
+    }
+  }
+}
+
+void gram_schmidt(torch::Tensor m, double eps) {
+  CHECK_GPU(m); CHECK_CONTIG(m);
+  TORCH_CHECK(m.dim() == 2 && m.scalar_type() == torch::kFloat32);
+  int n = (int)m.size(0), r = (int)m.size(1);
+  hipLaunchKernelGGL(gram_schmidt_kernel, dim3(1), dim3(256),
+                     16 * sizeof(float), current_stream(),
+                     m.data_ptr<float>(), n, r, (float)eps);
+}

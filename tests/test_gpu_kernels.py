@@ -180,3 +180,22 @@ def test_linear_bf16(dev):
 def test_graft_smoke(dev):
     import __graft_entry__
     __graft_entry__.smoke()
+
+
+def test_gram_schmidt_kernel(dev):
+    torch.manual_seed(11)
+    m = torch.randn(500, 4, device=dev)
+    ref = m.clone()
+    from coinstac_dinunet_amd.distrib.powersgd import orthogonalize
+    C.gram_schmidt(m, 1e-8)
+    # CPU reference Gram-Schmidt
+    n_cols = ref.shape[1]
+    for i in range(n_cols):
+        col = ref[:, i:i + 1]
+        col.div_(torch.norm(col) + 1e-8)
+        if i + 1 < n_cols:
+            ref[:, i + 1:].sub_(col @ (col.t() @ ref[:, i + 1:]))
+    torch.testing.assert_close(m, ref, rtol=1e-4, atol=1e-5)
+    gram = m.t() @ m
+    torch.testing.assert_close(gram, torch.eye(4, device=dev),
+                               rtol=1e-3, atol=1e-3)
