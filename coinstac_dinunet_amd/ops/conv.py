@@ -21,7 +21,7 @@ class _Conv3dFn(torch.autograd.Function):
         # are dense; else the igemm kernel wins (profiled in tools/bench_conv)
         ow = (xb.size(4) + 2 - 3) // stride + 1
         oh = (xb.size(3) + 2 - 3) // stride + 1
-        min_chunk = 256 if stride == 1 else 128
+        min_chunk = 64
         if (ow % 8 == 0 and xb.size(1) >= 16 and oh * ow >= min_chunk):
             out = C.conv3d_fwd_spatial(xb, wb, stride)
         else:
@@ -47,7 +47,7 @@ class _Conv3dFn(torch.autograd.Function):
             hsub = (xb.size(3) + 1) // 2
             if (ctx.stride == 1 and xb.size(4) % 8 == 0
                     and go.size(1) >= 16
-                    and xb.size(3) * xb.size(4) >= 256):
+                    and xb.size(3) * xb.size(4) >= 64):
                 gx = C.conv3d_dgrad_spatial(go, wb,
                                             list(xb.shape)).to(ctx.in_dtype)
             elif (ctx.stride == 2 and wsub % 8 == 0 and go.size(1) >= 32

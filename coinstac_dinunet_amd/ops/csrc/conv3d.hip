@@ -587,18 +587,18 @@ static torch::Tensor conv3d_dgrad_s2(torch::Tensor g, torch::Tensor wc,
 // Blocks grid-stride over spatial chunks; partials fold into dw by fp32
 // atomics once at the end.
 // ---------------------------------------------------------------------------
-template <int OWT, int STRIDE>
+template <int OWT, int STRIDE, int CHUNK = 128>
 __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
     const __bf16* __restrict__ x, const __bf16* __restrict__ go,
     float* __restrict__ dw, ConvDims cd, int64_t nchunks, int64_t zstride) {
-  constexpr int OHT = 128 / OWT;  // 128-m chunks
+  constexpr int OHT = CHUNK / OWT;
   constexpr int IW = STRIDE * OWT;
   constexpr int W2 = IW + (STRIDE == 1 ? 4 : 2);
   constexpr int H2 = STRIDE * (OHT - 1) + 3;
   constexpr int CT = STRIDE == 1 ? 32 : 16;     // ci tile
   constexpr int COT = STRIDE == 1 ? 32 : 64;    // co tile
   __shared__ __bf16 sX[CT][3][H2][W2];
-  __shared__ __bf16 sGo[COT][128 + LDA_PAD];
+  __shared__ __bf16 sGo[COT][CHUNK + LDA_PAD];
 
   const int co0 = blockIdx.x * COT;
   const int ci0 = blockIdx.y * CT;
@@ -684,9 +684,9 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
     }
     __syncthreads();
 
-    // ---- 4 m-subchunks x 27 taps ---------------------------------------
+    // ---- m-subchunks x 27 taps -----------------------------------------
 #pragma unroll 1
-    for (int ms = 0; ms < 4; ++ms) {
+    for (int ms = 0; ms < CHUNK / 32; ++ms) {
       bf16x8 afrag;   // go[co16][m32]
 #pragma unroll
       for (int j = 0; j < 8; ++j)
@@ -837,12 +837,17 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
   auto dw = torch::zeros({cd.Cout, (int64_t)K},
                          xc.options().dtype(torch::kFloat32));
 
-  if ((cd.OW % 8) == 0 && cd.Cin >= 16 && cd.OH * cd.OW >= 128 &&
+  if ((cd.OW % 8) == 0 && cd.Cin >= 16 && cd.OH * cd.OW >= 64 &&
       (stride == 1 || stride == 2)) {
     // tap-reuse path (stride-templated)
     int OWT = cd.OW % 32 == 0 ? 32 : (cd.OW % 16 == 0 ? 16 : 8);
+    int chunk = 128;
+    if (cd.OH * cd.OW < 128) {
+      chunk = 64;
+      OWT = 8;
+    }
     int wtiles = (cd.OW + OWT - 1) / OWT;
-    int OHT = 128 / OWT;
+    int OHT = chunk / OWT;
     int htiles = (cd.OH + OHT - 1) / OHT;
     int COT = stride == 1 ? 32 : 64, CT = stride == 1 ? 32 : 16;
     int co_t = (cd.Cout + COT - 1) / COT, ci_t = (cd.Cin + CT - 1) / CT;
@@ -857,11 +862,13 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
                          dw.data_ptr<float>(), cd, nchunks, zstride);
     };
     if (stride == 1) {
-      if (OWT == 32) L(conv3d_wgrad_s1_kernel<32, 1>);
+      if (chunk == 64) L(conv3d_wgrad_s1_kernel<8, 1, 64>);
+      else if (OWT == 32) L(conv3d_wgrad_s1_kernel<32, 1>);
       else if (OWT == 16) L(conv3d_wgrad_s1_kernel<16, 1>);
       else L(conv3d_wgrad_s1_kernel<8, 1>);
     } else {
-      if (OWT == 32) L(conv3d_wgrad_s1_kernel<32, 2>);
+      if (chunk == 64) L(conv3d_wgrad_s1_kernel<8, 2, 64>);
+      else if (OWT == 32) L(conv3d_wgrad_s1_kernel<32, 2>);
       else if (OWT == 16) L(conv3d_wgrad_s1_kernel<16, 2>);
       else L(conv3d_wgrad_s1_kernel<8, 2>);
     }

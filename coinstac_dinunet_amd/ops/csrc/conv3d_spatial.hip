@@ -33,16 +33,17 @@ struct SpDims {
 #define LDA_PAD 8
 #endif
 
-template <int OWT, int STRIDE, int CTILE>
+template <int OWT, int STRIDE, int CTILE,
+          int CHUNK = (STRIDE == 1 ? 256 : 128)>
 __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
     const __bf16* __restrict__ in, const __bf16* __restrict__ wb,
     __bf16* __restrict__ out, SpDims sd, int64_t nchunks) {
-  constexpr int OHT = (STRIDE == 1 ? 256 : 128) / OWT;
+  constexpr int OHT = CHUNK / OWT;
   constexpr int IW = STRIDE * OWT;                  // staged interior width
   constexpr int W2 = IW + (STRIDE == 1 ? 4 : 2);
   constexpr int H2 = STRIDE * (OHT - 1) + 3;
-  constexpr int CHUNK = OWT * OHT;                  // output positions
   constexpr int MPW = CHUNK / 64;                   // m-frags per wave
+  static_assert(MPW >= 1, "chunk too small");
   constexpr int KT_PAD = ((CTILE * 27 + 31) / 32) * 32;
   __shared__ __bf16 sX[CTILE][3][H2][W2];
   __shared__ unsigned short sKtab[KT_PAD + 8];
@@ -204,7 +205,13 @@ static torch::Tensor prep_wb(torch::Tensor w_flat2d, int KCH, int ctile) {
 static void launch_spatial(torch::Tensor in, torch::Tensor wb,
                            torch::Tensor out, SpDims sd, int stride) {
   int OWT = sd.TW % 32 == 0 ? 32 : (sd.TW % 16 == 0 ? 16 : 8);
-  int OHT = (stride == 1 ? 256 : 128) / OWT;
+  int chunk = stride == 1 ? 256 : 128;
+  // small images (d8-class): 64-position chunks keep the grid dense
+  if (sd.TH * sd.TW < chunk) {
+    chunk = 64;
+    OWT = 8;  // the chunk-64 instances are OWT=8
+  }
+  int OHT = chunk / OWT;
   int wtiles = (sd.TW + OWT - 1) / OWT;
   int htiles = (sd.TH + OHT - 1) / OHT;
   int64_t nchunks = (int64_t)sd.N * sd.TD * htiles * wtiles;
@@ -217,11 +224,13 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
     hipLaunchKernelGGL(kern, grid, dim3(256), 0, s, ip, wp, op, sd, nchunks);
   };
   if (stride == 1) {
-    if (OWT == 32) L(conv3d_spatial_kernel<32, 1, 32>);
+    if (chunk == 64) L(conv3d_spatial_kernel<8, 1, 32, 64>);
+    else if (OWT == 32) L(conv3d_spatial_kernel<32, 1, 32>);
     else if (OWT == 16) L(conv3d_spatial_kernel<16, 1, 32>);
     else L(conv3d_spatial_kernel<8, 1, 32>);
   } else {
-    if (OWT == 32) L(conv3d_spatial_kernel<32, 2, 16>);
+    if (chunk == 64) L(conv3d_spatial_kernel<8, 2, 16, 64>);
+    else if (OWT == 32) L(conv3d_spatial_kernel<32, 2, 16>);
     else if (OWT == 16) L(conv3d_spatial_kernel<16, 2, 16>);
     else L(conv3d_spatial_kernel<8, 2, 16>);
   }
