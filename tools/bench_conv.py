@@ -42,12 +42,12 @@ def main():
         go = torch.randn(B, Cout, OD, OD, OD, device=dev,
                          dtype=torch.bfloat16)
         ow = (D + 2 - 3) // s + 1
-        if ow % 8 == 0 and Cin >= 16 and ow * ow >= (256 if s == 1 else 128):
+        if ow % 8 == 0 and Cin >= 16 and ow * ow >= 64:
             ms_f = t(lambda: C.conv3d_fwd_spatial(x, w, s))
         else:
             ms_f = t(lambda: C.conv3d_fwd(x, w, s))
         wsub = (D + 1) // 2
-        if s == 1 and D % 8 == 0 and Cout >= 16 and D * D >= 256:
+        if s == 1 and D % 8 == 0 and Cout >= 16 and D * D >= 64:
             ms_d = t(lambda: C.conv3d_dgrad_spatial(go, w, list(x.shape)))
         elif s == 2 and wsub % 8 == 0 and Cout >= 32 and wsub * wsub >= 128:
             ms_d = t(lambda: C.conv3d_dgrad_s2_spatial(go, w, list(x.shape)))
