@@ -20,7 +20,6 @@ import os as _os
 import numpy as _np
 import torch as _torch
 
-from .. import config as _conf
 from ..utils import tensorutils as _tu
 from .learner import COINNLearner
 from .reducer import COINNReducer

@@ -1,5 +1,4 @@
 """Segmentation losses (parity: reference metrics/loss.py:1-23)."""
-import torch
 
 
 def dice_loss_binary(outputs, target, beta=1.0, weights=None, eps=1e-5):

@@ -6,7 +6,6 @@ dinunet_implementations repos, README.md:30-33); the FreeSurfer workload is
 linear layers route through the MFMA HIP GEMM (ops.linear) when the
 extension is loaded.
 """
-import torch
 import torch.nn as nn
 
 

@@ -5,11 +5,10 @@ The reference's GPU example workload is a 3D CNN over ~64^3 (up to
 This is our own architecture sized for that shape: conv3x3x3 stacks with
 BN/ReLU and stride-2 downsampling, global average pool, linear head.
 
-MI355X notes: channels_last_3d (NDHWC) memory format is used on GPU so the
-conv kernels (ops Conv3d implicit-GEMM / MIOpen) see coalescible layouts;
-bf16 autocast-friendly (no ops that silently upcast).
+MI355X notes: NCDHW layout (w-contiguous rows feed the spatial-slab conv
+kernels' 16-byte staging loads); conv runs on OpsConv3d (implicit-GEMM
+MFMA) and BN+ReLU is one fused pass (OpsBatchNorm3d).
 """
-import torch
 import torch.nn as nn
 
 from ..ops.bnorm import OpsBatchNorm3d

@@ -15,7 +15,6 @@ The extension is built in-tree (`python setup.py build_ext --inplace` or
 __graft_entry__.build()) for gfx950 only. On a GPU box the native path is
 MANDATORY: ops fail loudly rather than silently falling back to eager.
 """
-import os
 
 import torch
 

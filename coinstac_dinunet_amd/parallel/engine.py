@@ -26,7 +26,7 @@ import os
 import torch
 import torch.distributed as dist
 
-from ..config.keys import Mode
+
 from ..distrib.learner import COINNLearner
 from ..distrib.reducer import COINNReducer
 

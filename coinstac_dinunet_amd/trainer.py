@@ -11,7 +11,7 @@ from os import sep as _sep
 
 from . import config as _conf
 from . import metrics as _metrics
-from .config.keys import Key, Mode
+from .config.keys import Key
 from .nn.basetrainer import NNTrainer as _NNTrainer
 from .utils.utils import performance_improved_
 

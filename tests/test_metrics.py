@@ -77,3 +77,12 @@ def test_auc_reduce_sites_mean():
     m = AUCROCMetrics()
     m.reduce_sites([[0.8], [0.6]])
     assert m.auc == pytest.approx(0.7, abs=1e-4)
+
+
+def test_dice_loss_binary():
+    from coinstac_dinunet_amd.metrics import dice_loss_binary
+    perfect = torch.tensor([1.0, 0.0, 1.0, 1.0])
+    target = torch.tensor([1.0, 0.0, 1.0, 1.0])
+    assert float(dice_loss_binary(perfect, target)) < 1e-4
+    worst = torch.tensor([0.0, 1.0, 0.0, 0.0])
+    assert float(dice_loss_binary(worst, target)) > 0.9
