@@ -83,3 +83,41 @@ def test_rccl_cluster_two_ranks_gloo(tmp_path):
     w0 = np.load(os.path.join(result_dir, 'weights_rank0.npy'))
     w1 = np.load(os.path.join(result_dir, 'weights_rank1.npy'))
     np.testing.assert_allclose(w0, w1, rtol=1e-5, atol=1e-6)
+
+
+def _rank_kfold(rank, world, port, root, result_dir):
+    os.environ.update(MASTER_ADDR='127.0.0.1', MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank))
+    sys.path.insert(0, TESTS_DIR)
+    from computations import TabularDataset, TabularTrainer, make_site_data
+    from coinstac_dinunet_amd.config.keys import Key, Mode
+    from coinstac_dinunet_amd.parallel.cluster import RcclCluster
+
+    local_kw = dict(task_id='tab', mode=Mode.TRAIN, batch_size=4, epochs=1,
+                    validation_epochs=1, local_iterations=1, num_folds=3,
+                    data_dir='data', num_class=2, seed_all=True, patience=1,
+                    verbose=False)
+    cluster = RcclCluster(root, local_kw=local_kw)
+    make_site_data(cluster.site.as_dict(), n_samples=18, seed=rank)
+    success, out = cluster.run(TabularTrainer, dataset_cls=TabularDataset,
+                               max_rounds=600)
+    assert success, f'rank {rank}: k-fold run did not finish'
+    if rank == 0:
+        assert len(cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]) == 3
+    # per-fold checkpoints exist on this rank
+    for fold in range(3):
+        d = os.path.join(cluster.site.outputDirectory, 'tab', f'fold_{fold}')
+        assert f'latest.tab-{fold}.pt' in os.listdir(d)
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+
+def test_rccl_cluster_kfold_gloo(tmp_path):
+    """3-fold cross-validation on the persistent process group: model
+    re-init per fold, grad-arena rebuild, per-fold checkpoints."""
+    result_dir = str(tmp_path / 'results2')
+    os.makedirs(result_dir)
+    port = _free_port()
+    mp.spawn(_rank_kfold, args=(2, port, str(tmp_path / 'kf'), result_dir),
+             nprocs=2, join=True)

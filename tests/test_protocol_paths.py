@@ -1,0 +1,77 @@
+"""Protocol paths not covered elsewhere: local pretraining election +
+weights relay (PRE_COMPUTATION), gradient accumulation, test-only mode."""
+import os
+
+import numpy as np
+import torch
+
+from coinstac_dinunet_amd import COINNLocal, COINNRemote
+from coinstac_dinunet_amd.config.keys import Key, Mode
+from coinstac_dinunet_amd.simulator import LoopbackCluster
+
+from computations import TabularDataset, TabularTrainer, make_site_data
+
+
+def _cluster(tmp_path, sizes, **extra):
+    cluster = LoopbackCluster(
+        str(tmp_path), n_sites=len(sizes),
+        site_data=lambda s: make_site_data(
+            s.as_dict(), n_samples=sizes[int(s.clientId[-1])],
+            seed=int(s.clientId[-1])))
+    kw = dict(task_id='tab', mode=Mode.TRAIN, batch_size=4, epochs=1,
+              validation_epochs=1, local_iterations=1,
+              split_ratio=(0.6, 0.2, 0.2), data_dir='data', num_class=2,
+              seed_all=True, patience=1, verbose=False)
+    kw.update(extra)
+
+    def make_local(cache, input, state):
+        return COINNLocal(cache=cache, input=input, state=state, **kw)
+
+    def make_remote(cache, input, state):
+        return COINNRemote(cache=cache, input=input, state=state)
+
+    return cluster, make_local, make_remote
+
+
+def test_pretrain_election_and_weight_relay(tmp_path):
+    """The max-train-data site pretrains; its weights.tar relays through
+    the remote and every site loads it at PRE_COMPUTATION."""
+    cluster, make_local, make_remote = _cluster(
+        tmp_path, sizes=[12, 28],  # site 1 has more data -> elected
+        pretrain_args={'epochs': 2})
+    success, _ = cluster.run(make_local, make_remote, TabularTrainer,
+                             dataset_cls=TabularDataset, max_rounds=400)
+    assert success
+    # election went to the larger site and pretraining actually ran there
+    assert cluster.site_caches[1].get('pretrain') is True
+    assert cluster.site_caches[0].get('pretrain') is False
+    assert len(cluster.site_caches[1].get(Key.TRAIN_LOG, [])) > 0
+    # relay artifact existed (remote republished it as pretrained_*.tar)
+    relayed = [f for f in os.listdir(cluster.sites[0].baseDirectory)
+               if f.startswith('pretrained_')]
+    assert relayed, 'pretrained weights never reached site 0'
+
+
+def test_gradient_accumulation_local_iterations(tmp_path):
+    """local_iterations=2: two micro-batches per reduce round; the shipped
+    gradient is their SUM (reference semantics, learner.py:32-47)."""
+    cluster, make_local, make_remote = _cluster(
+        tmp_path, sizes=[16, 16], local_iterations=2)
+    success, _ = cluster.run(make_local, make_remote, TabularTrainer,
+                             dataset_cls=TabularDataset, max_rounds=400)
+    assert success
+    m0 = cluster.site_caches[0]['nn']['net']
+    m1 = cluster.site_caches[1]['nn']['net']
+    for p0, p1 in zip(m0.parameters(), m1.parameters()):
+        assert torch.allclose(p0, p1, atol=1e-6)
+
+
+def test_unequal_site_sizes_stay_lockstep(tmp_path):
+    """Sites with different data volumes do not deadlock: the lagged site
+    reshuffles (VALIDATION_WAITING) and keeps training until quorum."""
+    cluster, make_local, make_remote = _cluster(tmp_path, sizes=[12, 32],
+                                                epochs=2, patience=2)
+    success, _ = cluster.run(make_local, make_remote, TabularTrainer,
+                             dataset_cls=TabularDataset, max_rounds=500)
+    assert success
+    assert cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
