@@ -7,13 +7,13 @@ C = ops.require_native()
 
 which = sys.argv[1] if len(sys.argv) > 1 else 'fwd'
 dev = torch.device('cuda:0')
-B = 64
+B = 16
 x = torch.randn(B, 32, 64, 64, 64, device=dev, dtype=torch.bfloat16)
 w = torch.randn(32, 32, 3, 3, 3, device=dev, dtype=torch.bfloat16) * 0.1
 go = torch.randn(B, 32, 64, 64, 64, device=dev, dtype=torch.bfloat16)
-for _ in range(6):
+for _ in range(4):
     if which == 'fwd':
-        C.conv3d_fwd_spatial(x, w)
+        C.conv3d_fwd_spatial(x, w, 1)
     elif which == 'wgrad':
         C.conv3d_wgrad(x, go, 1)
 torch.cuda.synchronize()
