@@ -1,0 +1,52 @@
+"""Guard the driver's bench.py contract: flags, JSON line, torchrun path."""
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+REQUIRED_KEYS = {'metric', 'value', 'unit', 'n_gpus', 'steps', 'warmup',
+                 'ms_per_step', 'higher_is_better', 'scaling', 'vs_baseline',
+                 'dtype', 'data', 'config'}
+
+
+def _run(cmd, timeout=300):
+    r = subprocess.run(cmd, cwd=REPO, capture_output=True, text=True,
+                       timeout=timeout)
+    assert r.returncode == 0, r.stderr[-2000:]
+    lines = [l for l in r.stdout.splitlines() if l.startswith('{')]
+    assert len(lines) == 1, f'expected ONE json line, got {len(lines)}'
+    return json.loads(lines[0])
+
+
+def test_bench_single_process(tmp_path):
+    out = _run([sys.executable, 'bench.py', '--steps', '2', '--warmup', '1',
+                '--batch', '2', '--vol', '8'])
+    assert REQUIRED_KEYS <= set(out)
+    assert out['n_gpus'] == 1 and out['steps'] == 2 and out['warmup'] == 1
+    assert out['scaling'] == 'weak' and out['data'] == 'synthetic'
+    assert out['value'] > 0
+
+
+def test_bench_torchrun_two_ranks():
+    """The driver's N>1 launch shape (gloo here, RCCL on the GPU node)."""
+    import socket
+    s = socket.socket()
+    s.bind(('127.0.0.1', 0))
+    port = s.getsockname()[1]
+    s.close()
+    out = _run([sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
+                '--nproc-per-node', '2', '--master-addr', '127.0.0.1',
+                '--master-port', str(port), 'bench.py', '--gpus', '2',
+                '--steps', '2', '--warmup', '1', '--batch', '2',
+                '--vol', '8'], timeout=420)
+    assert out['n_gpus'] == 2
+    assert out['config']['parallelism'] == 'dsgd-dp2'
+    assert out['config']['global_batch'] == 4
+
+
+def test_bench_mlp_model():
+    out = _run([sys.executable, 'bench.py', '--model', 'mlp', '--steps', '2',
+                '--warmup', '1', '--batch', '8'])
+    assert 'mlp' in out['metric']
