@@ -188,11 +188,15 @@ class COINNLocal:
                     trainer.data_handle.get_train_dataset(dataset_cls),
                     trainer.data_handle.get_validation_dataset(dataset_cls)))
 
-        elif self.out['phase'] == Phase.PRE_COMPUTATION and \
-                self.input.get('pretrained_weights'):
-            trainer.load_checkpoint(
-                file_path=self.state['baseDirectory'] + _sep +
-                self.input['pretrained_weights'])
+        elif self.out['phase'] == Phase.PRE_COMPUTATION:
+            if self.input.get('pretrained_weights'):
+                trainer.load_checkpoint(
+                    file_path=self.state['baseDirectory'] + _sep +
+                    self.input['pretrained_weights'])
+            # Deviation from the reference (local.py:208-212): when the
+            # elected site's pretraining never improved, no weights.tar
+            # exists — the reference then echoes PRE_COMPUTATION forever
+            # (site/remote deadlock). Proceed to COMPUTATION instead.
             self.out['phase'] = Phase.COMPUTATION
 
         learner = self._get_learner_cls(learner_cls)(trainer=trainer,
