@@ -2,14 +2,16 @@
 
 Kernel inventory (MI355X equivalents of the ops the reference implicitly
 runs through stock PyTorch — SURVEY.md §2.9):
-  - fused multi-tensor Adam / SGD step            (K4)
-  - flat gradient bucket pack / unpack            (K5/K7)
-  - fused log_softmax + NLL loss fwd/bwd + argmax (K3/K16)
-  - Prf1a confusion counts / KxK histogram        (K12/K13)
-  - MFMA bf16/f32 GEMM for the MLP family         (K2)
-  - implicit-GEMM Conv3d fwd/dgrad/wgrad          (K1)
-  - PowerSGD Gram-Schmidt + skinny GEMMs          (K8/K9)
-  - rankDAD power-iteration + reconstruction      (K10/K11)
+  - Conv3d fwd/dgrad/wgrad (K1): MFMA spatial-slab tap-reuse kernels
+    (stride-1 + stride-2 incl. parity-decomposed dgrad) with implicit-GEMM
+    fallbacks for odd shapes — conv3d_spatial.hip / conv3d.hip
+  - fused BatchNorm3d(+ReLU) fwd/bwd              — bnorm.hip
+  - fused multi-tensor Adam / SGD step            (K4) — adam.hip
+  - flat gradient bucket pack / unpack            (K5/K7) — pack.hip
+  - fused log_softmax + NLL loss fwd/bwd + argmax (K3/K16) — lsnll.hip
+  - Prf1a confusion counts / KxK histogram        (K12/K13) — metrics.hip
+  - MFMA linear with fused bias+ReLU              (K2) — linear.hip
+  - fused Gram-Schmidt for PowerSGD               (K8) — linear.hip
 
 The extension is built in-tree (`python setup.py build_ext --inplace` or
 __graft_entry__.build()) for gfx950 only. On a GPU box the native path is
