@@ -129,3 +129,69 @@ class RcclCluster:
             if success:
                 break
         return success, out
+
+
+# --------------------------------------------------------------------------
+# Mid-run checkpoint/resume (a capability the reference lacks: its "resume"
+# granularity is one platform iteration and a crashed run restarts the fold
+# — SURVEY §5.4). Here every rank can snapshot its whole protocol state
+# (cache scalars + model/optimizer state_dicts + the pending input dict)
+# and resume lock-step from the next round.
+# --------------------------------------------------------------------------
+_UNPICKLABLE_KEYS = ('train_loader_iter', 'nn', 'device', 'optimizer',
+                     'dataset', '_rccl_grad_buffer', 'frozen_args')
+
+
+def save_cluster_state(cluster, path):
+    import torch as _torch
+    from ..utils import jsonable  # noqa: F401 (documentational)
+    cache = {k: v for k, v in cluster.site_cache.items()
+             if k not in _UNPICKLABLE_KEYS}
+    models = {k: m.state_dict()
+              for k, m in cluster.site_cache.get('nn', {}).items()}
+    optims = {k: o.state_dict()
+              for k, o in cluster.site_cache.get('optimizer', {}).items()}
+    frozen = dict(cluster.site_cache.get('frozen_args', {}))
+    state = {'cache': cache, 'models': models, 'optimizers': optims,
+             'frozen_args': frozen, 'input': dict(cluster.input),
+             'rounds': cluster.rounds}
+    if cluster.rank == 0:
+        remote_cache = {k: v for k, v in cluster.remote_cache.items()
+                        if k not in _UNPICKLABLE_KEYS}
+        state['remote_cache'] = remote_cache
+    _torch.save(state, path)
+
+
+def load_cluster_state(cluster, path, trainer_cls, dataset_cls=None):
+    """Restore a snapshot into a freshly constructed RcclCluster."""
+    import torch as _torch
+    from ..data import COINNDataHandle
+    from ..utils import FrozenDict
+    state = _torch.load(path, weights_only=False)
+    cluster.site_cache.clear()
+    cluster.site_cache.update(state['cache'])
+    cluster.site_cache['cursor'] = 0  # loader iterator cannot be restored
+    if state.get('frozen_args'):
+        cluster.site_cache['frozen_args'] = FrozenDict(state['frozen_args'])
+    cluster.input = dict(state['input'])
+    cluster.rounds = state['rounds']
+    if cluster.rank == 0 and 'remote_cache' in state:
+        cluster.remote_cache.clear()
+        cluster.remote_cache.update(state['remote_cache'])
+    # rebuild model/optimizer into the cache registries
+    handle = COINNDataHandle(cache=cluster.site_cache, input=cluster.input,
+                             state=cluster.site.as_dict())
+    trainer = trainer_cls(data_handle=handle)
+    trainer.init_nn(init_model=True, init_optim=True, set_devices=True)
+    for k, sd in state['models'].items():
+        if k in trainer.nn:
+            m = trainer.nn[k]
+            (m.module if hasattr(m, 'module') else m).load_state_dict(sd)
+    for k, sd in state['optimizers'].items():
+        if k in trainer.optimizer:
+            trainer.optimizer[k].load_state_dict(sd)
+    # rebuild the dataset registry (loader iterators don't survive)
+    if dataset_cls is not None and cluster.site_cache.get('split_file'):
+        handle.get_train_dataset(dataset_cls)
+        handle.get_validation_dataset(dataset_cls)
+    return cluster
