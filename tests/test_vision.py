@@ -55,3 +55,17 @@ def test_clahe_shapes():
 def test_pix_neigh():
     assert len(imageutils.get_pix_neigh(3, 3)) == 8
     assert len(imageutils.get_pix_neigh(3, 3, eight=False)) == 4
+
+
+def test_image_loader_roundtrip(tmp_path):
+    from PIL import Image as PILImage
+    from coinstac_dinunet_amd.vision.imageutils import Image
+    arr = (np.random.RandomState(1).rand(16, 16, 3) * 255).astype(np.uint8)
+    PILImage.fromarray(arr).save(tmp_path / 'img.png')
+    mask = (np.random.RandomState(2).rand(16, 16) > 0.5).astype(np.uint8) * 255
+    PILImage.fromarray(mask).save(tmp_path / 'img_mask.png')
+    im = Image(dir=str(tmp_path), file='img.png').load()
+    assert im.array.shape == (16, 16, 3)
+    im.load_mask(str(tmp_path), fget_mask=lambda f: f.replace('.png', '_mask.png'))
+    im.apply_mask()
+    assert (im.array[mask == 0] == 0).all()
