@@ -1,14 +1,17 @@
 """Aggregator phase-machine driver.
 
-API-parity: /root/reference/coinstac_dinunet/distrib/nodes/remote.py:22-310
-(COINNRemote + _gather/check/EmptyDataHandle): adopts shared_args from the
-first site, drives the fold queue, epoch/mode transitions, global score
-reduction, best-model signaling and the results zip.
+Behavior-parity: /root/reference/coinstac_dinunet/distrib/nodes/remote.py
+:22-310 — adopts shared_args from the first site, drives the reversed
+fold queue, epoch/mode transitions, global score reduction, the
+best-model signal and the results zip. The wire contract (out-dict keys,
+phase/mode strings, quorum rules) is identical; the implementation is
+organized as a table of quorum-gated phase handlers instead of the
+reference's inline if-chain.
 """
 import datetime as _datetime
 import os as _os
-import time as _time
 import shutil as _shutil
+import time as _time
 import traceback as _tback
 
 from ... import config as _conf
@@ -21,6 +24,8 @@ from ..reducer import COINNReducer as _dSGDReducer
 
 
 class EmptyDataHandle:
+    """Remote-side trainer plumbing: cache/input/state without data."""
+
     def __init__(self, cache, input, state):
         self.cache = cache
         self.input = input
@@ -28,22 +33,23 @@ class EmptyDataHandle:
 
 
 def _gather(keys, data, mode='append'):
+    """Collect per-site values for `keys` across site out-dicts."""
     assert mode in ('append', 'extend'), f'Invalid gather mode: {mode}'
-    data = list(data)
     res = {k: [] for k in keys}
-    for k in res:
-        for d in data:
-            if not d.get(k):
+    for d in list(data):
+        for k in keys:
+            value = d.get(k)
+            if not value:
                 continue
             if mode == 'append':
-                res[k].append(d[k])
+                res[k].append(value)
             else:
-                res[k] = res[k] + d[k]
+                res[k] = res[k] + value
     return res
 
 
 def check(logic, k, v, kw):
-    """Quorum predicate over per-site input dicts."""
+    """Quorum predicate: logic(site[k] == v for every site input)."""
     return logic(site_vars.get(k) == v for site_vars in kw.values())
 
 
@@ -55,84 +61,211 @@ class COINNRemote:
         self.input = _FrozenDict(input if input is not None else {})
         self.state = _FrozenDict(state if state is not None else {})
         self.cache['verbose'] = verbose
-        if not self.cache.get(Key.ARGS_CACHED):
-            site = list(self.input.values())[0]
-            self.cache.update(**site['shared_args'])
-            self.cache[Key.ARGS_CACHED] = True
+        self._adopt_shared_args()
 
-    # ---- run lifecycle ---------------------------------------------------
-    def _init_runs(self):
-        self.cache.update(seed=self.cache.setdefault('seed', _conf.current_seed))
+    def _adopt_shared_args(self):
+        """Single source of truth for hyperparameters: the sites' frozen
+        args, adopted wholesale on first contact."""
+        if self.cache.get(Key.ARGS_CACHED):
+            return
+        first_site = next(iter(self.input.values()))
+        self.cache.update(**first_site['shared_args'])
+        self.cache[Key.ARGS_CACHED] = True
+
+    # =====================================================================
+    # compute: quorum-gated dispatch
+    # =====================================================================
+    def compute(self, mp_pool, trainer_cls, reducer_cls=_dSGDReducer, **kw):
+        trainer = trainer_cls(data_handle=EmptyDataHandle(
+            cache=self.cache, input=self.input, state=self.state))
+
+        self.out['phase'] = self.input.get('phase', Phase.INIT_RUNS)
+
+        if check(all, 'phase', Phase.INIT_RUNS, self.input):
+            self._handle_init_runs(trainer)
+        if check(all, 'phase', Phase.PRE_COMPUTATION, self.input):
+            self._handle_pre_computation()
+
+        # every round re-publishes the per-site mode map
+        self.out['global_modes'] = self._mode_map()
+
+        if check(all, 'phase', Phase.COMPUTATION, self.input):
+            self._handle_computation(trainer, reducer_cls, mp_pool)
+        if check(all, 'phase', Phase.NEXT_RUN_WAITING, self.input):
+            self._handle_run_end(trainer)
+
+    # ---- INIT_RUNS ------------------------------------------------------
+    def _handle_init_runs(self, trainer):
+        self.cache.update(seed=self.cache.setdefault('seed',
+                                                     _conf.current_seed))
         self.cache[Key.GLOBAL_TEST_SERIALIZABLE] = []
-        self.cache['data_size'] = {}
-        for site, site_vars in self.input.items():
-            self.cache['data_size'][site] = site_vars.get('data_size')
-        self.cache['folds'] = [{'split_ix': str(f), 'seed': self.cache['seed']}
-                               for f in range(self.cache['num_folds'])][::-1]
+        self.cache['data_size'] = {
+            site: site_vars.get('data_size')
+            for site, site_vars in self.input.items()}
+        # reversed so .pop() walks folds in ascending order
+        self.cache['folds'] = [
+            {'split_ix': str(f), 'seed': self.cache['seed']}
+            for f in range(self.cache['num_folds'])][::-1]
+        self.out['global_runs'] = self._open_fold(trainer)
+        self.out['phase'] = Phase.NEXT_RUN
 
-    def _next_run(self, trainer):
-        """Pop a fold, reset best-score state, elect the pretrain site."""
-        self.cache['fold'] = self.cache['folds'].pop()
+    def _open_fold(self, trainer):
+        """Pop the next fold, reset best-score state, elect the pretrain
+        site (the one with the most training data for this fold)."""
+        fold = self.cache['fold'] = self.cache['folds'].pop()
         self.cache['log_dir'] = _os.path.join(
             self.state['outputDirectory'], self.cache['task_id'],
-            f"fold_{self.cache['fold']['split_ix']}")
+            f"fold_{fold['split_ix']}")
         _os.makedirs(self.cache['log_dir'], exist_ok=True)
         trainer.init_nn(set_devices=True)
 
-        self.cache.update(epoch=0, best_val_epoch=0)
-        self.cache.update(best_val_score=0
-                          if self.cache['metric_direction'] == 'maximize'
-                          else _conf.max_size)
-        self.cache[Key.TRAIN_LOG] = []
-        self.cache[Key.VALIDATION_LOG] = []
-        self.cache[Key.TEST_METRICS] = []
+        maximize = self.cache['metric_direction'] == 'maximize'
+        self.cache.update(epoch=0, best_val_epoch=0,
+                          best_val_score=0 if maximize else _conf.max_size)
+        for log_key in (Key.TRAIN_LOG, Key.VALIDATION_LOG, Key.TEST_METRICS):
+            self.cache[log_key] = []
 
-        out = {}
-        fold_ix = self.cache['fold']['split_ix']
-        data_sizes = {st: self.cache['data_size'][st][fold_ix]['train']
-                      for st in self.input}
-        max_data_site = max(data_sizes, key=data_sizes.get)
-        for site in self.input:
-            fold = {**self.cache['fold']}
-            fold['pretrain'] = site == max_data_site
-            out[site] = fold
-        return out
+        train_sizes = {
+            site: self.cache['data_size'][site][fold['split_ix']]['train']
+            for site in self.input}
+        elected = max(train_sizes, key=train_sizes.get)
+        return {site: {**fold, 'pretrain': site == elected}
+                for site in self.input}
 
-    # ---- epoch accounting --------------------------------------------------
-    def _accumulate_epoch_info(self, trainer):
-        out = {}
-        train_scores = _gather([Key.TRAIN_SERIALIZABLE], self.input.values(),
-                               'extend')
-        train_scores = _gather(['averages', 'metrics'],
-                               train_scores[Key.TRAIN_SERIALIZABLE], 'append')
-        out['train_averages'] = trainer.new_averages()
-        out['train_averages'].reduce_sites(train_scores['averages'])
-        out['train_metrics'] = trainer.new_metrics()
-        out['train_metrics'].reduce_sites(train_scores['metrics'])
+    # ---- PRE_COMPUTATION -------------------------------------------------
+    def _handle_pre_computation(self):
+        """Relay the elected site's pretrained weights to every site."""
+        for site, site_vars in self.input.items():
+            wfile = site_vars.get('weights_file')
+            if wfile is None:
+                continue
+            relayed = f'pretrained_{_conf.weights_file}'
+            _shutil.copy(
+                _os.path.join(self.state['baseDirectory'], site, wfile),
+                _os.path.join(self.state['transferDirectory'], relayed))
+            self.out['pretrained_weights'] = relayed
+            break
+        self.out['phase'] = Phase.PRE_COMPUTATION
 
-        val_scores = _gather([Key.VALIDATION_SERIALIZABLE],
-                             self.input.values(), 'extend')
-        val_scores = _gather(['averages', 'metrics'],
-                             val_scores[Key.VALIDATION_SERIALIZABLE], 'append')
-        out['val_averages'] = trainer.new_averages()
-        out['val_averages'].reduce_sites(val_scores['averages'])
-        out['val_metrics'] = trainer.new_metrics()
-        out['val_metrics'].reduce_sites(val_scores['metrics'])
-        return out
+    # ---- COMPUTATION -----------------------------------------------------
+    def _handle_computation(self, trainer, reducer_cls, mp_pool):
+        reducer = self._get_reducer_cls(reducer_cls)(trainer=trainer,
+                                                     mp_pool=mp_pool)
+        self.out['phase'] = Phase.COMPUTATION
 
-    def _on_epoch_end(self, reducer):
-        epoch_info = self._accumulate_epoch_info(reducer.trainer)
+        if check(all, 'reduce', True, self.input):
+            self.out.update(**reducer.reduce())
+
+        if check(all, 'mode', Mode.VALIDATION_WAITING, self.input):
+            # epoch boundary: validate on cadence, else straight back to train
+            self.cache['epoch'] += 1
+            on_cadence = (self.cache['epoch'] %
+                          self.cache['validation_epochs'] == 0)
+            self.out['global_modes'] = self._mode_map(
+                Mode.VALIDATION if on_cadence else Mode.TRAIN)
+
+        if check(all, 'mode', Mode.TRAIN_WAITING, self.input):
+            epoch_info = self._close_epoch(reducer)
+            self.out['global_modes'] = self._mode_map(
+                self._next_mode(**epoch_info))
+
+    def _close_epoch(self, reducer):
+        info = self._reduce_site_scores(reducer.trainer)
         self.cache[Key.TRAIN_LOG].append(
-            [*epoch_info['train_averages'].get(),
-             *epoch_info['train_metrics'].get()])
-        self._save_if_better(**epoch_info)
-        if epoch_info.get('val_averages'):
+            [*info['train_averages'].get(), *info['train_metrics'].get()])
+        self._signal_save_best(**info)
+        if info.get('val_averages'):
             self.cache[Key.VALIDATION_LOG].append(
-                [*epoch_info['val_averages'].get(),
-                 *epoch_info['val_metrics'].get()])
+                [*info['val_averages'].get(), *info['val_metrics'].get()])
         if _lazy_debug(self.cache['epoch']):
             self._plot_progress()
-        return epoch_info
+        return info
+
+    def _reduce_site_scores(self, trainer):
+        """Cross-site reduction of the serialized train/val scores."""
+        out = {}
+        for prefix, key in (('train', Key.TRAIN_SERIALIZABLE),
+                            ('val', Key.VALIDATION_SERIALIZABLE)):
+            rows = _gather([key], self.input.values(), 'extend')[key]
+            parts = _gather(['averages', 'metrics'], rows, 'append')
+            averages = trainer.new_averages()
+            averages.reduce_sites(parts['averages'])
+            metrics = trainer.new_metrics()
+            metrics.reduce_sites(parts['metrics'])
+            out[f'{prefix}_averages'] = averages
+            out[f'{prefix}_metrics'] = metrics
+        return out
+
+    def _signal_save_best(self, **kw):
+        if kw.get('val_metrics'):
+            score = kw['val_metrics'].extract(self.cache['monitor_metric'])
+            self.out['save_current_as_best'] = performance_improved_(
+                self.cache['epoch'], score, self.cache)
+
+    def _next_mode(self, **kw):
+        done = self.cache['epoch'] > self.cache['epochs']
+        return Mode.TEST if done or self._stop_early(**kw) else Mode.TRAIN
+
+    def _stop_early(self, **kw):
+        return stop_training_(self.cache['epoch'], self.cache)
+
+    def _mode_map(self, mode=None):
+        return {site: (mode if mode else site_vars.get('mode', 'N/A'))
+                for site, site_vars in self.input.items()}
+
+    # ---- NEXT_RUN_WAITING / SUCCESS --------------------------------------
+    def _handle_run_end(self, trainer):
+        self._save_fold_scores(trainer)
+        if self.cache['folds']:
+            self.out['global_runs'] = self._open_fold(trainer)
+            self.out['phase'] = Phase.NEXT_RUN
+        else:
+            self.out.update(**self._publish_results(trainer))
+            self.out['phase'] = Phase.SUCCESS
+
+    def _save_fold_scores(self, trainer):
+        rows = _gather([Key.TEST_SERIALIZABLE], self.input.values(),
+                       'extend')[Key.TEST_SERIALIZABLE]
+        parts = _gather(['averages', 'metrics'], rows, 'append')
+        averages = trainer.new_averages()
+        averages.reduce_sites(parts['averages'])
+        metrics = trainer.new_metrics()
+        metrics.reduce_sites(parts['metrics'])
+
+        self.cache[Key.TEST_METRICS].append(
+            [*averages.get(), *metrics.get()])
+        self.cache[Key.GLOBAL_TEST_SERIALIZABLE].append(
+            {'averages': averages.serialize(),
+             'metrics': metrics.serialize()})
+        self._plot_progress()
+        _utils.save_scores(self.cache, self.cache['log_dir'],
+                           file_keys=[Key.TEST_METRICS])
+        snapshot = {**self.cache,
+                    Key.GLOBAL_TEST_SERIALIZABLE:
+                        self.cache[Key.GLOBAL_TEST_SERIALIZABLE][-1]}
+        _utils.save_cache(snapshot, self.cache['log_dir'])
+
+    def _publish_results(self, trainer):
+        """Reduce all folds' test scores, save CSV, zip everything."""
+        parts = _gather(['averages', 'metrics'],
+                        self.cache[Key.GLOBAL_TEST_SERIALIZABLE], 'append')
+        averages = trainer.new_averages()
+        averages.reduce_sites(parts['averages'])
+        metrics = trainer.new_metrics()
+        metrics.reduce_sites(parts['metrics'])
+        self.cache[Key.GLOBAL_TEST_METRICS] = [
+            [*averages.get(), *metrics.get()]]
+        task_dir = _os.path.join(self.state['outputDirectory'],
+                                 self.cache['task_id'])
+        _utils.save_scores(self.cache, log_dir=task_dir,
+                           file_keys=[Key.GLOBAL_TEST_METRICS])
+
+        stamp = '_'.join(str(_datetime.datetime.now()).split(' '))
+        zip_name = f"{self.cache['task_id']}_{self.cache['agg_engine']}_{stamp}"
+        _shutil.make_archive(
+            _os.path.join(self.state['transferDirectory'], zip_name),
+            'zip', task_dir)
+        return {'results_zip': zip_name}
 
     def _plot_progress(self):
         try:
@@ -142,131 +275,7 @@ class COINNRemote:
         except Exception:
             pass
 
-    def _on_run_end(self, trainer):
-        """Save this fold's globally reduced test score."""
-        test_scores = _gather([Key.TEST_SERIALIZABLE], self.input.values(),
-                              'extend')
-        test_scores = _gather(['averages', 'metrics'],
-                              test_scores[Key.TEST_SERIALIZABLE], 'append')
-        test_averages = trainer.new_averages()
-        test_averages.reduce_sites(test_scores['averages'])
-        test_metrics = trainer.new_metrics()
-        test_metrics.reduce_sites(test_scores['metrics'])
-
-        self.cache[Key.TEST_METRICS].append(
-            [*test_averages.get(), *test_metrics.get()])
-        self.cache[Key.GLOBAL_TEST_SERIALIZABLE].append(
-            {'averages': test_averages.serialize(),
-             'metrics': test_metrics.serialize()})
-        self._plot_progress()
-        _utils.save_scores(self.cache, self.cache['log_dir'],
-                           file_keys=[Key.TEST_METRICS])
-        _cache = {**self.cache}
-        _cache[Key.GLOBAL_TEST_SERIALIZABLE] = \
-            _cache[Key.GLOBAL_TEST_SERIALIZABLE][-1]
-        _utils.save_cache(_cache, self.cache['log_dir'])
-
-    def _send_global_scores(self, trainer):
-        out = {}
-        scores = _gather(['averages', 'metrics'],
-                         self.cache[Key.GLOBAL_TEST_SERIALIZABLE], 'append')
-        averages = trainer.new_averages()
-        averages.reduce_sites(scores['averages'])
-        metrics = trainer.new_metrics()
-        metrics.reduce_sites(scores['metrics'])
-
-        self.cache[Key.GLOBAL_TEST_METRICS] = [[*averages.get(), *metrics.get()]]
-        _utils.save_scores(self.cache,
-                           log_dir=self.state['outputDirectory'] + _os.sep +
-                           self.cache['task_id'],
-                           file_keys=[Key.GLOBAL_TEST_METRICS])
-
-        out['results_zip'] = f"{self.cache['task_id']}_{self.cache['agg_engine']}_"
-        out['results_zip'] += '_'.join(str(_datetime.datetime.now()).split(' '))
-        _shutil.make_archive(
-            f"{self.state['transferDirectory']}{_os.sep}{out['results_zip']}",
-            'zip',
-            self.state['outputDirectory'] + _os.sep + self.cache['task_id'])
-        return out
-
-    def _set_mode(self, mode=None):
-        return {site: (mode if mode else site_vars.get('mode', 'N/A'))
-                for site, site_vars in self.input.items()}
-
-    def _pre_compute(self):
-        """Relay the elected site's pretrained weights to everyone."""
-        out = {}
-        pt_path = None
-        for site, site_vars in self.input.items():
-            if site_vars.get('weights_file') is not None:
-                pt_path = self.state['baseDirectory'] + _os.sep + site + \
-                    _os.sep + site_vars['weights_file']
-                break
-        if pt_path is not None:
-            out['pretrained_weights'] = f'pretrained_{_conf.weights_file}'
-            _shutil.copy(pt_path, self.state['transferDirectory'] + _os.sep +
-                         out['pretrained_weights'])
-        return out
-
-    # ---- main dispatch -----------------------------------------------------
-    def compute(self, mp_pool, trainer_cls, reducer_cls=_dSGDReducer, **kw):
-        trainer = trainer_cls(data_handle=EmptyDataHandle(
-            cache=self.cache, input=self.input, state=self.state))
-
-        self.out['phase'] = self.input.get('phase', Phase.INIT_RUNS)
-        if check(all, 'phase', Phase.INIT_RUNS, self.input):
-            self._init_runs()
-            self.out['global_runs'] = self._next_run(trainer)
-            self.out['phase'] = Phase.NEXT_RUN
-
-        if check(all, 'phase', Phase.PRE_COMPUTATION, self.input):
-            self.out.update(**self._pre_compute())
-            self.out['phase'] = Phase.PRE_COMPUTATION
-
-        self.out['global_modes'] = self._set_mode()
-        if check(all, 'phase', Phase.COMPUTATION, self.input):
-            reducer = self._get_reducer_cls(reducer_cls)(trainer=trainer,
-                                                         mp_pool=mp_pool)
-            self.out['phase'] = Phase.COMPUTATION
-            if check(all, 'reduce', True, self.input):
-                self.out.update(**reducer.reduce())
-
-            if check(all, 'mode', Mode.VALIDATION_WAITING, self.input):
-                self.cache['epoch'] += 1
-                if self.cache['epoch'] % self.cache['validation_epochs'] == 0:
-                    self.out['global_modes'] = self._set_mode(mode=Mode.VALIDATION)
-                else:
-                    self.out['global_modes'] = self._set_mode(mode=Mode.TRAIN)
-
-            if check(all, 'mode', Mode.TRAIN_WAITING, self.input):
-                epoch_info = self._on_epoch_end(reducer)
-                nxt_epoch = self._next_epoch(**epoch_info)
-                self.out['global_modes'] = self._set_mode(mode=nxt_epoch['mode'])
-
-        if check(all, 'phase', Phase.NEXT_RUN_WAITING, self.input):
-            self._on_run_end(trainer)
-            if len(self.cache['folds']) > 0:
-                self.out['global_runs'] = self._next_run(trainer)
-                self.out['phase'] = Phase.NEXT_RUN
-            else:
-                self.out.update(**self._send_global_scores(trainer))
-                self.out['phase'] = Phase.SUCCESS
-
-    def _next_epoch(self, **kw):
-        epochs_done = self.cache['epoch'] > self.cache['epochs']
-        if epochs_done or self._stop_early(**kw):
-            return {'mode': Mode.TEST}
-        return {'mode': Mode.TRAIN}
-
-    def _save_if_better(self, **kw):
-        if kw.get('val_metrics'):
-            val_score = kw['val_metrics'].extract(self.cache['monitor_metric'])
-            self.out['save_current_as_best'] = performance_improved_(
-                self.cache['epoch'], val_score, self.cache)
-
-    def _stop_early(self, **kw):
-        return stop_training_(self.cache['epoch'], self.cache)
-
+    # ---- engine selection --------------------------------------------------
     def _get_reducer_cls(self, reducer_cls):
         engine = self.cache.get('agg_engine')
         if engine == AGG_Engine.dSGD:
