@@ -35,6 +35,7 @@ CASES = [
     (1, 64, 128, 8, 8, 8, 2),
     (3, 4, 8, 9, 11, 13, 1),   # odd sizes
     (1, 16, 16, 7, 7, 7, 2),
+    (1, 32, 64, 32, 32, 32, 2),  # routes the parity-spatial s2 dgrad
 ]
 
 
@@ -76,6 +77,11 @@ def test_conv3d_dgrad(dev, case):
         dx2 = C.conv3d_dgrad_spatial(go.to(torch.bfloat16),
                                      w.to(torch.bfloat16), list(x.shape))
         torch.testing.assert_close(dx2.float(), x.grad, rtol=5e-2,
+                                   atol=5e-2 * (Cout * 27) ** 0.5 * 0.2)
+    if s == 2 and ((W + 1) // 2) % 8 == 0 and Cout >= 32:
+        dx3 = C.conv3d_dgrad_s2_spatial(go.to(torch.bfloat16),
+                                        w.to(torch.bfloat16), list(x.shape))
+        torch.testing.assert_close(dx3.float(), x.grad, rtol=5e-2,
                                    atol=5e-2 * (Cout * 27) ** 0.5 * 0.2)
 
 
