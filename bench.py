@@ -124,8 +124,10 @@ def main():
         dist.barrier()
     elapsed = time.perf_counter() - t0
 
-    # MAX over ranks (slowest rank defines the lock-step round time)
-    t = torch.tensor([elapsed], dtype=torch.float64)
+    # MAX over ranks (slowest rank defines the lock-step round time);
+    # the tensor must live on the device for the RCCL backend
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if on_gpu else 'cpu')
     if dist.is_initialized() and world > 1:
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
     elapsed = float(t.item())
