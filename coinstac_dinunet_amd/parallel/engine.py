@@ -48,7 +48,10 @@ def init_distributed(backend=None, timeout_s=300):
     if backend is None:
         backend = 'nccl' if torch.cuda.is_available() else 'gloo'
     if backend == 'nccl':
-        torch.cuda.set_device(int(os.environ.get('LOCAL_RANK', 0)))
+        # modulo: lets N ranks share fewer GPUs (dev boxes); on the 8-GPU
+        # node LOCAL_RANK < device_count and this is the identity
+        torch.cuda.set_device(int(os.environ.get('LOCAL_RANK', 0)) %
+                              max(1, torch.cuda.device_count()))
     dist.init_process_group(backend=backend,
                             timeout=_dt.timedelta(seconds=timeout_s))
     return dist.get_rank(), dist.get_world_size()
