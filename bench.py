@@ -76,8 +76,10 @@ def main():
 
     rank, world = init_distributed()
     on_gpu = torch.cuda.is_available()
-    device = torch.device(f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}"
-                          if on_gpu else 'cpu')
+    local = int(os.environ.get('LOCAL_RANK', 0))
+    device = torch.device(
+        f"cuda:{local % max(1, torch.cuda.device_count())}"
+        if on_gpu else 'cpu')
     if on_gpu:
         torch.cuda.set_device(device)
         torch.backends.cudnn.benchmark = True
