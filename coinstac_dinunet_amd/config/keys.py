@@ -45,7 +45,8 @@ class AGG_Engine:
 
 
 class GatherMode:
-    # Declared for API parity with the reference (config/keys.py:47-49);
-    # consumed by the RCCL reducer to choose all-reduce vs all-gather.
+    # Declared for API parity with the reference (config/keys.py:47-49).
+    # Like the reference, nothing consumes it yet; the engines choose their
+    # collective semantics themselves (dSGD/powerSGD reduce, rankDAD gather).
     REDUCE = 'reduce'
     GATHER = 'gather'

@@ -77,3 +77,42 @@ def test_unequal_site_sizes_stay_lockstep(tmp_path):
                              dataset_cls=TabularDataset, max_rounds=500)
     assert success
     assert cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
+
+
+def test_four_site_quorum(tmp_path):
+    """Quorum logic at N=4 (all earlier protocol tests use 2 sites)."""
+    cluster, make_local, make_remote = _cluster(
+        tmp_path, sizes=[12, 16, 20, 24])
+    success, _ = cluster.run(make_local, make_remote, TabularTrainer,
+                             dataset_cls=TabularDataset, max_rounds=500)
+    assert success
+    m0 = cluster.site_caches[0]['nn']['net']
+    for i in (1, 2, 3):
+        mi = cluster.site_caches[i]['nn']['net']
+        for p0, p1 in zip(m0.parameters(), mi.parameters()):
+            assert torch.allclose(p0, p1, atol=1e-6)
+
+
+def test_run_is_reproducible_with_fixed_seed(tmp_path):
+    """Same seed + same data => bitwise-identical final weights."""
+    finals = []
+    for rep in range(2):
+        cluster, make_local, make_remote = _cluster(
+            tmp_path / f'rep{rep}', sizes=[16, 16], seed=123, seed_all=True)
+        success, _ = cluster.run(make_local, make_remote, TabularTrainer,
+                                 dataset_cls=TabularDataset, max_rounds=400)
+        assert success
+        net = cluster.site_caches[0]['nn']['net']
+        finals.append(torch.cat([p.detach().reshape(-1)
+                                 for p in net.parameters()]))
+    assert torch.equal(finals[0], finals[1])
+
+
+def test_load_sparse_test_mode(tmp_path):
+    """load_sparse=True: one dataset per test file during TEST."""
+    cluster, make_local, make_remote = _cluster(
+        tmp_path, sizes=[16, 16], load_sparse=True)
+    success, _ = cluster.run(make_local, make_remote, TabularTrainer,
+                             dataset_cls=TabularDataset, max_rounds=400)
+    assert success
+    assert cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
