@@ -220,6 +220,13 @@ class COINNDataHandle:
             loader = self.get_loader(handle_key=handle_key, shuffle=shuffle,
                                      dataset=dataset, use_padded_sampler=True,
                                      total_size=self.cache.get('lockstep_total_size'))
+            if loader is None:
+                raise RuntimeError(
+                    f"'{handle_key}' split is empty on site "
+                    f"{self.state.get('clientId', '?')} — with k-fold "
+                    "splitting, k=2 leaves NO training files (test and "
+                    "validation take both parts); use num_folds >= 3 or a "
+                    "split_ratio.")
             self.cache['data_len'] = len(loader) * self.cache['batch_size']
             self.cache['train_loader_iter'] = iter(loader)
 
