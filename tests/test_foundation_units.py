@@ -1,0 +1,64 @@
+"""Foundation utils: persistence, logger, ops CPU fallbacks."""
+import json
+import os
+import time
+
+import torch
+
+from coinstac_dinunet_amd import ops, utils
+from coinstac_dinunet_amd.utils import logger
+
+
+def test_save_scores_csv(tmp_path):
+    cache = {'log_header': 'Loss|Acc', 'test_metrics': [[0.5, 0.9], [0.4, 0.92]]}
+    utils.save_scores(cache, str(tmp_path), file_keys=['test_metrics'])
+    lines = open(tmp_path / 'test_metrics.csv').read().strip().split('\n')
+    assert lines[0] == 'Loss|Acc'
+    assert lines[1] == '0.5,0.9'
+
+
+def test_save_cache_stringifies_tensors(tmp_path):
+    cache = {'a': 1, 'weird': torch.randn(3), 'nested': {'t': torch.zeros(2)},
+             'log_dir': str(tmp_path)}
+    utils.save_cache(cache, str(tmp_path))
+    loaded = json.load(open(tmp_path / 'logs.json'))
+    assert loaded['a'] == 1
+    assert isinstance(loaded['weird'], str)  # stringified, not crashed
+
+
+def test_logger_duration():
+    cache = {}
+    t0 = time.time()
+    time.sleep(0.01)
+    with logger.duration(cache, 'phase'):
+        time.sleep(0.01)
+    assert len(cache['phase']) == 1 and cache['phase'][0] >= 0.005
+    from coinstac_dinunet_amd.utils.utils import duration
+    duration(cache, t0, 'span')
+    assert cache['span'][0] >= 0.01
+
+
+def test_ops_cpu_fallbacks():
+    logits = torch.randn(6, 3)
+    target = torch.tensor([0, 1, 2, 0, 1, 2])
+    loss = ops.cross_entropy(logits, target)
+    ref = torch.nn.functional.cross_entropy(logits, target)
+    torch.testing.assert_close(loss, ref)
+    torch.testing.assert_close(ops.argmax_rows(logits),
+                               torch.argmax(logits, 1))
+    x, w, b = torch.randn(4, 5), torch.randn(2, 5), torch.randn(2)
+    torch.testing.assert_close(ops.linear(x, w, b),
+                               torch.nn.functional.linear(x, w, b))
+    torch.testing.assert_close(
+        ops.linear(x, w, b, relu=True),
+        torch.relu(torch.nn.functional.linear(x, w, b)))
+
+
+def test_native_available_false_on_cpu():
+    assert ops.native_available() is False  # no GPU in this container
+    try:
+        ops.require_native()
+        raised = False
+    except RuntimeError:
+        raised = True
+    assert raised
