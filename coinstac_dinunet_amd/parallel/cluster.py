@@ -94,11 +94,12 @@ class RcclCluster:
             shutil.rmtree(p) if os.path.isdir(p) else os.remove(p)
 
     # -- one protocol round -------------------------------------------------
-    def run_round(self, trainer_cls, dataset_cls=None, mp_pool=None, **kw):
+    def run_round(self, trainer_cls, dataset_cls=None, mp_pool=None,
+                  learner_cls=RcclLearner, reducer_cls=RcclReducer, **kw):
         local = COINNLocal(cache=self.site_cache, input=dict(self.input),
                            state=self.site.as_dict(), **self.local_kw)
         result = local(mp_pool, trainer_cls, dataset_cls=dataset_cls,
-                       learner_cls=RcclLearner, **kw)
+                       learner_cls=learner_cls, **kw)
         my_out = result['output']
 
         gathered = [None] * self.world_size
@@ -111,7 +112,7 @@ class RcclCluster:
             self._site_transfer_to_remote(site_outs)
             remote = COINNRemote(cache=self.remote_cache, input=site_outs,
                                  state=self.remote_state, **self.remote_kw)
-            rres = remote(mp_pool, trainer_cls, reducer_cls=RcclReducer)
+            rres = remote(mp_pool, trainer_cls, reducer_cls=reducer_cls)
             self._remote_transfer_to_sites(site_outs)
             bcast = [(rres['output'], rres.get('success', False))]
         dist.broadcast_object_list(bcast, src=0)
