@@ -42,7 +42,7 @@ class _Conv3dFn(torch.autograd.Function):
         xb, wb = ctx.saved_tensors
         go = grad_out.to(torch.bfloat16).contiguous()
         gx = gw = gb = None
-        if ctx.x_requires:
+        if ctx.needs_input_grad[0]:
             wsub = (xb.size(4) + 1) // 2
             hsub = (xb.size(3) + 1) // 2
             if (ctx.stride == 1 and xb.size(4) % 8 == 0
@@ -57,8 +57,9 @@ class _Conv3dFn(torch.autograd.Function):
             else:
                 gx = C.conv3d_dgrad(go, wb, list(xb.shape),
                                     ctx.stride).to(ctx.in_dtype)
-        gw = C.conv3d_wgrad(xb, go, ctx.stride).to(ctx.w_dtype)
-        if ctx.has_bias:
+        if ctx.needs_input_grad[1]:
+            gw = C.conv3d_wgrad(xb, go, ctx.stride).to(ctx.w_dtype)
+        if ctx.has_bias and ctx.needs_input_grad[2]:
             gb = C.channel_sum(go)
         return gx, gw, gb, None
 
