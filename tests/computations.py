@@ -14,17 +14,22 @@ from coinstac_dinunet_amd.models import FreeSurferMLP
 N_FEATURES = 16
 
 
-def make_site_data(state, n_samples=24, n_features=N_FEATURES, seed=0):
+def make_site_data(state, n_samples=24, n_features=N_FEATURES, seed=0,
+                   n_classes=2):
     """Write per-sample .npy files (features + label) into baseDirectory."""
     rng = np.random.RandomState(seed)
     data_dir = os.path.join(state['baseDirectory']
                             if isinstance(state, dict) else state.baseDirectory,
                             'data')
     os.makedirs(data_dir, exist_ok=True)
-    w = rng.randn(n_features)
+    w = rng.randn(n_features, n_classes - 1)
     for i in range(n_samples):
         x = rng.randn(n_features).astype(np.float32)
-        y = int(x @ w > 0)
+        if n_classes == 2:
+            y = int(x @ w[:, 0] > 0)
+        else:
+            scores = np.concatenate([[0.0], x @ w])
+            y = int(np.argmax(scores))
         np.save(os.path.join(data_dir, f'sample_{i:03d}.npy'),
                 {'x': x, 'y': y}, allow_pickle=True)
 
@@ -61,6 +66,9 @@ class TabularTrainer(COINNTrainer):
         avg = self.new_averages()
         avg.add(loss.item(), len(inputs))
         metrics = self.new_metrics()
-        metrics.add(pred, labels)
+        if self.cache.get('monitor_metric') == 'auc':
+            metrics.add(torch.softmax(out.detach(), 1)[:, 1], labels)
+        else:
+            metrics.add(pred, labels)
         return {'loss': loss, 'averages': avg, 'metrics': metrics,
                 'output': pred}

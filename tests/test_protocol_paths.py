@@ -116,3 +116,52 @@ def test_load_sparse_test_mode(tmp_path):
                              dataset_cls=TabularDataset, max_rounds=400)
     assert success
     assert cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
+
+
+def test_test_only_mode(tmp_path):
+    """mode='test': no training rounds — straight to distributed test."""
+    cluster, make_local, make_remote = _cluster(
+        tmp_path, sizes=[16, 16], mode=Mode.TEST)
+    success, _ = cluster.run(make_local, make_remote, TabularTrainer,
+                             dataset_cls=TabularDataset, max_rounds=60)
+    assert success, f'test-only run took >{cluster.rounds} rounds'
+    assert cluster.rounds < 20  # no epoch loop happened
+    assert cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
+
+
+def test_auc_monitor_metric(tmp_path):
+    """monitor_metric='auc' routes through AUCROCMetrics end to end."""
+    cluster, make_local, make_remote = _cluster(
+        tmp_path, sizes=[16, 16], monitor_metric='auc')
+    success, _ = cluster.run(make_local, make_remote, TabularTrainer,
+                             dataset_cls=TabularDataset, max_rounds=400)
+    assert success
+    scores = cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
+    assert scores and len(scores[0]['metrics']) == 1  # [auc]
+
+
+def test_multiclass_confusion_matrix_metrics(tmp_path):
+    """num_class=3 routes through ConfusionMatrix (serialize = [acc,
+    per-class prec, per-class rec])."""
+    import functools
+    cluster = LoopbackCluster(
+        str(tmp_path), n_sites=2,
+        site_data=lambda s: make_site_data(s.as_dict(), n_samples=18,
+                                           seed=int(s.clientId[-1]),
+                                           n_classes=3))
+    kw = dict(task_id='tab', mode=Mode.TRAIN, batch_size=4, epochs=1,
+              validation_epochs=1, local_iterations=1,
+              split_ratio=(0.6, 0.2, 0.2), data_dir='data', num_class=3,
+              seed_all=True, patience=1, verbose=False)
+
+    def make_local(cache, input, state):
+        return COINNLocal(cache=cache, input=input, state=state, **kw)
+
+    def make_remote(cache, input, state):
+        return COINNRemote(cache=cache, input=input, state=state)
+
+    success, _ = cluster.run(make_local, make_remote, TabularTrainer,
+                             dataset_cls=TabularDataset, max_rounds=400)
+    assert success
+    ser = cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE][0]['metrics']
+    assert len(ser) == 3 and len(ser[1]) == 3 and len(ser[2]) == 3
