@@ -34,13 +34,16 @@ DAD_FILE = 'dad_data.npy'
 DAD_AGG_FILE = 'reduced_dad_data.npy'
 
 
-def power_iteration_BC(B, C, rank=10, numiterations=5, tol=1e-3):
+def power_iteration_BC(B, C, rank=10, numiterations=5, tol=1e-3,
+                       generator=None):
     """Top-`rank` structure of G = B @ C^T without forming G.
 
     B [n, k], C [m, k] (k = flattened batch). Returns (Bf [n, r], Cf [m, r])
     with G ~= Bf @ Cf^T; Bf columns carry the singular values. Deflated
     power iteration: each new direction is orthogonalized against the
     already-extracted ones via the accumulated sigma^2-weighted projector.
+    `generator` (CPU) makes the start vectors deterministic so independent
+    ranks recompressing the same factors produce bit-identical results.
     """
     device = B.device
     n, k = B.shape
@@ -55,7 +58,10 @@ def power_iteration_BC(B, C, rank=10, numiterations=5, tol=1e-3):
     bs, cs, sigmas = [], [], []
     lam1 = None  # first (largest) eigenvalue of G G^T, sets the zero scale
     for _ in range(rank):
-        b = _torch.rand(n, device=device)
+        if generator is not None:
+            b = _torch.rand(n, generator=generator).to(device)
+        else:
+            b = _torch.rand(n, device=device)
         degenerate = False
         for _ in range(numiterations):
             if small_k:
