@@ -113,7 +113,9 @@ def clahe_equalize(img, clip_limit=2.0, grid=(8, 8)):
     return out.astype(_np.uint8)
 
 
-def rescale2d(arr, lo=0.0, hi=255.0):
+def rescale2d(arr, lo=0.0, hi=1.0):
+    """Min-max rescale to [lo, hi] (reference default: unit range,
+    imageutils.py:154-157)."""
     arr = arr.astype(_np.float64)
     mn, mx = arr.min(), arr.max()
     if mx - mn < 1e-12:
@@ -121,27 +123,93 @@ def rescale2d(arr, lo=0.0, hi=255.0):
     return (arr - mn) / (mx - mn) * (hi - lo) + lo
 
 
-def get_praf1(pred, true, eps=1e-5):
-    """Precision/recall/accuracy/F1 on binary arrays."""
-    pred = _np.asarray(pred).astype(bool)
-    true = _np.asarray(true).astype(bool)
-    tp = int((pred & true).sum())
-    fp = int((pred & ~true).sum())
-    fn = int((~pred & true).sum())
-    tn = int((~pred & ~true).sum())
-    p = tp / max(tp + fp, eps)
-    r = tp / max(tp + fn, eps)
-    a = (tp + tn) / max(tp + fp + fn + tn, eps)
-    f1 = (2 * p * r) / max(p + r, eps)
-    return {'precision': p, 'recall': r, 'accuracy': a, 'f1': f1}
+def rescale3d(arrays):
+    """Per-slice unit rescale (reference imageutils.py:160-161)."""
+    return [rescale2d(a) for a in arrays]
 
 
-def expand_and_mirror_patch(image, patch_region, pad):
-    """Mirror-pad the patch window (U-Net style context tiling)."""
-    (r0, r1), (c0, c1) = patch_region
-    pr0, pr1, pc0, pc1 = pad
-    padded = _np.pad(image, ((pr0, pr1), (pc0, pc1)), mode='reflect')
-    return padded[r0:r1 + pr0 + pr1, c0:c1 + pc0 + pc1]
+def get_signed_diff_int8(image_arr1, image_arr2):
+    """Signed pixel difference rescaled to uint8 for visual diffing
+    (reference imageutils.py:164-168)."""
+    signed = _np.asarray(image_arr1 - image_arr2, dtype=_np.int8)
+    shifted = _np.asarray(signed - signed.min(), dtype=_np.uint8)
+    return _np.asarray(rescale2d(shifted) * 255, dtype=_np.uint8)
+
+
+def whiten_image2d(img_arr2d):
+    """Zero-mean/unit-std whiten, then stretch back to uint8
+    (reference imageutils.py:171-174)."""
+    z = (img_arr2d - img_arr2d.mean()) / max(float(img_arr2d.std()), 1e-12)
+    return _np.asarray(rescale2d(z) * 255, dtype=_np.uint8)
+
+
+def _binarize_255(arr):
+    a = _np.asarray(arr).copy()
+    a[a == 255] = 1
+    return a.astype(_np.int64)
+
+
+def get_rgb_scores(arr_2d, truth):
+    """RGB overlay of prediction vs ground truth: white=TP, green=FP,
+    red=FN, black=TN (reference imageutils.py:88-107)."""
+    xy = _binarize_255(arr_2d) + 2 * _binarize_255(truth)
+    rgb = _np.zeros([xy.shape[0], xy.shape[1], 3], dtype=_np.uint8)
+    rgb[xy == 3] = [255, 255, 255]
+    rgb[xy == 1] = [0, 255, 0]
+    rgb[xy == 2] = [255, 0, 0]
+    return rgb
+
+
+def get_praf1(arr_2d, truth):
+    """Precision/recall/accuracy/F1 between binary (0/1 or 0/255) arrays,
+    rounded to 5 places (reference imageutils.py:110-151)."""
+    xy = _binarize_255(arr_2d) + 2 * _binarize_255(truth)
+    tp = int((xy == 3).sum())
+    fp = int((xy == 1).sum())
+    fn = int((xy == 2).sum())
+    tn = int((xy == 0).sum())
+    p = tp / (tp + fp) if tp + fp else 0
+    r = tp / (tp + fn) if tp + fn else 0
+    a = (tp + tn) / (tp + fp + fn + tn) if tp + fp + fn + tn else 0
+    f1 = 2 * p * r / (p + r) if p + r else 0
+    return {'Precision': round(p, 5), 'Recall': round(r, 5),
+            'Accuracy': round(a, 5), 'F1': round(f1, 5)}
+
+
+def map_img_to_img2d(map_to, img):
+    """Burn a binary overlay into an image (gray promoted to RGB; overlay
+    pixels turn red — reference imageutils.py:290-301)."""
+    arr = _np.asarray(map_to).copy()
+    if arr.ndim == 2:
+        rgb = _np.stack([arr, arr, arr], -1).astype(_np.uint8)
+    else:
+        rgb = arr.astype(_np.uint8)
+    on = _np.asarray(img) == 255
+    rgb[..., 0][on] = 255
+    rgb[..., 1][on] = 0
+    rgb[..., 2][on] = 0
+    return rgb
+
+
+def expand_and_mirror_patch(full_img_shape, orig_patch_indices, expand_by):
+    """Expanded window around a patch for U-Net-style context tiling:
+    returns the clamped window indices plus the reflect-pad spec covering
+    whatever margin fell outside the image (reference
+    imageutils.py:253-279). Apply via
+    np.pad(img[a:b, c:d], pad_spec, mode='reflect')."""
+    half_i, half_j = int(expand_by[0] / 2), int(expand_by[1] / 2)
+    p, q, r, s = orig_patch_indices
+    a, b, c, d = p - half_i, q + half_i, r - half_j, s + half_j
+    pad = [0, 0, 0, 0]
+    if a < 0:
+        pad[0], a = half_i - p, 0
+    if b > full_img_shape[0]:
+        pad[1], b = b - full_img_shape[0], full_img_shape[0]
+    if c < 0:
+        pad[2], c = half_j - r, 0
+    if d > full_img_shape[1]:
+        pad[3], d = d - full_img_shape[1], full_img_shape[1]
+    return a, b, c, d, [(pad[0], pad[1]), (pad[2], pad[3])]
 
 
 def get_chunk_indexes(img_shape, chunk_shape, offset=None):
@@ -192,9 +260,50 @@ def remove_small_cc(binary_arr, min_size=64):
     return keep[labels]
 
 
-def get_pix_neigh(i, j, eight=True):
-    """Pixel neighborhood coordinates (4- or 8-connectivity)."""
+def get_pix_neigh(i, j, eight=False):
+    """Pixel neighborhood coordinates (4- or 8-connectivity; reference
+    imageutils.py:328-348 defaults to 4)."""
     n4 = [(i - 1, j), (i + 1, j), (i, j - 1), (i, j + 1)]
     if not eight:
         return n4
     return n4 + [(i - 1, j - 1), (i - 1, j + 1), (i + 1, j - 1), (i + 1, j + 1)]
+
+
+def get_chunk_indices_by_index(img_shape, chunk_shape, indices):
+    """Windows of chunk_shape centered on the given (row, col) points,
+    clamped inside the image (reference imageutils.py:211-226)."""
+    x, y = chunk_shape
+    out = []
+    w, h = img_shape[:2]
+    for (c1, c2) in indices:
+        p, q = c1 - x // 2, c1 + x // 2
+        r, s = c2 - y // 2, c2 + y // 2
+        if p < 0:
+            p, q = 0, x
+        if q > w:
+            p, q = w - x, w
+        if r < 0:
+            r, s = 0, y
+        if s > h:
+            r, s = h - y, h
+        out.append([int(p), int(q), int(r), int(s)])
+    return out
+
+
+def remove_connected_comp(segmented_img, connected_comp_diam_limit=20):
+    """Zero out connected components whose bounding-box diagonal is below
+    the diameter limit (reference imageutils.py:304-325; vectorized via
+    per-label bounding boxes instead of a per-pixel loop)."""
+    assert _HAS_SCIPY, 'scipy is required for remove_connected_comp'
+    img = _np.asarray(segmented_img).copy()
+    labeled, n = _ndi.label(img, _np.ones((3, 3), dtype=int))
+    if n == 0:
+        return img
+    for sl, lab in zip(_ndi.find_objects(labeled), range(1, n + 1)):
+        if sl is None:
+            continue
+        dy = sl[0].stop - sl[0].start - 1
+        dx = sl[1].stop - sl[1].start - 1
+        if _math.sqrt(dy * dy + dx * dx) < connected_comp_diam_limit:
+            img[labeled == lab] = 0
+    return img
