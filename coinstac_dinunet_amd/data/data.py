@@ -216,7 +216,13 @@ class COINNDataHandle:
         flips out['mode'] to VALIDATION_WAITING and resets the cursor."""
         out = {}
         if self.cache.get('cursor', 0) == 0:
-            dataset = self.dataset[handle_key]
+            # registry first; fall back to an injected dataset
+            # (dataloader_args['train']['dataset']) — the reference KeyErrors
+            # here for injected train datasets (data.py:179 reads only the
+            # registry that get_dataset fills, data.py:99-103).
+            dataset = self.dataset.get(handle_key)
+            if dataset is None:
+                dataset = self.dataloader_args.get(handle_key, {}).get('dataset')
             loader = self.get_loader(handle_key=handle_key, shuffle=shuffle,
                                      dataset=dataset, use_padded_sampler=True,
                                      total_size=self.cache.get('lockstep_total_size'))

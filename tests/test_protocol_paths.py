@@ -165,3 +165,58 @@ def test_multiclass_confusion_matrix_metrics(tmp_path):
     assert success
     ser = cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE][0]['metrics']
     assert len(ser) == 3 and len(ser[1]) == 3 and len(ser[2]) == 3
+
+
+def test_fp16_wire_format(tmp_path):
+    """precision_bits=16: gradients ship as float16 (reference
+    learner.py:17 dtype rule); sites stay in sync and the run finishes."""
+    cluster, make_local, make_remote = _cluster(
+        tmp_path, sizes=[16, 16], precision_bits=16)
+    success, _ = cluster.run(make_local, make_remote, TabularTrainer,
+                             dataset_cls=TabularDataset, max_rounds=400)
+    assert success
+    m0 = cluster.site_caches[0]['nn']['net']
+    m1 = cluster.site_caches[1]['nn']['net']
+    for p0, p1 in zip(m0.parameters(), m1.parameters()):
+        assert torch.allclose(p0, p1, atol=1e-6)  # same fp16 avg everywhere
+    # the wire artifact itself is half precision
+    import glob
+    grads = sorted(glob.glob(os.path.join(
+        str(tmp_path), 'transfer_local0', '*grads*.npy')))
+    if grads:  # transfer dir layout is cluster-internal; check when present
+        arrs = np.load(grads[-1], allow_pickle=True)
+        assert arrs[0].dtype == np.float16
+
+
+def test_injected_datasets_via_dataloader_args(tmp_path):
+    """dataloader_args={'train': {'dataset': ...}} bypasses split files
+    entirely (reference data.py:105-139 injected branch)."""
+    cluster = LoopbackCluster(
+        str(tmp_path), n_sites=2,
+        site_data=lambda s: make_site_data(s.as_dict(), n_samples=16,
+                                           seed=int(s.clientId[-1])))
+    kw = dict(task_id='tab', mode=Mode.TRAIN, batch_size=4, epochs=1,
+              validation_epochs=1, local_iterations=1,
+              split_ratio=(0.6, 0.2, 0.2), data_dir='data', num_class=2,
+              seed_all=True, patience=1, verbose=False)
+
+    def make_local(cache, input, state):
+        files = sorted(os.listdir(os.path.join(state['baseDirectory'], 'data')))
+        dl_args = {}
+        for mode_key, sl in (('train', slice(0, 10)),
+                             ('validation', slice(10, 13)),
+                             ('test', slice(13, 16))):
+            ds = TabularDataset(mode=mode_key, cache=cache, input=input,
+                                state=state)
+            ds.add(files=files[sl])
+            dl_args[mode_key] = {'dataset': ds}
+        return COINNLocal(cache=cache, input=input, state=state,
+                          dataloader_args=dl_args, **kw)
+
+    def make_remote(cache, input, state):
+        return COINNRemote(cache=cache, input=input, state=state)
+
+    success, _ = cluster.run(make_local, make_remote, TabularTrainer,
+                             dataset_cls=TabularDataset, max_rounds=400)
+    assert success
+    assert cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
