@@ -15,7 +15,6 @@ import os as _os
 import shutil as _shutil
 import time as _time
 import traceback as _tback
-from os import sep as _sep
 
 from ... import config as _conf
 from ... import utils as _utils

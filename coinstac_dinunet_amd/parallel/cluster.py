@@ -145,7 +145,6 @@ _UNPICKLABLE_KEYS = ('train_loader_iter', 'nn', 'device', 'optimizer',
 
 def save_cluster_state(cluster, path):
     import torch as _torch
-    from ..utils import jsonable  # noqa: F401 (documentational)
     cache = {k: v for k, v in cluster.site_cache.items()
              if k not in _UNPICKLABLE_KEYS}
     models = {k: m.state_dict()

@@ -1,6 +1,4 @@
 """PowerSGD and rankDAD engines: unit math + full loopback protocol runs."""
-import numpy as np
-import pytest
 import torch
 
 from coinstac_dinunet_amd import COINNLocal, COINNRemote

@@ -1,6 +1,5 @@
 """Foundation utils: persistence, logger, ops CPU fallbacks."""
 import json
-import os
 import time
 
 import torch

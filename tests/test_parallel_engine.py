@@ -1,5 +1,4 @@
 """RCCL engine tests on CPU: flat grad arena + 2-process gloo cluster."""
-import json
 import os
 import socket
 import sys

@@ -143,7 +143,6 @@ def test_auc_monitor_metric(tmp_path):
 def test_multiclass_confusion_matrix_metrics(tmp_path):
     """num_class=3 routes through ConfusionMatrix (serialize = [acc,
     per-class prec, per-class rec])."""
-    import functools
     cluster = LoopbackCluster(
         str(tmp_path), n_sites=2,
         site_data=lambda s: make_site_data(s.as_dict(), n_samples=18,

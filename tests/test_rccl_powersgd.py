@@ -28,7 +28,6 @@ def _rank_main(rank, world, port, root, result_dir):
     from coinstac_dinunet_amd.parallel.cluster import RcclCluster
     from coinstac_dinunet_amd.parallel.powersgd import (RcclPowerSGDLearner,
                                                         RcclPowerSGDReducer)
-    import coinstac_dinunet_amd.parallel.cluster as cluster_mod
 
     local_kw = dict(task_id='tab', mode=Mode.TRAIN, batch_size=4, epochs=1,
                     validation_epochs=1, local_iterations=1,

@@ -1,9 +1,6 @@
 """Unit tests: checkpoint format, train_local, site runner, FedAvg engine."""
-import json
 import os
 
-import numpy as np
-import pytest
 import torch
 
 from coinstac_dinunet_amd import COINNLocal, COINNRemote
