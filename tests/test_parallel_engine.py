@@ -120,3 +120,17 @@ def test_rccl_cluster_kfold_gloo(tmp_path):
     port = _free_port()
     mp.spawn(_rank_kfold, args=(2, port, str(tmp_path / 'kf'), result_dir),
              nprocs=2, join=True)
+
+
+def test_rccl_cluster_three_ranks_gloo(tmp_path):
+    """Odd world size (3 sites, unequal data): quorum and the all-reduce
+    mean must not assume a power-of-2 group."""
+    result_dir = str(tmp_path / 'results3')
+    os.makedirs(result_dir)
+    port = _free_port()
+    mp.spawn(_rank_main, args=(3, port, str(tmp_path / 'cluster3'),
+                               result_dir), nprocs=3, join=True)
+    w0 = np.load(os.path.join(result_dir, 'weights_rank0.npy'))
+    for r in (1, 2):
+        wr = np.load(os.path.join(result_dir, f'weights_rank{r}.npy'))
+        np.testing.assert_allclose(w0, wr, rtol=1e-5, atol=1e-6)
