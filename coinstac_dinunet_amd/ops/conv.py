@@ -4,11 +4,18 @@ GPU path (bf16): ops._hip_ops.conv3d_{fwd,dgrad,wgrad}; CPU falls back to
 torch.nn.functional.conv3d. Restricted to the VBM workload's conv family:
 3x3x3 kernels, padding 1, stride 1 or 2, NCDHW.
 """
+import os
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
 from . import native_available, require_native
+
+# Opt-in (unvalidated this round): route Cin<16 stride-1 forwards through
+# the CTILE=1 spatial instances instead of the igemm fallback. Flip the
+# default once measured on hardware.
+_SPATIAL_CI1 = os.environ.get('COINN_SPATIAL_CI1', '0') == '1'
 
 
 class _Conv3dFn(torch.autograd.Function):
@@ -22,8 +29,10 @@ class _Conv3dFn(torch.autograd.Function):
         ow = (xb.size(4) + 2 - 3) // stride + 1
         oh = (xb.size(3) + 2 - 3) // stride + 1
         min_chunk = 64
-        if (ow % 8 == 0 and xb.size(1) >= 16 and oh * ow >= min_chunk):
-            out = C.conv3d_fwd_spatial(xb, wb, stride)
+        ci_ok = xb.size(1) >= 16 or (_SPATIAL_CI1 and stride == 1)
+        if (ow % 8 == 0 and ci_ok and oh * ow >= min_chunk):
+            out = C.conv3d_fwd_spatial(xb, wb, stride,
+                                       1 if xb.size(1) < 16 else 0)
         else:
             out = C.conv3d_fwd(xb, wb, stride)
         if bias is not None:

@@ -45,7 +45,7 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go, int64_t stride);
 torch::Tensor channel_sum(torch::Tensor go);
 // conv3d_spatial.hip
 torch::Tensor conv3d_fwd_spatial(torch::Tensor x, torch::Tensor w,
-                                 int64_t stride);
+                                 int64_t stride, int64_t ctile_opt);
 torch::Tensor conv3d_dgrad_spatial(torch::Tensor go, torch::Tensor w,
                                    std::vector<int64_t> in_shape);
 torch::Tensor conv3d_dgrad_s2_spatial(torch::Tensor go, torch::Tensor w,
@@ -80,7 +80,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3d_dgrad", &conv3d_dgrad);
   m.def("conv3d_wgrad", &conv3d_wgrad);
   m.def("channel_sum", &channel_sum);
-  m.def("conv3d_fwd_spatial", &conv3d_fwd_spatial);
+  m.def("conv3d_fwd_spatial", &conv3d_fwd_spatial,
+        py::arg("x"), py::arg("w"), py::arg("stride"),
+        py::arg("ctile_opt") = 0);
   m.def("conv3d_dgrad_spatial", &conv3d_dgrad_spatial);
   m.def("conv3d_dgrad_s2_spatial", &conv3d_dgrad_s2_spatial);
   m.def("bn3d_fwd", &bn3d_fwd);

@@ -203,7 +203,8 @@ static torch::Tensor prep_wb(torch::Tensor w_flat2d, int KCH, int ctile) {
 }
 
 static void launch_spatial(torch::Tensor in, torch::Tensor wb,
-                           torch::Tensor out, SpDims sd, int stride) {
+                           torch::Tensor out, SpDims sd, int stride,
+                           int ctile = 0) {
   int OWT = sd.TW % 32 == 0 ? 32 : (sd.TW % 16 == 0 ? 16 : 8);
   int chunk = stride == 1 ? 256 : 128;
   // small images (d8-class): 64-position chunks keep the grid dense
@@ -223,7 +224,14 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
   auto L = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, dim3(256), 0, s, ip, wp, op, sd, nchunks);
   };
-  if (stride == 1) {
+  if (stride == 1 && ctile == 1) {
+    // single-channel (first-layer) instances: one 32-k-step covers the
+    // whole 27-tap K, slab is [1][3][H2][W2]
+    if (chunk == 64) L(conv3d_spatial_kernel<8, 1, 1, 64>);
+    else if (OWT == 32) L(conv3d_spatial_kernel<32, 1, 1>);
+    else if (OWT == 16) L(conv3d_spatial_kernel<16, 1, 1>);
+    else L(conv3d_spatial_kernel<8, 1, 1>);
+  } else if (stride == 1) {
     if (chunk == 64) L(conv3d_spatial_kernel<8, 1, 32, 64>);
     else if (OWT == 32) L(conv3d_spatial_kernel<32, 1, 32>);
     else if (OWT == 16) L(conv3d_spatial_kernel<16, 1, 32>);
@@ -237,7 +245,7 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
 }
 
 torch::Tensor conv3d_fwd_spatial(torch::Tensor x, torch::Tensor w,
-                                 int64_t stride) {
+                                 int64_t stride, int64_t ctile_opt) {
   CHECK_GPU(x);
   auto xc = x.contiguous();
   auto wc = w.to(torch::kBFloat16).contiguous();
@@ -250,13 +258,15 @@ torch::Tensor conv3d_fwd_spatial(torch::Tensor x, torch::Tensor w,
   sd.TD = (sd.D + 2 - 3) / (int)stride + 1;
   sd.TH = (sd.H + 2 - 3) / (int)stride + 1;
   sd.TW = (sd.W + 2 - 3) / (int)stride + 1;
-  int ctile = stride == 1 ? 32 : 16;
+  // ctile_opt=1 opts into the CTILE=1 single-channel instances
+  // (stride 1 only); 0 = the default tiling.
+  int ctile = stride == 1 ? ((ctile_opt == 1) ? 1 : 32) : 16;
   auto wb = prep_wb(wc.reshape({sd.NCOL, (int64_t)sd.KCH * 27}), sd.KCH,
                     ctile);
   sd.Kpad = (int)wb.size(1);
   auto out = torch::empty({sd.N, sd.NCOL, sd.TD, sd.TH, sd.TW},
                           xc.options());
-  launch_spatial(xc, wb, out, sd, (int)stride);
+  launch_spatial(xc, wb, out, sd, (int)stride, ctile);
   return out;
 }
 

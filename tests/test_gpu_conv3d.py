@@ -116,3 +116,23 @@ def test_ops_conv3d_module_autograd(dev):
     assert x.grad is not None and m.weight.grad is not None \
         and m.bias.grad is not None
     assert torch.isfinite(m.weight.grad).all()
+
+
+@pytest.mark.parametrize('case', [(2, 1, 32, 16, 16, 16),
+                                  (1, 1, 8, 16, 32, 32),
+                                  (2, 4, 16, 8, 16, 16)])
+def test_conv3d_fwd_spatial_ctile1(dev, case):
+    """CTILE=1 single-channel spatial instances (opt-in routing for the
+    Cin<16 first layer; COINN_SPATIAL_CI1=1 enables them in the module).
+    The kernel entry is callable directly regardless of the env gate."""
+    import os
+    if os.environ.get('COINN_SPATIAL_CI1') != '1':
+        pytest.skip('CTILE=1 instances not opted in (COINN_SPATIAL_CI1=1)')
+    N, Cin, Cout, D, H, W = case
+    torch.manual_seed(4)
+    x = torch.randn(N, Cin, D, H, W, device=dev, dtype=torch.bfloat16)
+    w = torch.randn(Cout, Cin, 3, 3, 3, device=dev, dtype=torch.bfloat16) * 0.2
+    out = C.conv3d_fwd_spatial(x, w, 1, 1)
+    ref = _ref_conv(x, w, 1)
+    torch.testing.assert_close(out.float(), ref, rtol=5e-2,
+                               atol=5e-2 * (Cin * 27) ** 0.5 * 0.2)
