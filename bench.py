@@ -159,6 +159,10 @@ def main():
                 'volume': f'{args.vol}^3' if args.model == 'vbm' else None,
                 'local_iterations': args.local_iterations,
                 'parallelism': f'dsgd-dp{world}',
+                # BASELINE.json names "wall-clock/epoch" too: derived for a
+                # nominal 1024-sample per-site epoch at this step time
+                'wallclock_per_epoch_s_1024spp': round(
+                    -(-1024 // args.batch) * ms_per_step / 1000.0, 3),
             },
         }))
     if dist.is_initialized():
