@@ -193,7 +193,15 @@ class Prf1a(COINNMetrics):
 
 
 class ConfusionMatrix(COINNMetrics):
-    """Multi-class K x K confusion matrix; macro precision/recall/F1."""
+    """Multi-class K x K confusion matrix; macro precision/recall/F1.
+
+    Layout: matrix[true][pred] (standard). DOCUMENTED DEVIATION: the
+    reference builds matrix[pred][true] (metrics.py:243-249) yet divides
+    diag by column/row sums as if rows were true — so its precision() and
+    recall() are swapped relative to the standard definitions. We keep the
+    standard ones; accuracy and (macro) F1 are identical either way since
+    F1 is symmetric under the precision<->recall swap
+    (tests/test_reference_parity.py pins this equivalence)."""
 
     def __init__(self, num_classes=2, **kw):
         super().__init__(**kw)
@@ -246,11 +254,17 @@ class ConfusionMatrix(COINNMetrics):
             else [round(v, self.num_precision) for v in rec.tolist()]
 
     def f1(self, average=True):
+        if average:
+            # reference semantics (metrics.py:269-276): harmonic mean of the
+            # AVERAGED precision/recall, not the mean of per-class F1s —
+            # and invariant under the reference's p<->r swap
+            p, r = self.precision(True), self.recall(True)
+            return round(2 * p * r / max(p + r, self.eps),
+                         self.num_precision)
         p = np.asarray(self.precision(average=False))
         r = np.asarray(self.recall(average=False))
         f = (2 * p * r) / np.clip(p + r, self.eps, None)
-        return round(float(f.mean()), self.num_precision) if average \
-            else [round(v, self.num_precision) for v in f.tolist()]
+        return [round(v, self.num_precision) for v in f.tolist()]
 
     def get(self):
         return [self.accuracy, self.f1(), self.precision(), self.recall()]

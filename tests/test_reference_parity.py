@@ -1,0 +1,172 @@
+"""Differential parity: run the UPSTREAM reference implementation (mounted
+read-only at /root/reference) against ours on identical inputs and demand
+identical outputs — split generation, checkpoint-best predicates, and the
+metric serialize/reduce_sites protocol are the behaviors COINSTAC
+computations depend on bit-for-bit.
+
+These tests import the reference solely as a black-box oracle; they skip
+wherever the mount is absent (e.g. on GPU boxes, which only get /root/repo).
+"""
+import json
+import os
+import sys
+import types
+
+import numpy as np
+import pytest
+import torch
+
+REF_ROOT = '/root/reference'
+pytestmark = pytest.mark.skipif(
+    not os.path.isdir(os.path.join(REF_ROOT, 'coinstac_dinunet')),
+    reason='reference mount not present')
+
+
+@pytest.fixture(scope='module')
+def ref():
+    """Import the reference package (cv2 stubbed: not in this image)."""
+    if 'cv2' not in sys.modules:
+        cv2 = types.ModuleType('cv2')
+        cv2.createCLAHE = lambda **kw: None
+        sys.modules['cv2'] = cv2
+    if not hasattr(np, 'float'):  # reference uses the numpy<2 alias
+        np.float = float
+    sys.path.insert(0, REF_ROOT)
+    import coinstac_dinunet as r
+    yield r
+    sys.path.remove(REF_ROOT)
+
+
+def _files(n, prefix='subj'):
+    return [f'{prefix}_{i:03d}.npy' for i in range(n)]
+
+
+@pytest.mark.parametrize('n,ratio', [(20, (0.6, 0.2, 0.2)),
+                                     (17, (0.7, 0.15, 0.15)),
+                                     (9, (0.5, 0.5)),
+                                     (33, (0.8, 0.1, 0.1))])
+def test_ratio_split_bitwise(ref, tmp_path, n, ratio):
+    from coinstac_dinunet.data import datautils as ref_du
+    from coinstac_dinunet_amd.data import datautils as our_du
+    a, b = tmp_path / 'ref', tmp_path / 'ours'
+    a.mkdir(), b.mkdir()
+    ref_du.create_ratio_split(_files(n), {'split_dir': str(a),
+                                          'split_ratio': list(ratio)})
+    our_du.create_ratio_split(_files(n), {'split_dir': str(b),
+                                          'split_ratio': list(ratio)})
+    ra, rb = sorted(os.listdir(a)), sorted(os.listdir(b))
+    assert ra == rb
+    for f in ra:
+        sa = json.load(open(a / f))
+        sb = json.load(open(b / f))
+        assert sa == sb, f'{f} differs'
+
+
+@pytest.mark.parametrize('n,k', [(20, 4), (18, 3), (23, 5)])
+def test_kfold_split_bitwise(ref, tmp_path, n, k):
+    from coinstac_dinunet.data import datautils as ref_du
+    from coinstac_dinunet_amd.data import datautils as our_du
+    a, b = tmp_path / 'ref', tmp_path / 'ours'
+    a.mkdir(), b.mkdir()
+    ref_du.create_k_fold_splits(_files(n), {'num_folds': k,
+                                            'split_dir': str(a)})
+    our_du.create_k_fold_splits(_files(n), {'num_folds': k,
+                                            'split_dir': str(b)})
+    ra, rb = sorted(os.listdir(a)), sorted(os.listdir(b))
+    assert ra == rb
+    for f in ra:
+        assert json.load(open(a / f)) == json.load(open(b / f)), f
+    # every fold partitions the full set
+    for f in rb:
+        s = json.load(open(b / f))
+        assert sorted(s['train'] + s['validation'] + s['test']) == \
+            sorted(_files(n))
+
+
+@pytest.mark.parametrize('direction,scores', [
+    ('maximize', [0.5, 0.6, 0.60005, 0.7, 0.69, 0.7001]),
+    ('minimize', [1.0, 0.9, 0.89995, 0.5, 0.51, 0.4999]),
+])
+def test_performance_improved_matches(ref, direction, scores):
+    from coinstac_dinunet.utils.utils import (performance_improved_ as rp,
+                                              stop_training_ as rs)
+    from coinstac_dinunet_amd.utils.utils import (performance_improved_ as op,
+                                                  stop_training_ as os_)
+    ca = {'metric_direction': direction, 'patience': 2, 'epochs': 10,
+          'best_val_epoch': 0,
+          'best_val_score': 0 if direction == 'maximize' else 1e11}
+    cb = dict(ca)
+    for ep, sc in enumerate(scores):
+        assert rp(ep, sc, ca) == op(ep, sc, cb), (ep, sc)
+        assert rs(ep, ca) == os_(ep, cb), (ep, sc)
+        assert ca == cb  # identical cache mutations
+
+
+def test_prf1a_protocol_matches(ref):
+    from coinstac_dinunet.metrics import Prf1a as RefP
+    from coinstac_dinunet_amd.metrics import Prf1a as OurP
+    rng = np.random.RandomState(7)
+    ra, oa = RefP(), OurP()
+    rb, ob = RefP(), OurP()
+    for m_ref, m_our in ((ra, oa), (rb, ob)):
+        for _ in range(3):
+            pred = torch.from_numpy(rng.randint(0, 2, 50))
+            true = torch.from_numpy(rng.randint(0, 2, 50))
+            m_ref.add(pred.clone(), true.clone())
+            m_our.add(pred.clone(), true.clone())
+    assert ra.serialize() == oa.serialize()
+    assert [ra.f1, ra.accuracy, ra.precision, ra.recall] == \
+           [oa.f1, oa.accuracy, oa.precision, oa.recall]
+    # accumulate parity
+    ra.accumulate(rb), oa.accumulate(ob)
+    assert ra.serialize() == oa.serialize()
+    # reduce_sites parity (mutates the instance; serialized site scores
+    # -> floor values)
+    site_scores = [[0.8, 0.7, 0.9], [0.6, 0.5, 0.4]]
+    rr, ro = RefP(), OurP()
+    rr.reduce_sites(site_scores)
+    ro.reduce_sites(site_scores)
+    assert rr.serialize() == ro.serialize()
+    assert rr.f1 == ro.f1 and rr.accuracy == ro.accuracy
+
+
+def test_averages_protocol_matches(ref):
+    from coinstac_dinunet.metrics import COINNAverages as RefA
+    from coinstac_dinunet_amd.metrics import COINNAverages as OurA
+    ra, oa = RefA(num_averages=2), OurA(num_averages=2)
+    for v, n, i in [(0.5, 4, 0), (0.25, 8, 1), (1.5, 2, 0)]:
+        ra.add(v, n, index=i)
+        oa.add(v, n, index=i)
+    assert ra.serialize() == oa.serialize()
+    np.testing.assert_allclose(np.asarray(ra.get(), dtype=float),
+                               np.asarray(oa.get(), dtype=float))
+    sites = [ra.serialize(), [[1.0, 2.0], [3, 5]]]
+    rred, ored = RefA(num_averages=2), OurA(num_averages=2)
+    rred.reduce_sites(sites)
+    ored.reduce_sites(sites)
+    assert rred.serialize() == ored.serialize()
+
+
+def test_confusion_matrix_protocol_matches(ref):
+    """The reference indexes matrix[pred][true] (metrics.py:243-249), so
+    its precision()/recall() are swapped vs the standard definitions we
+    use — accuracy and macro-F1 are invariant under that swap (F1 is
+    symmetric in p<->r). Pin exactly that equivalence."""
+    from coinstac_dinunet.metrics import ConfusionMatrix as RefC
+    from coinstac_dinunet_amd.metrics import ConfusionMatrix as OurC
+    rng = np.random.RandomState(11)
+    rc, oc = RefC(num_classes=3), OurC(num_classes=3)
+    for _ in range(3):
+        pred = torch.from_numpy(rng.randint(0, 3, 60))
+        true = torch.from_numpy(rng.randint(0, 3, 60))
+        rc.add(pred.clone(), true.clone())
+        oc.add(pred.clone(), true.clone())
+    assert torch.equal(rc.matrix.long(), oc.matrix.t())  # transposed layouts
+    np.testing.assert_allclose(rc.accuracy(), oc.accuracy, atol=1e-4)
+    np.testing.assert_allclose(rc.f1(average=True),
+                               oc.f1(average=True), atol=1e-4)
+    # the reference's 'precision' is the standard recall and vice versa
+    np.testing.assert_allclose(rc.precision(average=False),
+                               oc.recall(average=False), atol=1e-4)
+    np.testing.assert_allclose(rc.recall(average=False),
+                               oc.precision(average=False), atol=1e-4)
