@@ -31,7 +31,10 @@ def initialize_weights(module):
     """
     for m in module.modules():
         if isinstance(m, (torch.nn.Conv2d, torch.nn.Conv3d, torch.nn.Linear)):
-            torch.nn.init.kaiming_normal_(m.weight, mode='fan_out', nonlinearity='relu')
+            # kaiming defaults (fan_in/leaky_relu) — bit-parity with the
+            # reference's init (tensorutils.py:32), pinned by
+            # tests/test_reference_parity.py::test_initialize_weights_bitwise
+            torch.nn.init.kaiming_normal_(m.weight)
             if m.bias is not None:
                 torch.nn.init.constant_(m.bias, 0)
         elif isinstance(m, (torch.nn.BatchNorm1d, torch.nn.BatchNorm2d, torch.nn.BatchNorm3d)):

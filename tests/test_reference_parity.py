@@ -170,3 +170,40 @@ def test_confusion_matrix_protocol_matches(ref):
                                oc.recall(average=False), atol=1e-4)
     np.testing.assert_allclose(rc.recall(average=False),
                                oc.precision(average=False), atol=1e-4)
+
+
+@pytest.mark.parametrize('n,bs,shuffle,drop_last',
+                         [(10, 4, True, False), (13, 4, False, False),
+                          (3, 8, True, False), (10, 4, True, True)])
+def test_padded_sampler_sequences_match(ref, n, bs, shuffle, drop_last):
+    """Identical padded index sequences per (seed, epoch) — the property
+    lock-step batch counts depend on."""
+    from coinstac_dinunet.data.data import COINNPaddedDataSampler as RefS
+    from coinstac_dinunet_amd.data.data import COINNPaddedDataSampler as OurS
+    ds = list(range(n))
+    rs = RefS(ds, bs, seed=5, shuffle=shuffle, drop_last=drop_last)
+    os_ = OurS(ds, bs, seed=5, shuffle=shuffle, drop_last=drop_last)
+    for epoch in range(3):
+        rs.set_epoch(epoch), os_.set_epoch(epoch)
+        assert list(iter(rs)) == list(iter(os_)), epoch
+    assert len(rs) == len(os_)
+
+
+def test_initialize_weights_bitwise(ref):
+    """Same torch seed => bit-identical Kaiming init on the module types
+    the reference covers (Conv2d/Linear/BatchNorm2d)."""
+    from coinstac_dinunet.utils.tensorutils import (initialize_weights as ri)
+    from coinstac_dinunet_amd.utils.tensorutils import (initialize_weights
+                                                        as oi)
+
+    def build():
+        torch.manual_seed(42)
+        return torch.nn.Sequential(
+            torch.nn.Conv2d(3, 8, 3), torch.nn.BatchNorm2d(8),
+            torch.nn.Flatten(), torch.nn.Linear(8, 4))
+
+    ma, mb = build(), build()
+    torch.manual_seed(99), ri(ma)
+    torch.manual_seed(99), oi(mb)
+    for pa, pb in zip(ma.parameters(), mb.parameters()):
+        assert torch.equal(pa, pb)
