@@ -25,15 +25,21 @@ class Mode:
 
 
 class Key:
-    ARGS_CACHED = 'args_cached'
+    # values match the reference wire/cache strings exactly
+    # (config/keys.py:22-38) — they appear in out-dicts, logs.json and CSVs
+    ARGS_CACHED = '_args_cached_'
     TRAIN_LOG = 'train_log'
+    TRAIN_METRICS = 'train_metrics'
+    TRAIN_SERIALIZABLE = 'serializable_train_scores'
     VALIDATION_LOG = 'validation_log'
+    VALIDATION_METRICS = 'validation_metrics'
+    VALIDATION_SERIALIZABLE = 'serializable_validation_scores'
+    TEST_LOG = 'test_log'
     TEST_METRICS = 'test_metrics'
+    TEST_SERIALIZABLE = 'serializable_test_scores'
+    GLOBAL_TEST_LOG = 'global_test_log'
     GLOBAL_TEST_METRICS = 'global_test_metrics'
-    GLOBAL_TEST_SERIALIZABLE = 'global_test_serializable'
-    TRAIN_SERIALIZABLE = 'train_serializable'
-    VALIDATION_SERIALIZABLE = 'validation_serializable'
-    TEST_SERIALIZABLE = 'test_serializable'
+    GLOBAL_TEST_SERIALIZABLE = 'serializable_global_test_scores'
     DATA_CURSOR = 'data_cursor'
     DATA_LEN = 'data_len'
 

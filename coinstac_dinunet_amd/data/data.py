@@ -204,6 +204,12 @@ class COINNDataHandle:
         if use_padded_sampler:
             loader_args['drop_last'] = False
             loader_args['shuffle'] = False
+            # DOCUMENTED DEVIATION: the requested shuffle flag reaches the
+            # sampler here (seeded, epoch-aware). The reference builds its
+            # sampler AFTER forcing loader shuffle False (data.py:163-171),
+            # so its padded train loader never shuffles despite the
+            # sampler's seeded-shuffle support — train batches repeat in
+            # file order every epoch.
             loader_args['sampler'] = COINNPaddedDataSampler(
                 loader_args['dataset'], loader_args['batch_size'],
                 seed=args.get('seed', 0), shuffle=bool(args.get('shuffle')),

@@ -1,0 +1,195 @@
+"""Whole-protocol differential: drive the UPSTREAM reference package's
+COINNLocal/COINNRemote through the in-process loopback relay on the SAME
+site data, seeds and hyperparameters as our stack, and require the two
+runs to agree — round count, reduced global scores, and final weights.
+
+This is the strongest drop-in-compatibility statement the CPU can make:
+the same user computation (module arch, iteration math) trained by both
+frameworks converges to the same model through the same protocol.
+Skips when /root/reference is not mounted (e.g. GPU boxes).
+"""
+import itertools
+import os
+import sys
+import types
+
+import numpy as np
+import pytest
+import torch
+
+TESTS_DIR = os.path.dirname(os.path.abspath(__file__))
+REF_ROOT = '/root/reference'
+pytestmark = pytest.mark.skipif(
+    not os.path.isdir(os.path.join(REF_ROOT, 'coinstac_dinunet')),
+    reason='reference mount not present')
+
+sys.path.insert(0, TESTS_DIR)
+
+
+class _FakePool:
+    """starmap-only stand-in for the platform's multiprocessing pool."""
+
+    @staticmethod
+    def starmap(fn, iterable):
+        return list(itertools.starmap(fn, iterable))
+
+
+def _import_reference():
+    if 'cv2' not in sys.modules:
+        cv2 = types.ModuleType('cv2')
+        cv2.createCLAHE = lambda **kw: None
+        sys.modules['cv2'] = cv2
+    if not hasattr(np, 'float'):
+        np.float = float
+    if REF_ROOT not in sys.path:
+        sys.path.insert(0, REF_ROOT)
+    import coinstac_dinunet  # noqa: F401
+
+
+_KW = dict(task_id='tab', batch_size=4, epochs=2, validation_epochs=1,
+           local_iterations=1, split_ratio=(0.6, 0.2, 0.2), data_dir='data',
+           num_class=2, seed_all=True, patience=2, verbose=False)
+
+
+def _make_reference_classes():
+    """Reference-API computation: same module arch as our TabularTrainer."""
+    from coinstac_dinunet import COINNDataset as RefDataset
+    from coinstac_dinunet import COINNTrainer as RefTrainer
+    from coinstac_dinunet_amd.models import FreeSurferMLP
+
+    class RefTabularDataset(RefDataset):
+        def load_index(self, file):
+            self.indices.append(file)
+
+        def __getitem__(self, ix):
+            rec = np.load(os.path.join(self.state['baseDirectory'],
+                                       self.cache.get('data_dir', 'data'),
+                                       self.indices[ix]),
+                          allow_pickle=True).item()
+            return {'inputs': torch.from_numpy(rec['x']),
+                    'labels': torch.tensor(rec['y'], dtype=torch.long)}
+
+    class RefTabularTrainer(RefTrainer):
+        def _init_nn_model(self):
+            self.nn['net'] = FreeSurferMLP(in_features=16,
+                                           hidden_sizes=(32, 16),
+                                           num_class=2, dropout=0.0)
+
+        def iteration(self, batch):
+            dev = self.device.get('gpu', torch.device('cpu'))
+            inputs = batch['inputs'].to(dev).float()
+            labels = batch['labels'].to(dev).long()
+            out = self.nn['net'](inputs)
+            loss = torch.nn.functional.cross_entropy(out, labels)
+            pred = torch.argmax(out, 1)
+            avg = self.new_averages()
+            avg.add(loss.item(), len(inputs))
+            metrics = self.new_metrics()
+            metrics.add(pred, labels)
+            return {'loss': loss, 'averages': avg, 'metrics': metrics,
+                    'output': pred}
+
+    return RefTabularTrainer, RefTabularDataset
+
+
+def _run_reference(root):
+    _import_reference()
+    from coinstac_dinunet import COINNLocal as RefLocal
+    from coinstac_dinunet import COINNRemote as RefRemote
+    from coinstac_dinunet_amd.simulator import LoopbackCluster
+    from computations import make_site_data
+
+    trainer_cls, dataset_cls = _make_reference_classes()
+    cluster = LoopbackCluster(
+        root, n_sites=2,
+        site_data=lambda s: make_site_data(s.as_dict(), n_samples=20,
+                                           seed=int(s.clientId[-1])))
+    cluster.remote_cache['seed'] = 2024  # else each package uses its own
+    success, _ = cluster.run(
+        lambda cache, input, state: RefLocal(cache=cache, input=input,
+                                             state=state, mode='train',
+                                             **_KW),
+        lambda cache, input, state: RefRemote(cache=cache, input=input,
+                                              state=state),
+        trainer_cls, dataset_cls=dataset_cls, mp_pool=_FakePool(),
+        max_rounds=400)
+    return success, cluster
+
+
+def _run_ours(root):
+    from coinstac_dinunet_amd import COINNLocal, COINNRemote
+    from coinstac_dinunet_amd.config.keys import Mode
+    from coinstac_dinunet_amd.distrib.learner import COINNLearner
+    from coinstac_dinunet_amd.distrib.reducer import COINNReducer \
+        as COINNReducerD
+    from coinstac_dinunet_amd.simulator import LoopbackCluster
+    from computations import TabularDataset, TabularTrainer, make_site_data
+
+    class UnshuffledLearner(COINNLearner):
+        """The reference's padded train loader never shuffles (its
+        get_loader forces shuffle False before building the sampler,
+        data.py:163-171); ours shuffles seeded by default. Align batch
+        order for the bitwise comparison."""
+
+        def backward(self):
+            out = {}
+            self.trainer.nn[self.first_model].train()
+            self.trainer.optimizer[self.first_optim].zero_grad()
+            its = []
+            for _ in range(self.cache.get('local_iterations', 1)):
+                batch, nxt = self.trainer.data_handle.next_iter(shuffle=False)
+                it = self.trainer.iteration(batch)
+                it['loss'].backward()
+                its.append(it)
+                out.update(**nxt)
+            return self.trainer.reduce_iteration(its), out
+
+    cluster = LoopbackCluster(
+        root, n_sites=2,
+        site_data=lambda s: make_site_data(s.as_dict(), n_samples=20,
+                                           seed=int(s.clientId[-1])))
+    cluster.remote_cache['seed'] = 2024  # match the reference run's seed
+    success, _ = cluster.run(
+        lambda cache, input, state: COINNLocal(cache=cache, input=input,
+                                               state=state, mode=Mode.TRAIN,
+                                               agg_engine='dSGD_unshuffled',
+                                               **_KW),
+        lambda cache, input, state: COINNRemote(cache=cache, input=input,
+                                                state=state),
+        TabularTrainer, dataset_cls=TabularDataset,
+        learner_cls=UnshuffledLearner, reducer_cls=COINNReducerD,
+        max_rounds=400)
+    return success, cluster
+
+
+def test_reference_and_ours_agree_end_to_end(tmp_path):
+    ok_ref, ref_cluster = _run_reference(str(tmp_path / 'ref'))
+    ok_our, our_cluster = _run_ours(str(tmp_path / 'ours'))
+    assert ok_ref and ok_our
+    assert ref_cluster.rounds == our_cluster.rounds, \
+        'protocols took different numbers of rounds'
+
+    # reduced global test scores agree
+    from coinstac_dinunet_amd.config.keys import Key
+    rs = ref_cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
+    os_ = our_cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
+    assert len(rs) == len(os_) == 1
+
+    def flat(x):
+        return np.asarray(x['averages'] + [x['metrics']], dtype=object)
+
+    ra = np.asarray(rs[0]['averages'], dtype=float)
+    oa = np.asarray(os_[0]['averages'], dtype=float)
+    np.testing.assert_allclose(ra, oa, rtol=1e-5)
+    np.testing.assert_allclose(np.asarray(rs[0]['metrics'], dtype=float),
+                               np.asarray(os_[0]['metrics'], dtype=float),
+                               atol=1e-6)
+
+    # final site models agree parameter by parameter
+    for i in range(2):
+        rnet = ref_cluster.site_caches[i]['nn']['net']
+        onet = our_cluster.site_caches[i]['nn']['net']
+        for (rn, rp), (on, op_) in zip(rnet.named_parameters(),
+                                       onet.named_parameters()):
+            assert rn == on
+            torch.testing.assert_close(rp, op_, rtol=1e-6, atol=1e-7)
