@@ -193,3 +193,55 @@ def test_reference_and_ours_agree_end_to_end(tmp_path):
                                        onet.named_parameters()):
             assert rn == on
             torch.testing.assert_close(rp, op_, rtol=1e-6, atol=1e-7)
+
+
+def test_reference_and_ours_agree_kfold(tmp_path):
+    """Same differential under 3-fold cross-validation: fold queue order,
+    per-fold re-init and the fold score table must all line up."""
+    kw = dict(_KW)
+    kw.update(split_ratio=None, num_folds=3, epochs=1, patience=1)
+
+    _import_reference()
+    from coinstac_dinunet import COINNLocal as RefLocal
+    from coinstac_dinunet import COINNRemote as RefRemote
+    from coinstac_dinunet_amd import COINNLocal, COINNRemote
+    from coinstac_dinunet_amd.config.keys import Key, Mode
+    from coinstac_dinunet_amd.simulator import LoopbackCluster
+    from computations import TabularDataset, TabularTrainer, make_site_data
+
+    rtc, rdc = _make_reference_classes()
+
+    def build(which):
+        c = LoopbackCluster(
+            str(tmp_path / which), n_sites=2,
+            site_data=lambda s: make_site_data(s.as_dict(), n_samples=18,
+                                               seed=int(s.clientId[-1])))
+        c.remote_cache['seed'] = 7
+        return c
+
+    cr = build('ref')
+    ok_ref, _ = cr.run(
+        lambda cache, input, state: RefLocal(cache=cache, input=input,
+                                             state=state, mode='train', **kw),
+        lambda cache, input, state: RefRemote(cache=cache, input=input,
+                                              state=state),
+        rtc, dataset_cls=rdc, mp_pool=_FakePool(), max_rounds=700)
+
+    co = build('ours')
+    ok_our, _ = co.run(
+        lambda cache, input, state: COINNLocal(cache=cache, input=input,
+                                               state=state, mode=Mode.TRAIN,
+                                               **kw),
+        lambda cache, input, state: COINNRemote(cache=cache, input=input,
+                                                state=state),
+        TabularTrainer, dataset_cls=TabularDataset, max_rounds=700)
+
+    assert ok_ref and ok_our
+    rs = cr.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
+    os_ = co.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
+    assert len(rs) == len(os_) == 3  # one entry per fold, same order
+    for fold_ref, fold_our in zip(rs, os_):
+        ra = np.asarray(fold_ref['averages'], dtype=float)
+        oa = np.asarray(fold_our['averages'], dtype=float)
+        assert ra.shape == oa.shape
+        np.testing.assert_allclose(ra[1], oa[1])  # same sample counts
