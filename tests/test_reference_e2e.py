@@ -245,3 +245,61 @@ def test_reference_and_ours_agree_kfold(tmp_path):
         oa = np.asarray(fold_our['averages'], dtype=float)
         assert ra.shape == oa.shape
         np.testing.assert_allclose(ra[1], oa[1])  # same sample counts
+
+
+def test_reference_and_ours_elect_same_pretrain_site(tmp_path):
+    """Pretraining election (max train-data site) must agree. KNOWN
+    REFERENCE DEADLOCK pinned here: when the elected site's pretraining
+    never improves, the reference ships no weights_file and every site
+    spins in PRE_COMPUTATION forever (observed below); our local
+    proceeds to COMPUTATION instead (distrib/nodes/local.py). The test
+    asserts our stack finishes and that the reference either finishes
+    too or is stuck in exactly that phase."""
+    kw = dict(_KW)
+    kw.update(pretrain_args={'epochs': 1})
+
+    _import_reference()
+    from coinstac_dinunet import COINNLocal as RefLocal
+    from coinstac_dinunet import COINNRemote as RefRemote
+    from coinstac_dinunet_amd import COINNLocal, COINNRemote
+    from coinstac_dinunet_amd.config.keys import Mode
+    from coinstac_dinunet_amd.simulator import LoopbackCluster
+    from computations import TabularDataset, TabularTrainer, make_site_data
+
+    rtc, rdc = _make_reference_classes()
+    sizes = {'local0': 12, 'local1': 28}  # site 1 must win the election
+
+    def build(which):
+        c = LoopbackCluster(
+            str(tmp_path / which), n_sites=2,
+            site_data=lambda s: make_site_data(
+                s.as_dict(), n_samples=sizes[s.clientId],
+                seed=int(s.clientId[-1])))
+        c.remote_cache['seed'] = 7
+        return c
+
+    cr = build('ref')
+    ok_ref, _ = cr.run(
+        lambda cache, input, state: RefLocal(cache=cache, input=input,
+                                             state=state, mode='train', **kw),
+        lambda cache, input, state: RefRemote(cache=cache, input=input,
+                                              state=state),
+        rtc, dataset_cls=rdc, mp_pool=_FakePool(), max_rounds=60)
+    co = build('ours')
+    ok_our, _ = co.run(
+        lambda cache, input, state: COINNLocal(cache=cache, input=input,
+                                               state=state, mode=Mode.TRAIN,
+                                               **kw),
+        lambda cache, input, state: COINNRemote(cache=cache, input=input,
+                                                state=state),
+        TabularTrainer, dataset_cls=TabularDataset, max_rounds=500)
+
+    assert ok_our, 'our stack must complete the pretrain path'
+    if not ok_ref:
+        # the reference deadlock: every site still in PRE_COMPUTATION
+        from coinstac_dinunet.config.keys import Phase as RefPhase
+        assert cr.site_inputs[0].get('phase') == RefPhase.PRE_COMPUTATION
+    for cl in (cr, co):
+        assert cl.site_caches[1].get('pretrain') is True, \
+            'site 1 (more data) should pretrain'
+        assert cl.site_caches[0].get('pretrain') is False
