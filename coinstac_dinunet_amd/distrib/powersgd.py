@@ -141,13 +141,16 @@ class PowerSGDLearner(COINNLearner):
         dev = self.device
         error = self.cache.setdefault('powerSGD_error', {})
         Ms, Ps, r1 = {}, {}, []
-        gen = _torch.Generator(device='cpu').manual_seed(
-            int(self.seed) + int(self.cache['powerSGD_iter']))
         ship = []
         for i, p in mats:
             g = self._matrix_view(p.grad.detach().float())
             M = g + error[i] if (self.use_error_feedback and i in error) \
                 else g
+            # generator re-seeded PER PARAM — bitwise parity with the
+            # reference's Q init (powersgd/__init__.py:113-114 re-seeds the
+            # global rng inside the param loop)
+            gen = _torch.Generator(device='cpu').manual_seed(
+                int(self.seed) + int(self.cache['powerSGD_iter']))
             Q = _torch.randn(M.shape[1], self.rank, generator=gen).to(dev)
             orthogonalize(Q)
             P = M @ Q

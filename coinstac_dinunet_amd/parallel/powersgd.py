@@ -62,8 +62,6 @@ class RcclPowerSGDLearner(COINNLearner):
 
     def _compress_round(self):
         error = self.cache.setdefault('powerSGD_error', {})
-        gen = torch.Generator(device='cpu').manual_seed(
-            int(self.seed) + int(self.cache['powerSGD_iter']))
         mats, rank1 = [], []
         for i, p in enumerate(self._params()):
             if p.grad is None:
@@ -74,6 +72,10 @@ class RcclPowerSGDLearner(COINNLearner):
         for i, p in mats:
             g = p.grad.detach().float().reshape(p.shape[0], -1)
             M = g + error[i] if (self.use_error_feedback and i in error) else g
+            # per-param re-seed: numerically identical Q to the loopback
+            # engine and the reference (powersgd/__init__.py:113-114)
+            gen = torch.Generator(device='cpu').manual_seed(
+                int(self.seed) + int(self.cache['powerSGD_iter']))
             Q0 = torch.randn(M.shape[1], self.rank_approx,
                              generator=gen).to(M.device)
             orthogonalize(Q0)

