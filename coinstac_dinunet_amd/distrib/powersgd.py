@@ -60,6 +60,7 @@ class PowerSGDLearner(COINNLearner):
         self.rank = self.cache.get('matrix_approximation_rank', 1)
         self.start_iter = self.cache.get('start_powerSGD_iter', 10)
         self.use_error_feedback = self.cache.get('use_error_feedback', True)
+        self.warm_start = self.cache.setdefault('warm_start', True)
         self.seed = self.cache.get('seed', 0)
         self.cache.setdefault('powerSGD_iter', 0)
 
@@ -107,6 +108,7 @@ class PowerSGDLearner(COINNLearner):
         Ps = self.cache['powerSGD_Ps']
         error = self.cache.setdefault('powerSGD_error', {})
         Ms = self.cache['powerSGD_Ms']
+        warm_qs = self.cache.setdefault('powerSGD_Qs', {})
         for (i, p), q in zip(mats, aggs):
             Q = _torch.tensor(_np.asarray(q), dtype=_torch.float32,
                               device=dev)
@@ -114,6 +116,8 @@ class PowerSGDLearner(COINNLearner):
             recon = P @ Q.t()
             if self.use_error_feedback:
                 error[i] = Ms[i] - recon
+            if self.warm_start:  # reuse next round (reference powersgd:110)
+                warm_qs[i] = Q
             p.grad = recon.reshape(p.shape).to(p.dtype)
         self.trainer.optimizer[self.first_optim].step()
         self.cache['powerSGD_iter'] += 1
@@ -146,12 +150,18 @@ class PowerSGDLearner(COINNLearner):
             g = self._matrix_view(p.grad.detach().float())
             M = g + error[i] if (self.use_error_feedback and i in error) \
                 else g
-            # generator re-seeded PER PARAM — bitwise parity with the
-            # reference's Q init (powersgd/__init__.py:113-114 re-seeds the
-            # global rng inside the param loop)
-            gen = _torch.Generator(device='cpu').manual_seed(
-                int(self.seed) + int(self.cache['powerSGD_iter']))
-            Q = _torch.randn(M.shape[1], self.rank, generator=gen).to(dev)
+            warm_qs = self.cache.get('powerSGD_Qs', {})
+            if self.warm_start and i in warm_qs:
+                # warm start: re-orthogonalize the last averaged Q
+                # (reference powersgd/__init__.py:110, 120-121)
+                Q = warm_qs[i].clone()
+            else:
+                # generator re-seeded PER PARAM — bitwise parity with the
+                # reference's Q init (powersgd/__init__.py:113-114 re-seeds
+                # the global rng inside the param loop)
+                gen = _torch.Generator(device='cpu').manual_seed(
+                    int(self.seed) + int(self.cache['powerSGD_iter']))
+                Q = _torch.randn(M.shape[1], self.rank, generator=gen).to(dev)
             orthogonalize(Q)
             P = M @ Q
             Ms[i], Ps[i] = M, P
