@@ -379,3 +379,56 @@ def test_reference_and_ours_agree_powersgd(tmp_path):
         onet = co.site_caches[i]['nn']['net']
         for rp, op_ in zip(rnet.parameters(), onet.parameters()):
             torch.testing.assert_close(rp, op_, rtol=1e-5, atol=1e-6)
+
+
+def test_reference_and_ours_agree_rankdad_protocol(tmp_path):
+    """rankDAD differential (structural): same phase trajectory and round
+    count. Factor VALUES legitimately differ — our power_iteration_BC
+    fixes the reference's deflation (garbage components past the true
+    rank; see distrib/rankdad.py docstring) — so this pins the protocol,
+    not the bits."""
+    kw = dict(_KW)
+    kw.update(agg_engine='rankDAD', dad_reduction_rank=3, epochs=1,
+              patience=1)
+
+    _import_reference()
+    from coinstac_dinunet import COINNLocal as RefLocal
+    from coinstac_dinunet import COINNRemote as RefRemote
+    from coinstac_dinunet_amd import COINNLocal, COINNRemote
+    from coinstac_dinunet_amd.config.keys import Key, Mode
+    from coinstac_dinunet_amd.simulator import LoopbackCluster
+    from computations import TabularDataset, TabularTrainer, make_site_data
+
+    rtc, rdc = _make_reference_classes()
+
+    def build(which):
+        c = LoopbackCluster(
+            str(tmp_path / which), n_sites=2,
+            site_data=lambda s: make_site_data(s.as_dict(), n_samples=20,
+                                               seed=int(s.clientId[-1])))
+        c.remote_cache['seed'] = 5
+        return c
+
+    cr = build('ref')
+    ok_ref, _ = cr.run(
+        lambda cache, input, state: RefLocal(cache=cache, input=input,
+                                             state=state, mode='train', **kw),
+        lambda cache, input, state: RefRemote(cache=cache, input=input,
+                                              state=state),
+        rtc, dataset_cls=rdc, mp_pool=_FakePool(), max_rounds=500)
+    co = build('ours')
+    ok_our, _ = co.run(
+        lambda cache, input, state: COINNLocal(cache=cache, input=input,
+                                               state=state, mode=Mode.TRAIN,
+                                               **kw),
+        lambda cache, input, state: COINNRemote(cache=cache, input=input,
+                                                state=state),
+        TabularTrainer, dataset_cls=TabularDataset, max_rounds=500)
+
+    assert ok_ref and ok_our
+    assert cr.rounds == co.rounds
+    rs = cr.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
+    os_ = co.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
+    assert len(rs) == len(os_) == 1
+    # same evaluated sample counts through the same protocol
+    assert rs[0]['averages'][1] == os_[0]['averages'][1]
