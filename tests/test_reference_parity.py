@@ -207,3 +207,38 @@ def test_initialize_weights_bitwise(ref):
     torch.manual_seed(99), oi(mb)
     for pa, pb in zip(ma.parameters(), mb.parameters()):
         assert torch.equal(pa, pb)
+
+
+def test_config_precedence_matches(ref, tmp_path):
+    """Three-source hyperparameter resolution (platform input >
+    {task}_args > {engine}_args > {task}_data_conf > ctor defaults,
+    local.py:92-118) must cache identical values in both stacks."""
+    from coinstac_dinunet import COINNLocal as RefLocal
+    from coinstac_dinunet_amd import COINNLocal as OurLocal
+
+    spec = {
+        'task_id': 'tsk', 'mode': 'train', 'agg_engine': 'dSGD',
+        'batch_size': 16,                       # platform input
+        'tsk_args': {'epochs': 7, 'learning_rate': 0.005},
+        'dSGD_args': {'local_iterations': 3},
+        'tsk_data_conf': {'batch_size': 99,      # loses: already in input?
+                          'data_dir': 'vols', 'num_folds': 4},
+    }
+    kw = dict(task_id='tsk', mode='train', batch_size=4, epochs=2,
+              local_iterations=1, split_ratio=(0.8, 0.1, 0.1),
+              data_dir='data', num_class=2, verbose=False)
+
+    caches = {}
+    for name, cls in (('ref', RefLocal), ('ours', OurLocal)):
+        cache = {}
+        cls(cache=cache, input=dict(spec), state={
+            'clientId': 'local0', 'baseDirectory': str(tmp_path / name),
+            'transferDirectory': str(tmp_path / name),
+            'outputDirectory': str(tmp_path / name),
+            'cacheDirectory': str(tmp_path / name)}, **kw)
+        caches[name] = cache
+
+    for k in ('batch_size', 'epochs', 'learning_rate', 'local_iterations',
+              'data_dir', 'num_folds', 'task_id', 'mode', 'agg_engine'):
+        assert caches['ref'].get(k) == caches['ours'].get(k), \
+            (k, caches['ref'].get(k), caches['ours'].get(k))
