@@ -173,19 +173,17 @@ class DADParallel(_torch.nn.Module):
             def fn(mod, inputs, output):
                 if inputs and inputs[0] is not None:
                     self._activations[key] = inputs[0]
-            return fn
-
-        def bk(key):
-            def fn(mod, grad_in, grad_out):
-                for g in grad_out:
-                    if g is not None:
-                        self._grads[key] = g
-                        break
+                # grad of the OUTPUT tensor == the reference's first
+                # non-None grad_out (spi.py:152-163) for these
+                # single-output leaves; a tensor hook avoids torch's
+                # full-backward-hook warning when inputs carry no grad
+                if isinstance(output, _torch.Tensor) and output.requires_grad:
+                    output.register_hook(
+                        lambda g, k=key: self._grads.__setitem__(k, g))
             return fn
 
         for name, mod in self._leaves():
             self._fw_handles.append(mod.register_forward_hook(fw(name)))
-            self._bk_handles.append(mod.register_full_backward_hook(bk(name)))
 
     def _unhook(self):
         for h in self._fw_handles + self._bk_handles:
