@@ -13,6 +13,12 @@ aggregator. Per round:
 
 The hot path therefore exchanges ONLY the fused gradient bucket per round;
 control costs one object gather+broadcast of a few KB.
+
+Scope: one node (the BASELINE configs — 8 GPU-sites over xGMI). The
+collectives themselves are multi-node-clean (torchrun + MASTER_ADDR),
+but the COLD path (pretrained weights.tar, results zip) relays over
+`root`, which must then be a shared filesystem; with a node-local root,
+keep the group on one node.
 """
 import os
 import shutil
