@@ -117,3 +117,55 @@ def test_averages_weighted_mean_property(vals):
     expect = sum(v * n for v, n in vals) / sum(n for _, n in vals)
     got = np.asarray(a.average, dtype=float).reshape(-1)[0]
     assert abs(got - round(expect, 5)) < 1e-4
+
+
+@settings(max_examples=30, deadline=None)
+@given(shapes=st.lists(
+    st.tuples(st.integers(1, 40), st.integers(0, 12)),  # (rows, cols|0=1D)
+    min_size=1, max_size=10),
+    bucket_kb=st.integers(1, 64))
+def test_flat_grad_buffer_partitions_any_model(shapes, bucket_kb):
+    """For ANY parameter geometry: views alias grads 1:1, buckets
+    partition the arena exactly, and a filled arena round-trips."""
+    import torch
+    from coinstac_dinunet_amd.parallel.engine import FlatGradBuffer
+    params = [torch.nn.Parameter(torch.randn(r, c) if c else torch.randn(r))
+              for r, c in shapes]
+    buf = FlatGradBuffer(params, bucket_bytes=bucket_kb * 1024, world_size=1)
+    total = sum(p.numel() for p in params)
+    assert buf.flat.numel() == total
+    # buckets partition [0, total)
+    spans = sorted((s, s + n) for s, n, _ in buf.buckets)
+    assert spans[0][0] == 0 and spans[-1][1] == total
+    for (a0, a1), (b0, b1) in zip(spans, spans[1:]):
+        assert a1 == b0
+    # every param's grad is a view into the arena
+    for p in params:
+        assert p.grad is not None
+        assert p.grad.data_ptr() >= buf.flat.data_ptr()
+        p.grad.fill_(3.0)
+    assert torch.all(buf.flat == 3.0)
+    buf.zero_()
+    assert torch.all(buf.flat == 0.0)
+    buf.remove_hooks()
+
+
+@settings(max_examples=25, deadline=None)
+@given(n=st.integers(4, 40), m=st.integers(4, 40),
+       true_rank=st.integers(1, 4), seed=st.integers(0, 9999))
+def test_power_iteration_recovers_low_rank(n, m, true_rank, seed):
+    """On an exactly rank-r product, rank-r extraction must reconstruct
+    it (to float tolerance) — the guarantee rankDAD's gradient fidelity
+    rests on."""
+    import torch
+    from coinstac_dinunet_amd.distrib.rankdad import power_iteration_BC
+    g = torch.Generator().manual_seed(seed)
+    k = true_rank
+    B = torch.randn(n, k, generator=g) @ torch.randn(k, 17, generator=g)
+    C = torch.randn(m, k, generator=g) @ torch.randn(k, 17, generator=g)
+    gen = torch.Generator().manual_seed(seed + 1)
+    bf, cf = power_iteration_BC(B, C, rank=k + 2, numiterations=30,
+                                tol=1e-6, generator=gen)
+    G = B @ C.t()
+    err = torch.norm(bf @ cf.t() - G) / max(torch.norm(G), 1e-8)
+    assert err < 5e-2, float(err)
