@@ -114,3 +114,43 @@ def test_fedavg_custom_engine(tmp_path):
     # after the last averaging round sites may have trained locally again
     # (weights equal only right after step); just assert the run finished
     assert cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
+
+
+def test_save_predictions_hook_overrides_accumulation(tmp_path):
+    """evaluation(save_pred=True): a save_predictions override may return
+    {'averages','metrics'} that replace the iteration's own."""
+    import torch
+    from coinstac_dinunet_amd.metrics import COINNAverages, Prf1a
+    from computations import TabularTrainer, TabularDataset, make_site_data
+    from coinstac_dinunet_amd.data import COINNDataHandle
+
+    state = {'clientId': 'local0', 'baseDirectory': str(tmp_path),
+             'outputDirectory': str(tmp_path),
+             'transferDirectory': str(tmp_path)}
+    make_site_data(state, n_samples=8)
+    cache = {'batch_size': 4, 'num_class': 2, 'data_dir': 'data',
+             'verbose': False}
+
+    class HookTrainer(TabularTrainer):
+        def __init__(self, **kw):
+            super().__init__(**kw)
+            self.hook_calls = 0
+
+        def save_predictions(self, dataset, its):
+            self.hook_calls += 1
+            avg = COINNAverages(num_averages=1)
+            avg.add(42.0, 1)
+            m = Prf1a()
+            m.add(torch.tensor([1]), torch.tensor([1]))
+            return {'averages': avg, 'metrics': m}
+
+    handle = COINNDataHandle(cache=cache, state=state)
+    trainer = HookTrainer(data_handle=handle)
+    trainer.init_nn(init_model=True, init_optim=True, set_devices=True)
+    ds = TabularDataset(mode='test', cache=cache, state=state)
+    ds.add(files=sorted(__import__('os').listdir(tmp_path / 'data')))
+    avg, metrics = trainer.evaluation(mode='test', dataset_list=[ds],
+                                      save_pred=True)
+    assert trainer.hook_calls == 2  # 8 samples / batch 4
+    assert avg.get()[0] == 42.0     # hook's values, not the loss
+    assert metrics.tp == 2
