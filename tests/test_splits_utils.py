@@ -96,3 +96,37 @@ def test_lazy_debug_monotone():
     hits = [x for x in range(1, 200) if lazy_debug(x)]
     assert len(hits) > 5
     assert len(hits) < 150
+
+
+def test_init_k_folds_existing_dir_wins(tmp_path):
+    """Pre-existing split dir in baseDirectory overrides everything."""
+    import json
+    from coinstac_dinunet_amd.data import datautils
+    base = tmp_path / 'base'
+    out = tmp_path / 'out'
+    (base / 'splits').mkdir(parents=True)
+    custom = {'train': ['a'], 'validation': ['b'], 'test': ['c']}
+    (base / 'splits' / 'SPLIT_custom.json').write_text(json.dumps(custom))
+    cache = {'task_id': 'tsk', 'num_folds': 4,          # would lose
+             'split_ratio': (0.8, 0.1, 0.1)}            # would lose
+    state = {'baseDirectory': str(base), 'outputDirectory': str(out)}
+    datautils.init_k_folds(['a', 'b', 'c'], cache, state)
+    assert cache['splits'] == {'0': 'SPLIT_custom.json'}
+    got = json.load(open(os.path.join(cache['split_dir'],
+                                      'SPLIT_custom.json')))
+    assert got == custom
+
+
+def test_init_k_folds_split_files_beat_num_folds(tmp_path):
+    import json
+    from coinstac_dinunet_amd.data import datautils
+    base = tmp_path / 'base'
+    out = tmp_path / 'out'
+    base.mkdir()
+    custom = {'train': ['x'], 'validation': ['y'], 'test': ['z']}
+    (base / 'given.json').write_text(json.dumps(custom))
+    cache = {'task_id': 'tsk', 'split_files': ['given.json'],
+             'num_folds': 3}
+    state = {'baseDirectory': str(base), 'outputDirectory': str(out)}
+    datautils.init_k_folds(['x', 'y', 'z'], cache, state)
+    assert cache['splits'] == {'0': 'given.json'}
