@@ -1,0 +1,67 @@
+"""3D U-Net for volumetric segmentation — the reference ecosystem's
+segmentation use case (dice_loss_binary in metrics/loss.py, the U-Net
+patch utilities in vision/imageutils.py, safe_concat in
+utils/tensorutils.py all exist to serve it; the model itself lives in
+the external implementation repos).
+
+MI355X mapping: every conv is a 3x3x3/pad-1 stride-1 or stride-2
+OpsConv3d, so the whole encoder/decoder runs on the spatial-slab MFMA
+kernels; BN+ReLU are the fused bnorm kernels. Downsampling is stride-2
+convolution (not pooling) and upsampling is nearest-neighbor interp +
+conv — both stay inside the supported kernel family, no transposed-conv
+kernel needed. Skip connections go through safe_concat (center-crop) so
+odd input sizes work.
+"""
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..ops.bnorm import OpsBatchNorm3d
+from ..ops.conv import OpsConv3d
+from ..utils.tensorutils import safe_concat
+
+
+class _Block(nn.Module):
+    """conv(3x3x3) -> fused BN+ReLU, twice; optional stride-2 entry."""
+
+    def __init__(self, cin, cout, stride=1):
+        super().__init__()
+        self.c1 = OpsConv3d(cin, cout, 3, stride=stride, padding=1,
+                            bias=False)
+        self.b1 = OpsBatchNorm3d(cout, relu=True)
+        self.c2 = OpsConv3d(cout, cout, 3, stride=1, padding=1, bias=False)
+        self.b2 = OpsBatchNorm3d(cout, relu=True)
+
+    def forward(self, x):
+        return self.b2(self.c2(self.b1(self.c1(x))))
+
+
+class UNet3D(nn.Module):
+    """Encoder-decoder with skip connections; `widths` sets depth."""
+
+    def __init__(self, in_channels=1, num_class=2, widths=(16, 32, 64)):
+        super().__init__()
+        self.stem = _Block(in_channels, widths[0])
+        self.down = nn.ModuleList(
+            _Block(widths[i], widths[i + 1], stride=2)
+            for i in range(len(widths) - 1))
+        self.up_conv = nn.ModuleList(
+            OpsConv3d(widths[i + 1], widths[i], 3, padding=1, bias=False)
+            for i in reversed(range(len(widths) - 1)))
+        self.up_block = nn.ModuleList(
+            _Block(2 * widths[i], widths[i])
+            for i in reversed(range(len(widths) - 1)))
+        self.head = nn.Conv3d(widths[0], num_class, 1)
+
+    def forward(self, x):
+        skips = []
+        x = self.stem(x)
+        for d in self.down:
+            skips.append(x)
+            x = d(x)
+        for conv, block in zip(self.up_conv, self.up_block):
+            skip = skips.pop()
+            x = F.interpolate(x, size=skip.shape[2:], mode='nearest')
+            x = conv(x)
+            x = block(safe_concat(skip, x))
+        return self.head(x)
