@@ -55,3 +55,78 @@ def test_unet_gpu_matches_cpu_reference():
     # loose: bf16 conv error accumulates through 5 conv layers; a layout
     # bug would still be O(1) wrong
     torch.testing.assert_close(out, ref, rtol=1e-1, atol=1e-1)
+
+
+def test_unet_segmentation_full_protocol(tmp_path):
+    """2-site lock-step dSGD SEGMENTATION run: UNet3D + dice loss +
+    voxelwise Prf1a through the whole phase machine."""
+    import os
+    import numpy as np
+    from coinstac_dinunet_amd import (COINNDataset, COINNLocal, COINNRemote,
+                                      COINNTrainer)
+    from coinstac_dinunet_amd.config.keys import Key, Mode
+    from coinstac_dinunet_amd.simulator import LoopbackCluster
+
+    def make_volumes(state, n=8, seed=0):
+        rng = np.random.RandomState(seed)
+        d = os.path.join(state.baseDirectory, 'data')
+        os.makedirs(d, exist_ok=True)
+        g = np.stack(np.meshgrid(*([np.arange(8.0)] * 3), indexing='ij'))
+        for i in range(n):
+            c = rng.uniform(2.5, 5.5, size=3)
+            mask = (((g - c[:, None, None, None]) ** 2).sum(0) < 4.0)
+            x = mask.astype(np.float32) + \
+                0.3 * rng.randn(8, 8, 8).astype(np.float32)
+            np.save(os.path.join(d, f'v{i:02d}.npy'),
+                    {'x': x, 'y': mask.astype(np.int64)}, allow_pickle=True)
+
+    class SegDataset(COINNDataset):
+        def load_index(self, file):
+            self.indices.append(file)
+
+        def __getitem__(self, ix):
+            import torch as T
+            rec = np.load(os.path.join(self.state['baseDirectory'],
+                                       self.cache['data_dir'],
+                                       self.indices[ix]),
+                          allow_pickle=True).item()
+            return {'inputs': T.from_numpy(rec['x']).unsqueeze(0),
+                    'labels': T.from_numpy(rec['y'])}
+
+    class SegTrainer(COINNTrainer):
+        def _init_nn_model(self):
+            from coinstac_dinunet_amd.models import UNet3D
+            self.nn['net'] = UNet3D(in_channels=1, num_class=1,
+                                    widths=(2, 4))
+
+        def iteration(self, batch):
+            import torch as T
+            from coinstac_dinunet_amd.metrics.loss import dice_loss_binary
+            dev = self.device['gpu']
+            x = batch['inputs'].to(dev).float()
+            y = batch['labels'].to(dev).float()
+            prob = T.sigmoid(self.nn['net'](x)).squeeze(1)
+            loss = dice_loss_binary(prob, y)
+            avg = self.new_averages()
+            avg.add(loss.item(), len(x))
+            metrics = self.new_metrics()
+            metrics.add((prob.detach() > 0.5).long().reshape(-1),
+                        y.long().reshape(-1))
+            return {'loss': loss, 'averages': avg, 'metrics': metrics,
+                    'output': prob}
+
+    cluster = LoopbackCluster(
+        str(tmp_path), n_sites=2,
+        site_data=lambda s: make_volumes(s, seed=int(s.clientId[-1])))
+    kw = dict(task_id='seg', mode=Mode.TRAIN, batch_size=2, epochs=1,
+              validation_epochs=1, local_iterations=1,
+              split_ratio=(0.5, 0.25, 0.25), data_dir='data', num_class=2,
+              seed_all=True, patience=1, verbose=False)
+    success, _ = cluster.run(
+        lambda cache, input, state: COINNLocal(cache=cache, input=input,
+                                               state=state, **kw),
+        lambda cache, input, state: COINNRemote(cache=cache, input=input,
+                                                state=state),
+        SegTrainer, dataset_cls=SegDataset, max_rounds=300)
+    assert success
+    assert cluster.remote_cache[Key.GLOBAL_TEST_SERIALIZABLE]
