@@ -26,7 +26,11 @@ class SiteRunner:
         if _os.path.exists(spec_path):
             with open(spec_path) as f:
                 spec = _json.load(f)
-            # inputspec.json layout: {key: {"value": ...}, ...}
+            # platform layout: a LIST with one {key: {"value": ...}} dict
+            # per site (reference site_runner.py:13-15); a bare dict is
+            # accepted too for hand-written specs
+            if isinstance(spec, list):
+                spec = spec[site_index]
             for k, v in spec.items():
                 self.inputspec[k] = v.get('value') if isinstance(v, dict) else v
 

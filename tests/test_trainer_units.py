@@ -154,3 +154,22 @@ def test_save_predictions_hook_overrides_accumulation(tmp_path):
     assert trainer.hook_calls == 2  # 8 samples / batch 4
     assert avg.get()[0] == 42.0     # hook's values, not the loss
     assert metrics.tp == 2
+
+
+def test_site_runner_platform_inputspec_list(tmp_path):
+    """The platform writes inputspec.json as a LIST (one entry per site,
+    values wrapped in {'value': ...}); SiteRunner must consume it."""
+    import json
+    from coinstac_dinunet_amd.site_runner import SiteRunner
+    from computations import TabularDataset, TabularTrainer, make_site_data
+
+    spec = [{'task_id': {'value': 'tab'}, 'batch_size': {'value': 4},
+             'epochs': {'value': 1}, 'num_class': {'value': 2},
+             'data_dir': {'value': 'data'},
+             'split_ratio': {'value': [0.6, 0.2, 0.2]}}]
+    (tmp_path / 'inputspec.json').write_text(json.dumps(spec))
+    runner = SiteRunner(task_id='tab', data_path=str(tmp_path),
+                        mode='train', verbose=False)
+    make_site_data(runner.state, n_samples=12)
+    runner.run(TabularTrainer, dataset_cls=TabularDataset)
+    assert runner.cache['batch_size'] == 4  # came from the spec list
