@@ -12,10 +12,13 @@ import torch.nn.functional as F
 
 from . import native_available, require_native
 
-# Opt-in (unvalidated this round): route Cin<16 stride-1 forwards through
-# the CTILE=1 spatial instances instead of the igemm fallback. Flip the
-# default once measured on hardware.
+# Opt-in: route Cin<16 stride-1 forwards through the CTILE=1 spatial
+# instances instead of the igemm fallback (hardware-validated
+# numerically; default off until measured).
 _SPATIAL_CI1 = os.environ.get('COINN_SPATIAL_CI1', '0') == '1'
+# Opt-in (compiled, unvalidated): double-buffered CTILE=16 stride-1
+# forward instances — stage slab t+1 during slab t's MFMA k-steps.
+_SPATIAL_DB = os.environ.get('COINN_SPATIAL_DB', '0') == '1'
 
 
 class _Conv3dFn(torch.autograd.Function):
@@ -31,8 +34,13 @@ class _Conv3dFn(torch.autograd.Function):
         min_chunk = 64
         ci_ok = xb.size(1) >= 16 or (_SPATIAL_CI1 and stride == 1)
         if (ow % 8 == 0 and ci_ok and oh * ow >= min_chunk):
-            out = C.conv3d_fwd_spatial(xb, wb, stride,
-                                       1 if xb.size(1) < 16 else 0)
+            if xb.size(1) < 16:
+                ctile_opt = 1
+            elif _SPATIAL_DB and stride == 1:
+                ctile_opt = 16
+            else:
+                ctile_opt = 0
+            out = C.conv3d_fwd_spatial(xb, wb, stride, ctile_opt)
         else:
             out = C.conv3d_fwd(xb, wb, stride)
         if bias is not None:

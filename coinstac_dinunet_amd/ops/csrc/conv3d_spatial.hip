@@ -185,6 +185,168 @@ __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Double-buffered variant (experimental; routed only via ctile_opt=16):
+// stage slab kt+1 into the spare LDS buffer while the 14 k-steps of slab
+// kt run, one barrier per channel tile instead of two. CTILE=16 so two
+// buffers fit LDS. Same math, same WB layout (KT_PAD blocks at ctile 16).
+// ---------------------------------------------------------------------------
+template <int OWT, int STRIDE, int CTILE,
+          int CHUNK = (STRIDE == 1 ? 256 : 128)>
+__global__ __launch_bounds__(256) void conv3d_spatial_db_kernel(
+    const __bf16* __restrict__ in, const __bf16* __restrict__ wb,
+    __bf16* __restrict__ out, SpDims sd, int64_t nchunks) {
+  constexpr int OHT = CHUNK / OWT;
+  constexpr int IW = STRIDE * OWT;
+  constexpr int W2 = IW + (STRIDE == 1 ? 4 : 2);
+  constexpr int H2 = STRIDE * (OHT - 1) + 3;
+  constexpr int MPW = CHUNK / 64;
+  static_assert(MPW >= 1, "chunk too small");
+  constexpr int KT_PAD = ((CTILE * 27 + 31) / 32) * 32;
+  __shared__ __bf16 sX[2][CTILE][3][H2][W2];
+  __shared__ unsigned short sKtab[KT_PAD + 8];
+
+  const int ncol0 = blockIdx.y * 32;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int row = lane & 15, kg = lane >> 4;
+
+  const int wtiles = (sd.TW + OWT - 1) / OWT;
+  const int htiles = (sd.TH + OHT - 1) / OHT;
+
+  int64_t t = blockIdx.x;
+  const int wt = (int)(t % wtiles);
+  t /= wtiles;
+  const int ht = (int)(t % htiles);
+  t /= htiles;
+  const int td = (int)(t % sd.TD);
+  const int n = (int)(t / sd.TD);
+  const int oh0 = ht * OHT, ow0 = wt * OWT;
+
+  for (int k = tid; k < KT_PAD; k += 256) {
+    unsigned short off = 0;
+    if (k < CTILE * 27) {
+      const int cl = k / 27;
+      const int r27 = k - cl * 27;
+      const int a = r27 / 9, b = (r27 / 3) % 3, c = r27 % 3;
+      off = (unsigned short)(((cl * 3 + a) * H2 + b) * W2 + c);
+    }
+    sKtab[k] = off;
+  }
+
+  f32x4 acc[MPW][2];
+#pragma unroll
+  for (int i = 0; i < MPW; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int64_t HW = (int64_t)sd.H * sd.W;
+  const int64_t in_n = (int64_t)n * sd.KCH * sd.D * HW;
+  const int kts = (sd.KCH + CTILE - 1) / CTILE;
+
+  auto stage = [&](int kt, int buf) {
+    constexpr int NROWS = CTILE * 3 * H2;
+    for (int r = tid; r < NROWS; r += 256) {
+      const int hrow = r % H2;
+      const int a = (r / H2) % 3;
+      const int c = r / (3 * H2);
+      const int id = STRIDE * td - 1 + a;
+      const int ih = STRIDE * oh0 - 1 + hrow;
+      const int ch = kt * CTILE + c;
+      __bf16* dst = &sX[buf][c][a][hrow][0];
+      const bool row_ok = (unsigned)id < (unsigned)sd.D &&
+                          (unsigned)ih < (unsigned)sd.H && ch < sd.KCH;
+      if (!row_ok) {
+#pragma unroll
+        for (int col = 0; col < W2; ++col) dst[col] = (__bf16)0.f;
+        continue;
+      }
+      const __bf16* src = in + in_n + ((int64_t)ch * sd.D + id) * HW +
+                          (int64_t)ih * sd.W;
+      const int iw0 = STRIDE * ow0;
+      dst[0] = (iw0 > 0) ? src[iw0 - 1] : (__bf16)0.f;
+#pragma unroll
+      for (int v = 0; v < IW / 8; ++v) {
+        bf16x8 vec = *reinterpret_cast<const bf16x8*>(src + iw0 + v * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) dst[1 + v * 8 + j] = vec[j];
+      }
+#pragma unroll
+      for (int e = 0; e < W2 - IW - 1; ++e) {
+        const int iw = iw0 + IW + e;
+        dst[1 + IW + e] = (iw < sd.W) ? src[iw] : (__bf16)0.f;
+      }
+    }
+  };
+
+  auto compute = [&](int kt, int buf) {
+    const int kbase_g = kt * KT_PAD;
+    const __bf16* slab = &sX[buf][0][0][0][0];
+#pragma unroll 1
+    for (int ks = 0; ks < KT_PAD / 32; ++ks) {
+      bf16x8 afrag[MPW];
+      {
+        const int kb = ks * 32 + kg * 8;
+        const u16x8 kt8 = *reinterpret_cast<const u16x8*>(&sKtab[kb]);
+#pragma unroll
+        for (int i = 0; i < MPW; ++i) {
+          const int m = (wave * MPW + i) * 16 + row;
+          const int base = (STRIDE * (m / OWT)) * W2 + STRIDE * (m % OWT);
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            afrag[i][j] = slab[base + kt8[j]];
+        }
+      }
+      bf16x8 bfrag[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        const int col = ncol0 + i * 16 + row;
+        const int64_t off =
+            (int64_t)col * sd.Kpad + kbase_g + ks * 32 + kg * 8;
+        bfrag[i] = (col < sd.NCOL)
+                       ? *reinterpret_cast<const bf16x8*>(wb + off)
+                       : bf16x8{};
+      }
+#pragma unroll
+      for (int i = 0; i < MPW; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+  for (int kt = 0; kt < kts; ++kt) {
+    if (kt + 1 < kts) stage(kt + 1, (kt + 1) & 1);
+    compute(kt, kt & 1);
+    __syncthreads();  // next buffer staged AND this buffer's reads done
+  }
+
+  const int64_t THW = (int64_t)sd.TH * sd.TW;
+  const int64_t out_n = (int64_t)n * sd.NCOL * sd.TD * THW;
+  const int ccol = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < MPW; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int col = ncol0 + j * 16 + ccol;
+      if (col >= sd.NCOL) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = (wave * MPW + i) * 16 + crow0 + r;
+        const int oh = oh0 + m / OWT;
+        const int ow = ow0 + m % OWT;
+        if (oh < sd.TH && ow < sd.TW)
+          out[out_n + ((int64_t)col * sd.TD + td) * THW +
+              (int64_t)oh * sd.TW + ow] = (__bf16)(acc[i][j][r]);
+      }
+    }
+  }
+}
+
 // ---- host -----------------------------------------------------------------
 // WB layout: [NCOL][kts * KT_PAD]; channel-tile block kt holds the CTILE*27
 // weights for channels [kt*CTILE, (kt+1)*CTILE), zero-padded to KT_PAD.
@@ -224,7 +386,13 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
   auto L = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, dim3(256), 0, s, ip, wp, op, sd, nchunks);
   };
-  if (stride == 1 && ctile == 1) {
+  if (stride == 1 && ctile == 16) {
+    // experimental double-buffered CTILE=16 instances (ctile_opt=16)
+    if (chunk == 64) L(conv3d_spatial_db_kernel<8, 1, 16, 64>);
+    else if (OWT == 32) L(conv3d_spatial_db_kernel<32, 1, 16>);
+    else if (OWT == 16) L(conv3d_spatial_db_kernel<16, 1, 16>);
+    else L(conv3d_spatial_db_kernel<8, 1, 16>);
+  } else if (stride == 1 && ctile == 1) {
     // single-channel (first-layer) instances: one 32-k-step covers the
     // whole 27-tap K, slab is [1][3][H2][W2]
     if (chunk == 64) L(conv3d_spatial_kernel<8, 1, 1, 64>);
@@ -258,9 +426,12 @@ torch::Tensor conv3d_fwd_spatial(torch::Tensor x, torch::Tensor w,
   sd.TD = (sd.D + 2 - 3) / (int)stride + 1;
   sd.TH = (sd.H + 2 - 3) / (int)stride + 1;
   sd.TW = (sd.W + 2 - 3) / (int)stride + 1;
-  // ctile_opt=1 opts into the CTILE=1 single-channel instances
-  // (stride 1 only); 0 = the default tiling.
-  int ctile = stride == 1 ? ((ctile_opt == 1) ? 1 : 32) : 16;
+  // ctile_opt=1 opts into the CTILE=1 single-channel instances;
+  // ctile_opt=16 into the double-buffered CTILE=16 instances
+  // (both stride 1 only); 0 = the default tiling.
+  int ctile = stride == 1
+                  ? ((ctile_opt == 1 || ctile_opt == 16) ? (int)ctile_opt : 32)
+                  : 16;
   auto wb = prep_wb(wc.reshape({sd.NCOL, (int64_t)sd.KCH * 27}), sd.KCH,
                     ctile);
   sd.Kpad = (int)wb.size(1);
