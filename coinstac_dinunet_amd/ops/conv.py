@@ -19,6 +19,8 @@ _SPATIAL_CI1 = os.environ.get('COINN_SPATIAL_CI1', '0') == '1'
 # Opt-in (compiled, unvalidated): double-buffered CTILE=16 stride-1
 # forward instances — stage slab t+1 during slab t's MFMA k-steps.
 _SPATIAL_DB = os.environ.get('COINN_SPATIAL_DB', '0') == '1'
+# Opt-in (compiled, unvalidated): double-buffered stride-1 wgrad.
+_WGRAD_DB = os.environ.get('COINN_WGRAD_DB', '0') == '1'
 
 
 class _Conv3dFn(torch.autograd.Function):
@@ -75,7 +77,9 @@ class _Conv3dFn(torch.autograd.Function):
                 gx = C.conv3d_dgrad(go, wb, list(xb.shape),
                                     ctx.stride).to(ctx.in_dtype)
         if ctx.needs_input_grad[1]:
-            gw = C.conv3d_wgrad(xb, go, ctx.stride).to(ctx.w_dtype)
+            variant = 1 if (_WGRAD_DB and ctx.stride == 1) else 0
+            gw = C.conv3d_wgrad(xb, go, ctx.stride,
+                                variant).to(ctx.w_dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             gb = C.channel_sum(go)
         return gx, gw, gb, None

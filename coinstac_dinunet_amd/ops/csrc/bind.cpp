@@ -41,7 +41,8 @@ torch::Tensor mfma_probe_gemm(torch::Tensor A, torch::Tensor B);
 torch::Tensor conv3d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride);
 torch::Tensor conv3d_dgrad(torch::Tensor go, torch::Tensor w,
                            std::vector<int64_t> in_shape, int64_t stride);
-torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go, int64_t stride);
+torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go, int64_t stride,
+                           int64_t variant);
 torch::Tensor channel_sum(torch::Tensor go);
 // conv3d_spatial.hip
 torch::Tensor conv3d_fwd_spatial(torch::Tensor x, torch::Tensor w,
@@ -78,7 +79,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe_gemm", &mfma_probe_gemm);
   m.def("conv3d_fwd", &conv3d_fwd);
   m.def("conv3d_dgrad", &conv3d_dgrad);
-  m.def("conv3d_wgrad", &conv3d_wgrad);
+  m.def("conv3d_wgrad", &conv3d_wgrad, py::arg("x"), py::arg("go"),
+        py::arg("stride"), py::arg("variant") = 0);
   m.def("channel_sum", &channel_sum);
   m.def("conv3d_fwd_spatial", &conv3d_fwd_spatial,
         py::arg("x"), py::arg("w"), py::arg("stride"),

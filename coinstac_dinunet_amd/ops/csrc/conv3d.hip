@@ -732,6 +732,164 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Double-buffered wgrad (experimental; routed via variant=1, stride 1):
+// stage the (x slab, go tile) of chunk z+zstride while chunk z's
+// 4 m-subchunks x 27 tap-MFMAs run — one barrier per chunk instead of
+// two, attacking the 67% SQ_WAIT_ANY the profiles show. LDS doubles to
+// ~114 KB at OWT=32 (1 block/CU): whether intra-block overlap beats the
+// lost co-residency is the round-2 measurement.
+// ---------------------------------------------------------------------------
+template <int OWT, int STRIDE, int CHUNK = 128>
+__global__ __launch_bounds__(256) void conv3d_wgrad_s1_db_kernel(
+    const __bf16* __restrict__ x, const __bf16* __restrict__ go,
+    float* __restrict__ dw, ConvDims cd, int64_t nchunks, int64_t zstride) {
+  constexpr int OHT = CHUNK / OWT;
+  constexpr int IW = STRIDE * OWT;
+  constexpr int W2 = IW + (STRIDE == 1 ? 4 : 2);
+  constexpr int H2 = STRIDE * (OHT - 1) + 3;
+  constexpr int CT = STRIDE == 1 ? 32 : 16;
+  constexpr int COT = STRIDE == 1 ? 32 : 64;
+  __shared__ __bf16 sX[2][CT][3][H2][W2];
+  __shared__ __bf16 sGo[2][COT][CHUNK + LDA_PAD];
+
+  const int co0 = blockIdx.x * COT;
+  const int ci0 = blockIdx.y * CT;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wi = (STRIDE == 1) ? (wave >> 1) : wave;
+  const int wj = (STRIDE == 1) ? (wave & 1) : 0;
+  const int row = lane & 15, kg = lane >> 4;
+
+  const int wtiles = (cd.OW + OWT - 1) / OWT;
+  const int htiles = (cd.OH + OHT - 1) / OHT;
+
+  f32x4 acc[27];
+#pragma unroll
+  for (int t = 0; t < 27; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
+
+  const int64_t HW = (int64_t)cd.H * cd.W;
+  const int64_t OHW = (int64_t)cd.OH * cd.OW;
+
+  auto stage = [&](int64_t z, int buf) {
+    int64_t t = z;
+    const int wt = (int)(t % wtiles);
+    t /= wtiles;
+    const int ht = (int)(t % htiles);
+    t /= htiles;
+    const int od = (int)(t % cd.OD);
+    const int n = (int)(t / cd.OD);
+    const int oh0 = ht * OHT, ow0 = wt * OWT;
+
+    constexpr int NXROWS = CT * 3 * H2;
+    const __bf16* xn = x + (int64_t)n * cd.Cin * cd.D * HW;
+    for (int r = tid; r < NXROWS; r += 256) {
+      const int hrow = r % H2;
+      const int a = (r / H2) % 3;
+      const int ci = r / (3 * H2);
+      const int id = STRIDE * od - 1 + a;
+      const int ih = STRIDE * oh0 - 1 + hrow;
+      __bf16* dst = &sX[buf][ci][a][hrow][0];
+      const bool row_ok = (unsigned)id < (unsigned)cd.D &&
+                          (unsigned)ih < (unsigned)cd.H &&
+                          (ci0 + ci) < cd.Cin;
+      if (!row_ok) {
+#pragma unroll
+        for (int col = 0; col < W2; ++col) dst[col] = (__bf16)0.f;
+        continue;
+      }
+      const __bf16* src = xn + ((int64_t)(ci0 + ci) * cd.D + id) * HW +
+                          (int64_t)ih * cd.W;
+      const int iw0 = STRIDE * ow0;
+      dst[0] = (iw0 > 0) ? src[iw0 - 1] : (__bf16)0.f;
+#pragma unroll
+      for (int v = 0; v < IW / 8; ++v) {
+        bf16x8 vec = *reinterpret_cast<const bf16x8*>(src + iw0 + v * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) dst[1 + v * 8 + j] = vec[j];
+      }
+#pragma unroll
+      for (int e = 0; e < W2 - IW - 1; ++e) {
+        const int iw = iw0 + IW + e;
+        dst[1 + IW + e] = (iw < cd.W) ? src[iw] : (__bf16)0.f;
+      }
+    }
+    const __bf16* gon = go + (int64_t)n * cd.Cout * cd.OD * OHW;
+    for (int r = tid; r < COT * OHT; r += 256) {
+      const int oh_off = r % OHT;
+      const int co = r / OHT;
+      __bf16* dst = &sGo[buf][co][oh_off * OWT];
+      const int oh = oh0 + oh_off;
+      if ((co0 + co) >= cd.Cout || oh >= cd.OH) {
+#pragma unroll
+        for (int j = 0; j < OWT; ++j) dst[j] = (__bf16)0.f;
+        continue;
+      }
+      const __bf16* src = gon + ((int64_t)(co0 + co) * cd.OD + od) * OHW +
+                          (int64_t)oh * cd.OW + ow0;
+#pragma unroll
+      for (int v = 0; v < OWT / 8; ++v)
+        *reinterpret_cast<bf16x8*>(dst + v * 8) =
+            *reinterpret_cast<const bf16x8*>(src + v * 8);
+    }
+  };
+
+  auto compute = [&](int buf) {
+#pragma unroll 1
+    for (int ms = 0; ms < CHUNK / 32; ++ms) {
+      bf16x8 afrag;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        afrag[j] = sGo[buf][wi * 16 + row][ms * 32 + kg * 8 + j];
+      const int mbase = ms * 32 + kg * 8;
+      const int oh_off = mbase / OWT;
+      const int ow_off = mbase % OWT;
+#pragma unroll
+      for (int kd = 0; kd < 3; ++kd) {
+#pragma unroll
+        for (int kh = 0; kh < 3; ++kh) {
+#pragma unroll
+          for (int kw = 0; kw < 3; ++kw) {
+            bf16x8 bfrag;
+            const __bf16* src = &sX[buf][wj * 16 + row][kd]
+                                   [STRIDE * oh_off + kh]
+                                   [STRIDE * ow_off + kw];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) bfrag[j] = src[STRIDE * j];
+            acc[(kd * 3 + kh) * 3 + kw] =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    afrag, bfrag, acc[(kd * 3 + kh) * 3 + kw], 0, 0, 0);
+          }
+        }
+      }
+    }
+  };
+
+  int buf = 0;
+  if (blockIdx.z < nchunks) stage(blockIdx.z, 0);
+  __syncthreads();
+  for (int64_t z = blockIdx.z; z < nchunks; z += zstride) {
+    if (z + zstride < nchunks) stage(z + zstride, buf ^ 1);
+    compute(buf);
+    __syncthreads();  // next chunk staged AND this buffer's reads done
+    buf ^= 1;
+  }
+
+  const int K = cd.Cin * 27;
+  const int ccol = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+#pragma unroll 1
+  for (int tp = 0; tp < 27; ++tp) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int co = co0 + wi * 16 + crow0 + r;
+      const int ci = ci0 + wj * 16 + ccol;
+      if (co < cd.Cout && ci < cd.Cin)
+        atomicAdd(&dw[(int64_t)co * K + ci * 27 + tp], acc[tp][r]);
+    }
+  }
+}
+
 // bias grad + (optionally) any channelwise sums: dB[co] = sum over m of go
 __global__ void channel_sum_kernel(const __bf16* __restrict__ go,
                                    float* __restrict__ db, int N, int C,
@@ -822,7 +980,7 @@ torch::Tensor conv3d_dgrad(torch::Tensor go, torch::Tensor w,
 }
 
 torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
-                           int64_t stride) {
+                           int64_t stride, int64_t variant) {
   CHECK_GPU(x);
   auto xc = x.to(torch::kBFloat16).contiguous();
   auto g = go.to(torch::kBFloat16).contiguous();
@@ -861,7 +1019,13 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
                          reinterpret_cast<const __bf16*>(g.data_ptr()),
                          dw.data_ptr<float>(), cd, nchunks, zstride);
     };
-    if (stride == 1) {
+    if (stride == 1 && variant == 1) {
+      // experimental double-buffered instances
+      if (chunk == 64) L(conv3d_wgrad_s1_db_kernel<8, 1, 64>);
+      else if (OWT == 32) L(conv3d_wgrad_s1_db_kernel<32, 1>);
+      else if (OWT == 16) L(conv3d_wgrad_s1_db_kernel<16, 1>);
+      else L(conv3d_wgrad_s1_db_kernel<8, 1>);
+    } else if (stride == 1) {
       if (chunk == 64) L(conv3d_wgrad_s1_kernel<8, 1, 64>);
       else if (OWT == 32) L(conv3d_wgrad_s1_kernel<32, 1>);
       else if (OWT == 16) L(conv3d_wgrad_s1_kernel<16, 1>);

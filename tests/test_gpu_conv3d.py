@@ -160,3 +160,32 @@ def test_conv3d_fwd_spatial_double_buffered(dev, case):
     # CTILE=16 math is not guaranteed; only the reference bound is.
     torch.testing.assert_close(out_db.float(), out_sb.float(),
                                rtol=5e-2, atol=tol)
+
+
+@pytest.mark.parametrize('case', [(2, 32, 32, 12, 16, 16),
+                                  (1, 48, 64, 8, 16, 32),
+                                  (2, 16, 32, 6, 8, 8)])
+def test_conv3d_wgrad_double_buffered(dev, case):
+    """Double-buffered stride-1 wgrad (opt-in: COINN_WGRAD_DB=1) vs the
+    validated single-buffered kernel and the torch fp32 reference."""
+    import os
+    if os.environ.get('COINN_WGRAD_DB') != '1':
+        pytest.skip('wgrad DB instances not opted in (COINN_WGRAD_DB=1)')
+    N, Cin, Cout, D, H, W = case
+    torch.manual_seed(8)
+    x = torch.randn(N, Cin, D, H, W, device=dev, dtype=torch.bfloat16)
+    go_shape = (N, Cout, D, H, W)
+    go = torch.randn(*go_shape, device=dev, dtype=torch.bfloat16) * 0.1
+    dw_db = C.conv3d_wgrad(x, go, 1, 1)
+    dw_sb = C.conv3d_wgrad(x, go, 1, 0)
+    # torch reference
+    xf = x.float().requires_grad_(True)
+    w0 = torch.zeros(Cout, Cin, 3, 3, 3, device=dev, requires_grad=True)
+    out = torch.nn.functional.conv3d(xf, w0, padding=1)
+    out.backward(go.float())
+    ref = w0.grad
+    m = (N * D * H * W) ** 0.5
+    torch.testing.assert_close(dw_db.float(), ref, rtol=5e-2,
+                               atol=5e-2 * m * 0.1)
+    torch.testing.assert_close(dw_db.float(), dw_sb.float(), rtol=2e-2,
+                               atol=2e-2 * m * 0.1)
