@@ -146,6 +146,23 @@ class NNTrainer:
                    use_padded_sampler=False, **kw):
         for k in self.nn:
             self.nn[k].eval()
+        # opt-in inference BN folding: evaluate against conv+BN-fused
+        # copies (every BN kernel = one full activation read+write saved)
+        restore = None
+        if self.cache.get('fuse_bn_eval'):
+            from ..ops.fuse import fuse_conv_bn_eval
+            restore = dict(self.nn)
+            for k in self.nn:
+                self.nn[k] = fuse_conv_bn_eval(self.nn[k])
+        try:
+            return self._evaluation_impl(mode, dataset_list, save_pred,
+                                         use_padded_sampler, **kw)
+        finally:
+            if restore is not None:
+                self.nn.update(restore)
+
+    def _evaluation_impl(self, mode='eval', dataset_list=None,
+                         save_pred=False, use_padded_sampler=False, **kw):
 
         eval_avg, eval_metrics = self.new_averages(), self.new_metrics()
         eval_loaders = []
