@@ -234,3 +234,7 @@ def linear(x, weight, bias=None, relu=False):
         return _LinearFn.apply(x, weight, bias, relu)
     out = torch.nn.functional.linear(x, weight, bias)
     return torch.nn.functional.relu(out) if relu else out
+
+
+# inference-time conv+BN folding (host-side; see ops/fuse.py)
+from .fuse import fold_bn, fuse_conv_bn_eval  # noqa: E402
