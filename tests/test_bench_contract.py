@@ -50,3 +50,21 @@ def test_bench_mlp_model():
     out = _run([sys.executable, 'bench.py', '--model', 'mlp', '--steps', '2',
                 '--warmup', '1', '--batch', '8'])
     assert 'mlp' in out['metric']
+
+
+def test_bench_torchrun_eight_ranks():
+    """Exactly the driver's SCALE shape at N=8 (gloo on CPU; RCCL on the
+    node): rendezvous, device-modulo, MAX-over-ranks, one JSON line."""
+    import socket
+    s = socket.socket()
+    s.bind(('127.0.0.1', 0))
+    port = s.getsockname()[1]
+    s.close()
+    out = _run([sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
+                '--nproc-per-node', '8', '--master-addr', '127.0.0.1',
+                '--master-port', str(port), 'bench.py', '--gpus', '8',
+                '--steps', '1', '--warmup', '1', '--batch', '2',
+                '--vol', '8', '--model', 'mlp'], timeout=420)
+    assert out['n_gpus'] == 8
+    assert out['config']['parallelism'] == 'dsgd-dp8'
+    assert out['config']['global_batch'] == 16
