@@ -24,6 +24,12 @@ class COINNMetrics:
         self.eps = EPS
         self.num_precision = PRECISION
 
+    @property
+    def time(self):
+        """Wall clock, for user metrics that timestamp (metrics.py:69-70)."""
+        import time as _t
+        return _t.time()
+
     def add(self, *args, **kw):
         raise NotImplementedError
 
