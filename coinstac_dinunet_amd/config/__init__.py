@@ -12,8 +12,9 @@ import sys as _sys
 import torch as _torch
 
 # Artifact names (loopback transport + checkpoints).
-grads_file = 'grads.npy'
-avg_grads_file = 'avg_grads.npy'
+grad_file_ext = '.npy'
+grads_file = f'grads{grad_file_ext}'
+avg_grads_file = f'avg_grads{grad_file_ext}'
 weights_file = 'weights.tar'
 
 # Metrics formatting.
@@ -22,6 +23,8 @@ metrics_num_precision = 5
 
 # Minimum improvement for "performance improved" checkpoint selection.
 score_delta = 1e-4
+score_high = 1.0
+score_low = 0.0
 
 # Unbounded sentinel (load_limit default, worst-possible minimize score).
 max_size = _sys.maxsize

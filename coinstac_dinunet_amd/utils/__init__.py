@@ -79,3 +79,6 @@ def save_cache(cache, log_dir=None, name='logs'):
 def lazy_debug(x, add=1):
     """Log-frequency thinner: true ~logarithmically often in x."""
     return x % int(_math.log(x + 1) + add) == 0
+
+# re-exports for parity with the reference utils namespace
+from .utils import duration, performance_improved_, stop_training_  # noqa: E402,F401

@@ -37,3 +37,10 @@ class duration:
     def __exit__(self, *exc):
         self.cache.setdefault(self.key, []).append(_time.time() - self.t0)
         return False
+
+
+def lazy_debug(x, add=1):
+    """Log-frequency thinner (also exported from utils; reference
+    logger.py:23-24)."""
+    from . import lazy_debug as _ld
+    return _ld(x, add)

@@ -84,3 +84,8 @@ def flatten_params(model):
     """Total element count and per-param (shape, numel) manifest for bucketing."""
     manifest = [(tuple(p.shape), p.numel()) for p in model.parameters()]
     return sum(n for _, n in manifest), manifest
+
+
+def caste_ndarray(a, dtype='float32'):
+    """Cast helper kept for API parity (reference tensorutils.py:40-41)."""
+    return a.astype(dtype)

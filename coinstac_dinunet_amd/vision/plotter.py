@@ -13,6 +13,11 @@ matplotlib.use('Agg')
 import matplotlib.pyplot as _plt  # noqa: E402
 
 
+COLORS = ['blue', 'maroon', 'magenta', 'teal', 'red',
+          'blueviolet', 'brown', 'cadetblue', 'darkgreen',
+          'darkorange', 'black', 'crimson']
+
+
 def _rolling_mean(x, w=10):
     if len(x) < 2:
         return x
@@ -48,9 +53,10 @@ def plot_progress(cache, log_dir, plot_keys=(), epoch=None):
                 if col >= arr.shape[1]:
                     break
                 y = arr[:, col]
-                ax.plot(y, alpha=0.3)
+                c = COLORS[col % len(COLORS)]
+                ax.plot(y, alpha=0.3, color=c)
                 ax.plot(_np.arange(len(y) - len(_rolling_mean(y)), len(y)),
-                        _rolling_mean(y), label=label)
+                        _rolling_mean(y), label=label, color=c)
                 col += 1
             ax.legend()
             ax.grid(True, alpha=0.3)
