@@ -242,3 +242,24 @@ def test_config_precedence_matches(ref, tmp_path):
               'data_dir', 'num_folds', 'task_id', 'mode', 'agg_engine'):
         assert caches['ref'].get(k) == caches['ours'].get(k), \
             (k, caches['ref'].get(k), caches['ours'].get(k))
+
+
+def test_aucroc_protocol_matches(ref):
+    from coinstac_dinunet.metrics import AUCROCMetrics as RefA
+    from coinstac_dinunet_amd.metrics import AUCROCMetrics as OurA
+    rng = np.random.RandomState(5)
+    ra, oa = RefA(), OurA()
+    for _ in range(3):
+        prob = torch.from_numpy(rng.rand(40).astype(np.float32))
+        true = torch.from_numpy((rng.rand(40) > 0.4).astype(np.int64))
+        ra.add(prob.clone(), true.clone())
+        oa.add(prob.clone(), true.clone())
+    # reference exposes auc() as a method; ours as a property
+    assert abs(ra.auc() - oa.auc) < 1e-6
+    np.testing.assert_allclose(np.asarray(ra.serialize(), dtype=float),
+                               np.asarray(oa.serialize(), dtype=float),
+                               atol=1e-5)
+    rr, ro = RefA(), OurA()
+    rr.reduce_sites([0.7, 0.9])
+    ro.reduce_sites([0.7, 0.9])
+    assert abs(rr.auc() - ro.auc) < 1e-9
