@@ -43,7 +43,7 @@ class COINNMetrics:
         raise NotImplementedError
 
     def extract(self, name):
-        value = getattr(self, name)
+        value = getattr(self, name.lower())  # case-tolerant (metrics.py:73)
         if callable(value):
             value = value()
         return value
