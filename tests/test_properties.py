@@ -169,3 +169,43 @@ def test_power_iteration_recovers_low_rank(n, m, true_rank, seed):
     G = B @ C.t()
     err = torch.norm(bf @ cf.t() - G) / max(torch.norm(G), 1e-8)
     assert err < 5e-2, float(err)
+
+
+@settings(max_examples=30, deadline=None)
+@given(items=st.lists(
+    st.one_of(st.none(), st.booleans(),
+              st.integers(1, 5)),  # falsy values get dropped
+    min_size=0, max_size=12))
+def test_safe_collate_drops_falsy(items):
+    """Corrupt-sample tolerance: falsy items never reach default_collate."""
+    import torch
+    from coinstac_dinunet_amd.data.data import safe_collate
+    batch = [{'x': torch.tensor([float(v)])} if v else v for v in items]
+    kept = [b for b in batch if b]
+    if not kept:
+        return  # nothing to collate either way
+    out = safe_collate(batch)
+    assert out['x'].shape[0] == len(kept)
+
+
+@settings(max_examples=30, deadline=None)
+@given(depth=st.integers(0, 3), seed=st.integers(0, 999))
+def test_save_cache_always_json_serializable(tmp_path_factory, depth, seed):
+    """save_cache must never crash: arbitrary nests of tensors/arrays/
+    callables stringify instead of raising."""
+    import json
+    import os
+    import torch
+    from coinstac_dinunet_amd import utils
+    rng = np.random.RandomState(seed)
+
+    def make(d):
+        if d == 0:
+            opts = [1, 'a', None, 3.5, torch.randn(2), np.arange(3), len]
+            return opts[rng.randint(len(opts))]
+        return {f'k{i}': make(d - 1) for i in range(2)}
+
+    d = tmp_path_factory.mktemp('cache')
+    cache = {'nested': make(depth), 'log_dir': str(d)}
+    utils.save_cache(cache, str(d))
+    json.load(open(os.path.join(str(d), 'logs.json')))  # parses
