@@ -20,6 +20,30 @@ def _run(cmd, timeout=300):
     return json.loads(lines[0])
 
 
+def _free_port():
+    import socket
+    s = socket.socket()
+    s.bind(('127.0.0.1', 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _run_torchrun(nproc, bench_args, timeout=420):
+    """torchrun with one retry on a fresh port: many-process rendezvous is
+    occasionally flaky on a loaded machine; a deterministic failure still
+    fails both attempts."""
+    cmd = lambda port: [sys.executable, '-m', 'torch.distributed.run',
+                        '--nnodes=1', '--nproc-per-node', str(nproc),
+                        '--master-addr', '127.0.0.1',
+                        '--master-port', str(port), 'bench.py',
+                        '--gpus', str(nproc)] + bench_args
+    try:
+        return _run(cmd(_free_port()), timeout=timeout)
+    except (AssertionError, subprocess.TimeoutExpired):
+        return _run(cmd(_free_port()), timeout=timeout)
+
+
 def test_bench_single_process(tmp_path):
     out = _run([sys.executable, 'bench.py', '--steps', '2', '--warmup', '1',
                 '--batch', '2', '--vol', '8'])
@@ -31,16 +55,8 @@ def test_bench_single_process(tmp_path):
 
 def test_bench_torchrun_two_ranks():
     """The driver's N>1 launch shape (gloo here, RCCL on the GPU node)."""
-    import socket
-    s = socket.socket()
-    s.bind(('127.0.0.1', 0))
-    port = s.getsockname()[1]
-    s.close()
-    out = _run([sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
-                '--nproc-per-node', '2', '--master-addr', '127.0.0.1',
-                '--master-port', str(port), 'bench.py', '--gpus', '2',
-                '--steps', '2', '--warmup', '1', '--batch', '2',
-                '--vol', '8'], timeout=420)
+    out = _run_torchrun(2, ['--steps', '2', '--warmup', '1',
+                            '--batch', '2', '--vol', '8'])
     assert out['n_gpus'] == 2
     assert out['config']['parallelism'] == 'dsgd-dp2'
     assert out['config']['global_batch'] == 4
@@ -55,16 +71,8 @@ def test_bench_mlp_model():
 def test_bench_torchrun_eight_ranks():
     """Exactly the driver's SCALE shape at N=8 (gloo on CPU; RCCL on the
     node): rendezvous, device-modulo, MAX-over-ranks, one JSON line."""
-    import socket
-    s = socket.socket()
-    s.bind(('127.0.0.1', 0))
-    port = s.getsockname()[1]
-    s.close()
-    out = _run([sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
-                '--nproc-per-node', '8', '--master-addr', '127.0.0.1',
-                '--master-port', str(port), 'bench.py', '--gpus', '8',
-                '--steps', '1', '--warmup', '1', '--batch', '2',
-                '--vol', '8', '--model', 'mlp'], timeout=420)
+    out = _run_torchrun(8, ['--steps', '1', '--warmup', '1',
+                            '--batch', '2', '--vol', '8', '--model', 'mlp'])
     assert out['n_gpus'] == 8
     assert out['config']['parallelism'] == 'dsgd-dp8'
     assert out['config']['global_batch'] == 16
