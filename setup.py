@@ -34,5 +34,11 @@ if __name__ == '__main__':
         packages=find_packages(include=['coinstac_dinunet_amd*']),
         package_data={'coinstac_dinunet_amd.ops': ['*.so', 'csrc/*']},
         python_requires='>=3.8',
+        # torch (ROCm build) is expected from the environment, like the
+        # reference (setup.py:34 lists only pure-python deps)
         install_requires=['numpy'],
+        extras_require={
+            'vision': ['pillow', 'matplotlib', 'scipy'],
+            'metrics': ['scikit-learn'],
+        },
     )
