@@ -215,6 +215,12 @@ class COINNDataHandle:
                 seed=args.get('seed', 0), shuffle=bool(args.get('shuffle')),
                 drop_last=bool(args.get('drop_last')),
                 total_size=args.get('total_size'))
+            # epoch-aware shuffle: the cursor rebuilds this loader every
+            # epoch, so the epoch counter must come from outside or the
+            # "seeded per-epoch" shuffle degenerates to one fixed order
+            # (ADVICE r1). All sites share the counter (lock-step epochs),
+            # so sequences stay identical across sites.
+            loader_args['sampler'].set_epoch(int(args.get('epoch', 0)))
         return _DataLoader(collate_fn=safe_collate, **loader_args)
 
     def next_iter(self, handle_key=Mode.TRAIN, shuffle=True):
@@ -231,7 +237,8 @@ class COINNDataHandle:
                 dataset = self.dataloader_args.get(handle_key, {}).get('dataset')
             loader = self.get_loader(handle_key=handle_key, shuffle=shuffle,
                                      dataset=dataset, use_padded_sampler=True,
-                                     total_size=self.cache.get('lockstep_total_size'))
+                                     total_size=self.cache.get('lockstep_total_size'),
+                                     epoch=self.cache.get('train_epoch', 0))
             if loader is None:
                 raise RuntimeError(
                     f"'{handle_key}' split is empty on site "
@@ -248,6 +255,8 @@ class COINNDataHandle:
         if self.cache['cursor'] >= self.cache['data_len']:
             out['mode'] = Mode.VALIDATION_WAITING
             self.cache['cursor'] = 0
+            # next rebuild of the train loader gets a fresh shuffle order
+            self.cache['train_epoch'] = self.cache.get('train_epoch', 0) + 1
         return batch, out
 
     # ---- discovery / splits --------------------------------------------

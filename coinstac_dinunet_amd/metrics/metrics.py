@@ -331,12 +331,20 @@ class AUCROCMetrics(COINNMetrics):
             fpr, tpr, _ = roc_curve(self.labels, self.probabilities)
             return round(float(_auc(fpr, tpr)), self.num_precision)
         except ImportError:
-            # rank-based (Mann-Whitney) AUC, identical value
+            # rank-based (Mann-Whitney) AUC with midranks for ties —
+            # identical value to sklearn's trapezoidal roc_curve/auc
             y = np.asarray(self.labels)
-            p = np.asarray(self.probabilities)
+            p = np.asarray(self.probabilities, dtype=np.float64)
             order = np.argsort(p, kind='mergesort')
-            ranks = np.empty_like(order, dtype=np.float64)
-            ranks[order] = np.arange(1, len(p) + 1)
+            ranks = np.empty(len(p), dtype=np.float64)
+            sp = p[order]
+            i = 0
+            while i < len(sp):
+                j = i
+                while j + 1 < len(sp) and sp[j + 1] == sp[i]:
+                    j += 1
+                ranks[order[i:j + 1]] = 0.5 * (i + j) + 1.0
+                i = j + 1
             n1 = y.sum()
             n0 = len(y) - n1
             return round(float((ranks[y == 1].sum() - n1 * (n1 + 1) / 2) / (n0 * n1)),

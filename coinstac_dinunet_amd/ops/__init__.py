@@ -73,7 +73,11 @@ class FusedAdam(torch.optim.Optimizer):
         loss = closure() if closure is not None else None
         C = require_native()
         for group in self.param_groups:
-            params, grads, exp_avgs, exp_avg_sqs = [], [], [], []
+            # bias correction is per-parameter step count in torch.optim.Adam:
+            # bucket params by their own step so a late-appearing grad gets
+            # the right correction. Normally one bucket -> one
+            # kernel launch, same as before.
+            by_step = {}
             for p in group['params']:
                 if p.grad is None:
                     continue
@@ -83,17 +87,16 @@ class FusedAdam(torch.optim.Optimizer):
                     state['exp_avg'] = torch.zeros_like(p, dtype=torch.float32)
                     state['exp_avg_sq'] = torch.zeros_like(p, dtype=torch.float32)
                 state['step'] += 1
-                params.append(p)
-                grads.append(p.grad)
-                exp_avgs.append(state['exp_avg'])
-                exp_avg_sqs.append(state['exp_avg_sq'])
-            if not params:
-                continue
-            step = self.state[params[0]]['step']
+                b = by_step.setdefault(state['step'], ([], [], [], []))
+                b[0].append(p)
+                b[1].append(p.grad)
+                b[2].append(state['exp_avg'])
+                b[3].append(state['exp_avg_sq'])
             beta1, beta2 = group['betas']
-            C.fused_adam(params, grads, exp_avgs, exp_avg_sqs,
-                         group['lr'], beta1, beta2, group['eps'],
-                         group['weight_decay'], step)
+            for step, (params, grads, exp_avgs, exp_avg_sqs) in by_step.items():
+                C.fused_adam(params, grads, exp_avgs, exp_avg_sqs,
+                             group['lr'], beta1, beta2, group['eps'],
+                             group['weight_decay'], step)
         return loss
 
 

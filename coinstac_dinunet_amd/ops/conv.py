@@ -85,6 +85,40 @@ class _Conv3dFn(torch.autograd.Function):
         return gx, gw, gb, None
 
 
+class _ConvPw3dFn(torch.autograd.Function):
+    """1x1x1 (pointwise) conv — a per-position channel GEMM on the
+    bandwidth-shaped pointwise kernels (UNet3D segmentation head)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        C = require_native()
+        xb = x.to(torch.bfloat16)
+        wb = weight.to(torch.bfloat16)
+        empty = torch.empty(0, device=x.device)
+        out = C.conv3d_pw_fwd(xb, wb.view(wb.size(0), wb.size(1)),
+                              bias if bias is not None else empty)
+        ctx.save_for_backward(xb, wb)
+        ctx.has_bias = bias is not None
+        ctx.in_dtype = x.dtype
+        ctx.w_dtype = weight.dtype
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        C = require_native()
+        xb, wb = ctx.saved_tensors
+        go = grad_out.to(torch.bfloat16).contiguous()
+        gx = gw = gb = None
+        w2d = wb.view(wb.size(0), wb.size(1))
+        if ctx.needs_input_grad[0]:
+            gx = C.conv3d_pw_dgrad(go, w2d).to(ctx.in_dtype)
+        if ctx.needs_input_grad[1]:
+            gw = C.conv3d_pw_wgrad(xb, go).view_as(wb).to(ctx.w_dtype)
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            gb = C.channel_sum(go)
+        return gx, gw, gb
+
+
 def conv3d(x, weight, bias=None, stride=1):
     if x.is_cuda and native_available() and weight.shape[2:] == (3, 3, 3):
         return _Conv3dFn.apply(x, weight, bias, int(stride))
@@ -107,4 +141,14 @@ class OpsConv3d(nn.Conv3d):
                 and self.dilation == (1, 1, 1) and self.groups == 1):
             return _Conv3dFn.apply(x, self.weight, self.bias,
                                    int(self.stride[0]))
+        if (x.is_cuda and native_available()
+                and self.kernel_size == (1, 1, 1)
+                and self.padding == (0, 0, 0)
+                and self.stride == (1, 1, 1)
+                and self.dilation == (1, 1, 1) and self.groups == 1):
+            return _ConvPw3dFn.apply(x, self.weight, self.bias)
+        # library fallback must still accept the bf16 activations the fused
+        # kernels upstream emit (bf16 act into fp32 head crashed)
+        if x.is_cuda and x.dtype != self.weight.dtype:
+            x = x.to(self.weight.dtype)
         return super().forward(x)

@@ -51,6 +51,11 @@ torch::Tensor conv3d_dgrad_spatial(torch::Tensor go, torch::Tensor w,
                                    std::vector<int64_t> in_shape);
 torch::Tensor conv3d_dgrad_s2_spatial(torch::Tensor go, torch::Tensor w,
                                       std::vector<int64_t> in_shape);
+// pointwise.hip
+torch::Tensor conv3d_pw_fwd(torch::Tensor x, torch::Tensor w,
+                            torch::Tensor bias);
+torch::Tensor conv3d_pw_dgrad(torch::Tensor go, torch::Tensor w);
+torch::Tensor conv3d_pw_wgrad(torch::Tensor x, torch::Tensor go);
 // bnorm.hip
 std::vector<torch::Tensor> bn3d_fwd(torch::Tensor x, torch::Tensor gamma,
                                     torch::Tensor beta, double eps, bool relu);
@@ -87,6 +92,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("ctile_opt") = 0);
   m.def("conv3d_dgrad_spatial", &conv3d_dgrad_spatial);
   m.def("conv3d_dgrad_s2_spatial", &conv3d_dgrad_s2_spatial);
+  m.def("conv3d_pw_fwd", &conv3d_pw_fwd);
+  m.def("conv3d_pw_dgrad", &conv3d_pw_dgrad);
+  m.def("conv3d_pw_wgrad", &conv3d_pw_wgrad);
   m.def("bn3d_fwd", &bn3d_fwd);
   m.def("bn3d_infer", &bn3d_infer);
   m.def("bn3d_bwd", &bn3d_bwd);

@@ -1,0 +1,191 @@
+// Pointwise (1x1x1) convolution kernels — the UNet3D segmentation head and
+// any channel-mixing conv. A 1x1 conv in NCDHW is a per-position channel
+// GEMM: out[n,co,s] = sum_ci w[co,ci] * x[n,ci,s]. At the shapes that hit
+// this path (head: Cin<=64, Cout=num_class) the op is purely HBM-bound —
+// read x once, write out once — so the kernels are bandwidth-shaped
+// (coalesced along s, weights staged in LDS, fp32 accumulate), not MFMA:
+// the arithmetic intensity (Cout MACs per loaded element, Cout ~ 2) is far
+// below the memory roofline crossover.
+// Reference counterpart: the stock nn.Conv3d head the external UNet
+// computations use (reference utils/tensorutils.py:10-25 safe_concat and
+// metrics/loss.py:1 dice_loss exist to serve that model family).
+#include "common.h"
+
+// out[n,co,s] (+= bias) from x[n,ci,s]; TRANSPOSE_W=true computes
+// dgrad: gx[n,ci,s] = sum_co w[co,ci] * go[n,co,s] with Cin/Cout roles
+// swapped by the caller (then "cin" here is Cout of the conv).
+template <bool TRANSPOSE_W>
+__global__ void pw_conv_kernel(const __bf16* __restrict__ x,
+                               const __bf16* __restrict__ w,
+                               const float* __restrict__ bias,
+                               __bf16* __restrict__ out, int N, int Cin,
+                               int Cout, int64_t S, int co0, int co_chunk) {
+  // W chunk in LDS: co_chunk x Cin bf16 (<= 32x512 = 32 KB worst case;
+  // callers cap co_chunk so this fits comfortably).
+  extern __shared__ __bf16 wlds[];
+  for (int i = threadIdx.x; i < co_chunk * Cin; i += blockDim.x) {
+    int co = co0 + i / Cin, ci = i % Cin;
+    wlds[i] = TRANSPOSE_W ? w[(int64_t)ci * Cout + co]
+                          : w[(int64_t)co * Cin + ci];
+  }
+  __syncthreads();
+
+  const int64_t total = (int64_t)N * S;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t n = idx / S, s = idx % S;
+    float acc[32];
+#pragma unroll
+    for (int j = 0; j < 32; ++j) acc[j] = 0.f;
+    const __bf16* xp = x + ((int64_t)n * Cin) * S + s;
+    for (int ci = 0; ci < Cin; ++ci) {
+      const float xv = (float)xp[(int64_t)ci * S];  // coalesced along s
+      for (int j = 0; j < co_chunk; ++j)
+        acc[j] += (float)wlds[j * Cin + ci] * xv;
+    }
+    __bf16* op = out + ((int64_t)n * Cout + co0) * S + s;
+    for (int j = 0; j < co_chunk; ++j) {
+      float v = acc[j];
+      if (bias != nullptr) v += bias[co0 + j];
+      op[(int64_t)j * S] = (__bf16)v;
+    }
+  }
+}
+
+// gw[co,ci] = sum_{n,s} go[n,co,s] * x[n,ci,s]. Output is tiny (Co x Ci);
+// each block accumulates a per-block fp32 partial tile over its slice of
+// (n,s) in registers, reduces through LDS, then atomically adds into gw.
+// CO_T x CI_T accumulators per thread; grid.y tiles (co, ci) chunks.
+template <int CO_T, int CI_T>
+__global__ void pw_wgrad_kernel(const __bf16* __restrict__ x,
+                                const __bf16* __restrict__ go,
+                                float* __restrict__ gw, int N, int Cin,
+                                int Cout, int64_t S) {
+  const int nco = (Cout + CO_T - 1) / CO_T;
+  const int co0 = (blockIdx.y % nco) * CO_T;
+  const int ci0 = (blockIdx.y / nco) * CI_T;
+  const int cot = min(CO_T, Cout - co0), cit = min(CI_T, Cin - ci0);
+
+  float acc[CO_T][CI_T];
+#pragma unroll
+  for (int a = 0; a < CO_T; ++a)
+#pragma unroll
+    for (int b = 0; b < CI_T; ++b) acc[a][b] = 0.f;
+
+  const int64_t total = (int64_t)N * S;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t n = idx / S, s = idx % S;
+    float gv[CO_T], xv[CI_T];
+    for (int a = 0; a < cot; ++a)
+      gv[a] = (float)go[((int64_t)n * Cout + co0 + a) * S + s];
+    for (int b = 0; b < cit; ++b)
+      xv[b] = (float)x[((int64_t)n * Cin + ci0 + b) * S + s];
+    for (int a = 0; a < cot; ++a)
+      for (int b = 0; b < cit; ++b) acc[a][b] += gv[a] * xv[b];
+  }
+
+  // block reduction: waves fold their lanes, wave leaders sum in LDS
+  __shared__ float part[4][CO_T][CI_T];
+  for (int a = 0; a < cot; ++a)
+    for (int b = 0; b < cit; ++b) {
+      float v = acc[a][b];
+      for (int off = WAVE_SIZE / 2; off > 0; off >>= 1)
+        v += __shfl_down(v, off);
+      if ((threadIdx.x & 63) == 0) part[threadIdx.x >> 6][a][b] = v;
+    }
+  __syncthreads();
+  if (threadIdx.x < (unsigned)(cot * cit)) {
+    const int a = threadIdx.x / cit, b = threadIdx.x % cit;
+    float t = 0.f;
+    for (int wv = 0; wv < (int)(blockDim.x >> 6); ++wv) t += part[wv][a][b];
+    atomicAdd(&gw[(int64_t)(co0 + a) * Cin + ci0 + b], t);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+static int pw_grid(int64_t total) {
+  int64_t blocks = (total + ELEM_BLOCK - 1) / ELEM_BLOCK;
+  if (blocks > 4096) blocks = 4096;  // grid-stride covers the rest
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+torch::Tensor conv3d_pw_fwd(torch::Tensor x, torch::Tensor w,
+                            torch::Tensor bias) {
+  CHECK_GPU(x);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "pw_fwd wants bf16 x");
+  auto xc = x.contiguous();
+  auto wc = w.to(torch::kBFloat16).contiguous();
+  const int N = (int)xc.size(0), Cin = (int)xc.size(1);
+  const int Cout = (int)wc.size(0);
+  const int64_t S = xc.numel() / ((int64_t)N * Cin);
+  TORCH_CHECK(wc.numel() == (int64_t)Cout * Cin, "1x1x1 weight expected");
+  auto sizes = xc.sizes().vec();
+  sizes[1] = Cout;
+  auto out = torch::empty(sizes, xc.options());
+  const float* bp = nullptr;
+  torch::Tensor bc;
+  if (bias.defined() && bias.numel() > 0) {
+    bc = bias.to(torch::kFloat32).contiguous();
+    bp = bc.data_ptr<float>();
+  }
+  for (int co0 = 0; co0 < Cout; co0 += 32) {
+    const int chunk = std::min(32, Cout - co0);
+    hipLaunchKernelGGL((pw_conv_kernel<false>), dim3(pw_grid((int64_t)N * S)),
+                       dim3(ELEM_BLOCK), chunk * Cin * sizeof(__bf16),
+                       current_stream(),
+                       reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                       reinterpret_cast<const __bf16*>(wc.data_ptr()), bp,
+                       reinterpret_cast<__bf16*>(out.data_ptr()), N, Cin,
+                       Cout, S, co0, chunk);
+  }
+  return out;
+}
+
+torch::Tensor conv3d_pw_dgrad(torch::Tensor go, torch::Tensor w) {
+  CHECK_GPU(go);
+  auto g = go.to(torch::kBFloat16).contiguous();
+  auto wc = w.to(torch::kBFloat16).contiguous();
+  const int N = (int)g.size(0), Cout = (int)g.size(1);
+  const int Cin = (int)wc.size(1);
+  const int64_t S = g.numel() / ((int64_t)N * Cout);
+  auto sizes = g.sizes().vec();
+  sizes[1] = Cin;
+  auto gx = torch::empty(sizes, g.options());
+  // roles swapped: iterate over ci chunks of the *output* gx
+  for (int ci0 = 0; ci0 < Cin; ci0 += 32) {
+    const int chunk = std::min(32, Cin - ci0);
+    hipLaunchKernelGGL((pw_conv_kernel<true>), dim3(pw_grid((int64_t)N * S)),
+                       dim3(ELEM_BLOCK), chunk * Cout * sizeof(__bf16),
+                       current_stream(),
+                       reinterpret_cast<const __bf16*>(g.data_ptr()),
+                       reinterpret_cast<const __bf16*>(wc.data_ptr()), nullptr,
+                       reinterpret_cast<__bf16*>(gx.data_ptr()), N, Cout, Cin,
+                       S, ci0, chunk);
+  }
+  return gx;
+}
+
+torch::Tensor conv3d_pw_wgrad(torch::Tensor x, torch::Tensor go) {
+  CHECK_GPU(x);
+  auto xc = x.to(torch::kBFloat16).contiguous();
+  auto g = go.to(torch::kBFloat16).contiguous();
+  const int N = (int)xc.size(0), Cin = (int)xc.size(1);
+  const int Cout = (int)g.size(1);
+  const int64_t S = xc.numel() / ((int64_t)N * Cin);
+  auto gw = torch::zeros({Cout, Cin},
+                         xc.options().dtype(torch::kFloat32));
+  constexpr int CO_T = 4, CI_T = 16;
+  const int nco = (Cout + CO_T - 1) / CO_T;
+  const int nci = (Cin + CI_T - 1) / CI_T;
+  hipLaunchKernelGGL((pw_wgrad_kernel<CO_T, CI_T>),
+                     dim3(pw_grid((int64_t)N * S), nco * nci),
+                     dim3(ELEM_BLOCK), 0, current_stream(),
+                     reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                     reinterpret_cast<const __bf16*>(g.data_ptr()),
+                     gw.data_ptr<float>(), N, Cin, Cout, S);
+  return gw;
+}
