@@ -34,6 +34,8 @@ torch::Tensor confusion_matrix(torch::Tensor pred, torch::Tensor true_,
 // linear.hip
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor weight,
                          torch::Tensor bias, bool relu);
+torch::Tensor linear_dgrad(torch::Tensor go, torch::Tensor weight);
+torch::Tensor linear_wgrad(torch::Tensor go, torch::Tensor x);
 torch::Tensor colsum(torch::Tensor g);
 void gram_schmidt(torch::Tensor m, double eps);
 // conv3d.hip
@@ -79,6 +81,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("prf1a_counts", &prf1a_counts);
   m.def("confusion_matrix", &confusion_matrix);
   m.def("linear_fwd", &linear_fwd);
+  m.def("linear_dgrad", &linear_dgrad);
+  m.def("linear_wgrad", &linear_wgrad);
   m.def("colsum", &colsum);
   m.def("gram_schmidt", &gram_schmidt);
   m.def("mfma_probe_gemm", &mfma_probe_gemm);

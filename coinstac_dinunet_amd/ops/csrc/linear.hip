@@ -28,12 +28,21 @@ __device__ inline float ldf<__hip_bfloat16>(const __hip_bfloat16* p) {
   return __bfloat162float(*p);
 }
 
+template <typename T>
+__device__ inline T stf(float v) { return (T)v; }
+template <>
+__device__ inline __hip_bfloat16 stf<__hip_bfloat16>(float v) {
+  return __float2bfloat16(v);
+}
+
 // C[M,N] = A[M,K] (row-major, lda) @ B[N,K]^T (row-major, ldb = weight
 // layout) + bias; optional ReLU. One wave computes 32x32 via 2x2 fragments.
-template <typename TA, typename TB, bool RELU, bool HAS_BIAS>
+// TC = output element type (bf16 epilogue writes directly — no fp32
+// round-trip tensor).
+template <typename TA, typename TB, typename TC, bool RELU, bool HAS_BIAS>
 __global__ __launch_bounds__(256) void linear_fwd_kernel(
     const TA* __restrict__ A, const TB* __restrict__ B,
-    const float* __restrict__ bias, float* __restrict__ C, int M, int N,
+    const float* __restrict__ bias, TC* __restrict__ C, int M, int N,
     int K, int lda, int ldb, int ldc) {
   __shared__ float sA[BM][LDK];
   __shared__ float sB[BN][LDK];
@@ -98,34 +107,35 @@ __global__ __launch_bounds__(256) void linear_fwd_kernel(
           float v = acc[i][j][r];
           if (HAS_BIAS) v += bias[gn];
           if (RELU) v = fmaxf(v, 0.f);
-          C[(int64_t)gm * ldc + gn] = v;
+          C[(int64_t)gm * ldc + gn] = stf<TC>(v);
         }
       }
     }
   }
 }
 
-template <typename TA, typename TB>
+template <typename TA, typename TB, typename TC>
 static void launch_linear(const TA* A, const TB* B, const float* bias,
-                          float* C, int M, int N, int K, int lda, int ldb,
+                          TC* C, int M, int N, int K, int lda, int ldb,
                           int ldc, bool relu, hipStream_t stream) {
   dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
   dim3 block(256);
   if (bias) {
     if (relu)
-      hipLaunchKernelGGL((linear_fwd_kernel<TA, TB, true, true>), grid, block,
-                         0, stream, A, B, bias, C, M, N, K, lda, ldb, ldc);
+      hipLaunchKernelGGL((linear_fwd_kernel<TA, TB, TC, true, true>), grid,
+                         block, 0, stream, A, B, bias, C, M, N, K, lda, ldb,
+                         ldc);
     else
-      hipLaunchKernelGGL((linear_fwd_kernel<TA, TB, false, true>), grid,
+      hipLaunchKernelGGL((linear_fwd_kernel<TA, TB, TC, false, true>), grid,
                          block, 0, stream, A, B, bias, C, M, N, K, lda, ldb,
                          ldc);
   } else {
     if (relu)
-      hipLaunchKernelGGL((linear_fwd_kernel<TA, TB, true, false>), grid,
+      hipLaunchKernelGGL((linear_fwd_kernel<TA, TB, TC, true, false>), grid,
                          block, 0, stream, A, B, nullptr, C, M, N, K, lda,
                          ldb, ldc);
     else
-      hipLaunchKernelGGL((linear_fwd_kernel<TA, TB, false, false>), grid,
+      hipLaunchKernelGGL((linear_fwd_kernel<TA, TB, TC, false, false>), grid,
                          block, 0, stream, A, B, nullptr, C, M, N, K, lda,
                          ldb, ldc);
   }
@@ -140,7 +150,7 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor weight,
   auto b = has_bias ? bias.to(torch::kFloat32).contiguous() : bias;
   int64_t M = x2.size(0), K = x2.size(1), N = w.size(0);
   TORCH_CHECK(w.size(1) == K, "weight/input K mismatch");
-  auto out = torch::empty({M, N}, x2.options().dtype(torch::kFloat32));
+  auto out = torch::empty({M, N}, x2.options());
   auto stream = current_stream();
   const float* bp = has_bias ? b.data_ptr<float>() : nullptr;
 
@@ -153,12 +163,157 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor weight,
              w.scalar_type() == torch::kBFloat16) {
     launch_linear(reinterpret_cast<const __hip_bfloat16*>(x2.data_ptr()),
                   reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()), bp,
-                  out.data_ptr<float>(), (int)M, (int)N, (int)K, (int)K,
-                  (int)K, (int)N, relu, stream);
+                  reinterpret_cast<__hip_bfloat16*>(out.data_ptr()), (int)M,
+                  (int)N, (int)K, (int)K, (int)K, (int)N, relu, stream);
   } else {
     TORCH_CHECK(false, "linear_fwd: unsupported dtype combination");
   }
-  return out.to(x2.scalar_type());
+  return out;
+}
+
+// ---------------------------------------------------------------------------
+// K2 backward — MFMA kernels for the two backward GEMMs (replaces the
+// rocBLAS calls the round-1 build used).
+//   dgrad:  gx[M,K] = go[M,N] @ W[N,K]        (contraction over N, A direct)
+//   wgrad:  gw[N,K] = go[M,N]^T @ x[M,K]      (contraction over M, A transp)
+// Unified form: C[i,j] = sum_c opA(i,c) * B[c,j] with B row-major [C,ldb]
+// contracted along its rows. Staging differs from the fwd kernel: B tiles
+// are read coalesced along j and written transposed into LDS.
+// ---------------------------------------------------------------------------
+template <typename TA, typename TB, typename TC, bool TRANS_A>
+__global__ __launch_bounds__(256) void linear_bwd_kernel(
+    const TA* __restrict__ A, const TB* __restrict__ B, TC* __restrict__ C,
+    int M, int N, int K, int lda, int ldb, int ldc) {
+  // M = rows of C, N = cols of C, K = contraction length
+  __shared__ float sA[BM][LDK];
+  __shared__ float sB[BN][LDK];
+
+  const int bm = blockIdx.x * BM;
+  const int bn = blockIdx.y * BN;
+  const int tid = threadIdx.x;
+  const int wave = tid / WAVE_SIZE;
+  const int lane = tid % WAVE_SIZE;
+  const int wm = (wave >> 1) * 32;
+  const int wn = (wave & 1) * 32;
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int idx = tid + r * 256;  // 0..1023 = BM*BK
+      if (TRANS_A) {
+        // A[c,i] layout: read coalesced along i, store sA[i][c]
+        int c = idx / BM, row = idx % BM;
+        int gm = bm + row, gk = k0 + c;
+        sA[row][c] =
+            (gm < M && gk < K) ? ldf(&A[(int64_t)gk * lda + gm]) : 0.f;
+      } else {
+        int row = idx / BK, col = idx % BK;
+        int gm = bm + row, gk = k0 + col;
+        sA[row][col] =
+            (gm < M && gk < K) ? ldf(&A[(int64_t)gm * lda + gk]) : 0.f;
+      }
+      // B[c,j]: read coalesced along j, store transposed sB[j][c]
+      int c = idx / BN, col = idx % BN;
+      int gn = bn + col, gk = k0 + c;
+      sB[col][c] = (gn < N && gk < K) ? ldf(&B[(int64_t)gk * ldb + gn]) : 0.f;
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 4) {
+      const int ar = lane & 15, ak = lane >> 4;
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+          float a = sA[wm + i * 16 + ar][kk + ak];
+          float b = sB[wn + j * 16 + ar][kk + ak];
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[i][j],
+                                                           0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  const int ccol = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int gm = bm + wm + i * 16 + crow0 + r;
+        int gn = bn + wn + j * 16 + ccol;
+        if (gm < M && gn < N)
+          C[(int64_t)gm * ldc + gn] = stf<TC>(acc[i][j][r]);
+      }
+}
+
+template <typename TA, typename TB, typename TC, bool TRANS_A>
+static void launch_linear_bwd(const TA* A, const TB* B, TC* C, int M, int N,
+                              int K, int lda, int ldb, int ldc,
+                              hipStream_t stream) {
+  dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
+  hipLaunchKernelGGL((linear_bwd_kernel<TA, TB, TC, TRANS_A>), grid,
+                     dim3(256), 0, stream, A, B, C, M, N, K, lda, ldb, ldc);
+}
+
+// gx[M,K] = go[M,N] @ W[N,K]; output dtype follows go.
+torch::Tensor linear_dgrad(torch::Tensor go, torch::Tensor weight) {
+  CHECK_GPU(go);
+  auto g = go.contiguous();
+  auto w = weight.to(g.scalar_type()).contiguous();
+  int64_t M = g.size(0), N = g.size(1), K = w.size(1);
+  TORCH_CHECK(w.size(0) == N, "weight/grad N mismatch");
+  auto gx = torch::empty({M, K}, g.options());
+  auto stream = current_stream();
+  if (g.scalar_type() == torch::kFloat32) {
+    launch_linear_bwd<float, float, float, false>(
+        g.data_ptr<float>(), w.data_ptr<float>(), gx.data_ptr<float>(),
+        (int)M, (int)K, (int)N, (int)N, (int)K, (int)K, stream);
+  } else if (g.scalar_type() == torch::kBFloat16) {
+    launch_linear_bwd<__hip_bfloat16, __hip_bfloat16, __hip_bfloat16, false>(
+        reinterpret_cast<const __hip_bfloat16*>(g.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+        reinterpret_cast<__hip_bfloat16*>(gx.data_ptr()), (int)M, (int)K,
+        (int)N, (int)N, (int)K, (int)K, stream);
+  } else {
+    TORCH_CHECK(false, "linear_dgrad: fp32 or bf16 only");
+  }
+  return gx;
+}
+
+// gw[N,K] = go[M,N]^T @ x[M,K]; fp32 out (master-weight gradient).
+torch::Tensor linear_wgrad(torch::Tensor go, torch::Tensor x) {
+  CHECK_GPU(go);
+  auto g = go.contiguous();
+  auto x2 = x.to(g.scalar_type()).contiguous();
+  int64_t M = g.size(0), N = g.size(1), K = x2.size(1);
+  TORCH_CHECK(x2.size(0) == M, "x/grad M mismatch");
+  auto gw = torch::empty({N, K}, g.options().dtype(torch::kFloat32));
+  auto stream = current_stream();
+  if (g.scalar_type() == torch::kFloat32) {
+    launch_linear_bwd<float, float, float, true>(
+        g.data_ptr<float>(), x2.data_ptr<float>(), gw.data_ptr<float>(),
+        (int)N, (int)K, (int)M, (int)N, (int)K, (int)K, stream);
+  } else if (g.scalar_type() == torch::kBFloat16) {
+    launch_linear_bwd<__hip_bfloat16, __hip_bfloat16, float, true>(
+        reinterpret_cast<const __hip_bfloat16*>(g.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(x2.data_ptr()),
+        gw.data_ptr<float>(), (int)N, (int)K, (int)M, (int)N, (int)K,
+        (int)K, stream);
+  } else {
+    TORCH_CHECK(false, "linear_wgrad: fp32 or bf16 only");
+  }
+  return gw;
 }
 
 // bias gradient: column sum of grad_out [M,N] -> [N]

@@ -177,6 +177,39 @@ def test_linear_bf16(dev):
     torch.testing.assert_close(out.float(), ref, rtol=2e-2, atol=2e-2)
 
 
+def test_linear_bwd_kernels_vs_fp32(dev):
+    """linear_dgrad/linear_wgrad MFMA kernels directly vs fp32 matmul;
+    asymmetric spikes catch transposed staging."""
+    C = ops.require_native()
+    torch.manual_seed(21)
+    for M, N, K in [(64, 64, 64), (100, 33, 17), (7, 130, 66), (1024, 256, 66)]:
+        go = torch.randn(M, N, device=dev)
+        w = torch.randn(N, K, device=dev)
+        x = torch.randn(M, K, device=dev)
+        go[min(2, M - 1), min(5, N - 1)] += 7.0
+        w[min(3, N - 1), min(1, K - 1)] += 5.0
+        gx = C.linear_dgrad(go, w)
+        torch.testing.assert_close(gx, go @ w, rtol=1e-4, atol=1e-3)
+        gw = C.linear_wgrad(go, x)
+        torch.testing.assert_close(gw, go.t() @ x, rtol=1e-4, atol=1e-3)
+
+
+def test_linear_bwd_kernels_bf16(dev):
+    C = ops.require_native()
+    torch.manual_seed(22)
+    go = torch.randn(64, 32, device=dev, dtype=torch.bfloat16)
+    w = torch.randn(32, 66, device=dev, dtype=torch.bfloat16)
+    x = torch.randn(64, 66, device=dev, dtype=torch.bfloat16)
+    gx = C.linear_dgrad(go, w)
+    assert gx.dtype == torch.bfloat16
+    torch.testing.assert_close(gx.float(), go.float() @ w.float(),
+                               rtol=2e-2, atol=2e-2)
+    gw = C.linear_wgrad(go, x)
+    assert gw.dtype == torch.float32  # master-weight gradient
+    torch.testing.assert_close(gw, go.float().t() @ x.float(),
+                               rtol=2e-2, atol=2e-2)
+
+
 def test_graft_smoke(dev):
     import __graft_entry__
     __graft_entry__.smoke()

@@ -6,12 +6,14 @@ set -euo pipefail
 mkdir -p gpurun_out
 
 echo "== full GPU suite (incl. new pointwise/unet) =="
-timeout 600 python -m pytest tests/ -q -m gpu 2>&1 | tee gpurun_out/r2_gpu_suite.log | tail -5
+timeout 600 python -m pytest tests/ -q -m gpu 2>&1 \
+    | tee gpurun_out/r2_gpu_suite.log | tail -5 || true
 
 echo "== gated numerics (CTILE=1, fwd-DB, wgrad-DB) =="
 COINN_SPATIAL_CI1=1 COINN_SPATIAL_DB=1 COINN_WGRAD_DB=1 \
-    timeout 420 python -m pytest tests/test_gpu_conv3d.py -q -x \
-    -k "ctile1 or double_buffered" 2>&1 | tee gpurun_out/r2_scaffold_tests.log
+    timeout 420 python -m pytest tests/test_gpu_conv3d.py -q \
+    -k "ctile1 or double_buffered" 2>&1 \
+    | tee gpurun_out/r2_scaffold_tests.log || true
 
 echo "== A/B bench: baseline =="
 timeout 180 python bench.py --steps 15 --warmup 5 \
