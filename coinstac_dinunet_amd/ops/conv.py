@@ -12,14 +12,19 @@ import torch.nn.functional as F
 
 from . import native_available, require_native
 
-# Opt-in: route Cin<16 stride-1 forwards through the CTILE=1 spatial
-# instances instead of the igemm fallback (hardware-validated
-# numerically; default off until measured).
-_SPATIAL_CI1 = os.environ.get('COINN_SPATIAL_CI1', '0') == '1'
-# Opt-in (compiled, unvalidated): double-buffered CTILE=16 stride-1
-# forward instances — stage slab t+1 during slab t's MFMA k-steps.
+# Cin<16 stride-1 forwards route through the CTILE=1 spatial instances
+# instead of the igemm fallback. DEFAULT ON since round 2: measured A/B on
+# MI355X (profiles/r2_scaffold_ab.md) — 43.79 vs 44.46 ms/step on the VBM
+# flagship. COINN_SPATIAL_CI1=0 restores the igemm routing.
+_SPATIAL_CI1 = os.environ.get('COINN_SPATIAL_CI1', '1') == '1'
+# Double-buffered CTILE=16 stride-1 forward instances — numerically
+# validated on hardware but measured FLAT on the flagship (44.59 vs 44.46
+# ms/step, profiles/r2_scaffold_ab.md): block-level overlap already hides
+# the staging at multi-block occupancy. Kept compiled behind the env flag.
 _SPATIAL_DB = os.environ.get('COINN_SPATIAL_DB', '0') == '1'
-# Opt-in (compiled, unvalidated): double-buffered stride-1 wgrad.
+# Double-buffered stride-1 wgrad: numerically validated, measured WORSE
+# (45.89 ms/step — LDS doubling costs co-residency more than intra-block
+# overlap buys). Kept compiled behind the env flag.
 _WGRAD_DB = os.environ.get('COINN_WGRAD_DB', '0') == '1'
 
 

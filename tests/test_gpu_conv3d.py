@@ -122,12 +122,8 @@ def test_ops_conv3d_module_autograd(dev):
                                   (1, 1, 8, 16, 32, 32),
                                   (2, 4, 16, 8, 16, 16)])
 def test_conv3d_fwd_spatial_ctile1(dev, case):
-    """CTILE=1 single-channel spatial instances (opt-in routing for the
-    Cin<16 first layer; COINN_SPATIAL_CI1=1 enables them in the module).
-    The kernel entry is callable directly regardless of the env gate."""
-    import os
-    if os.environ.get('COINN_SPATIAL_CI1') != '1':
-        pytest.skip('CTILE=1 instances not opted in (COINN_SPATIAL_CI1=1)')
+    """CTILE=1 single-channel spatial instances (default routing for the
+    Cin<16 first layer since r2 — measured A/B in profiles/r2_scaffold_ab.md)."""
     N, Cin, Cout, D, H, W = case
     torch.manual_seed(4)
     x = torch.randn(N, Cin, D, H, W, device=dev, dtype=torch.bfloat16)
@@ -142,11 +138,8 @@ def test_conv3d_fwd_spatial_ctile1(dev, case):
                                   (1, 48, 64, 8, 16, 32),
                                   (1, 16, 32, 6, 8, 8)])
 def test_conv3d_fwd_spatial_double_buffered(dev, case):
-    """Double-buffered CTILE=16 instances (opt-in: COINN_SPATIAL_DB=1).
+    """Double-buffered CTILE=16 instances (compiled, routing off by default).
     Must agree with the validated single-buffered CTILE=32 kernel."""
-    import os
-    if os.environ.get('COINN_SPATIAL_DB') != '1':
-        pytest.skip('DB instances not opted in (COINN_SPATIAL_DB=1)')
     N, Cin, Cout, D, H, W = case
     torch.manual_seed(6)
     x = torch.randn(N, Cin, D, H, W, device=dev, dtype=torch.bfloat16)
@@ -166,11 +159,8 @@ def test_conv3d_fwd_spatial_double_buffered(dev, case):
                                   (1, 48, 64, 8, 16, 32),
                                   (2, 16, 32, 6, 8, 8)])
 def test_conv3d_wgrad_double_buffered(dev, case):
-    """Double-buffered stride-1 wgrad (opt-in: COINN_WGRAD_DB=1) vs the
+    """Double-buffered stride-1 wgrad (compiled, routing off by default) vs the
     validated single-buffered kernel and the torch fp32 reference."""
-    import os
-    if os.environ.get('COINN_WGRAD_DB') != '1':
-        pytest.skip('wgrad DB instances not opted in (COINN_WGRAD_DB=1)')
     N, Cin, Cout, D, H, W = case
     torch.manual_seed(8)
     x = torch.randn(N, Cin, D, H, W, device=dev, dtype=torch.bfloat16)
