@@ -21,6 +21,7 @@ import numpy as _np
 import torch as _torch
 
 from ..utils import tensorutils as _tu
+from .. import ops as _ops
 from .learner import COINNLearner
 from .reducer import COINNReducer
 
@@ -113,7 +114,7 @@ class PowerSGDLearner(COINNLearner):
             Q = _torch.tensor(_np.asarray(q), dtype=_torch.float32,
                               device=dev)
             P = Ps[i]
-            recon = P @ Q.t()
+            recon = _ops.matmul_abT(P, Q)
             if self.use_error_feedback:
                 error[i] = Ms[i] - recon
             if self.warm_start:  # reuse next round (reference powersgd:110)
@@ -163,7 +164,7 @@ class PowerSGDLearner(COINNLearner):
                     int(self.seed) + int(self.cache['powerSGD_iter']))
                 Q = _torch.randn(M.shape[1], self.rank, generator=gen).to(dev)
             orthogonalize(Q)
-            P = M @ Q
+            P = _ops.matmul_ab(M, Q)
             Ms[i], Ps[i] = M, P
             ship.append(P.cpu().numpy().astype(self.dtype))
         for i, p in rank1:
@@ -192,7 +193,7 @@ class PowerSGDLearner(COINNLearner):
                               device=dev)
             orthogonalize(P)
             Ps[i] = P
-            Q = M.t() @ P
+            Q = _ops.matmul_aTb(M, P)
             ship.append(Q.cpu().numpy().astype(self.dtype))
         _tu.save_arrays(self.state['transferDirectory'] + _os.sep + Q_FILE,
                         ship)

@@ -44,8 +44,37 @@ def power_iteration_BC(B, C, rank=10, numiterations=5, tol=1e-3,
     already-extracted ones via the accumulated sigma^2-weighted projector.
     `generator` (CPU) makes the start vectors deterministic so independent
     ranks recompressing the same factors produce bit-identical results.
+
+    On GPU the whole extraction runs as ONE fused HIP launch (K10,
+    ops/csrc/rankdad.hip) with the Gram precompute on the in-tree MFMA
+    linear kernels — replacing ~rank*numiterations tiny launches; the
+    torch chain below is the CPU path and the oversized-shape fallback.
     """
     device = B.device
+    if B.is_cuda:
+        from .. import ops as _ops
+        if _ops.native_available():
+            n, k = B.shape
+            m = C.shape[0]
+            # one-workgroup kernel: route only launch-bound small shapes
+            per_iter = n * n if k <= m else (n * k * 2 + k * k)
+            if rank <= 32 and per_iter <= (1 << 22):
+                CN = _ops.require_native()
+                if generator is not None:
+                    starts = _torch.stack(
+                        [_torch.rand(n, generator=generator)
+                         for _ in range(rank)]).to(device)
+                else:
+                    starts = _torch.rand(rank, n, device=device)
+                Bf, Cf, nc = CN.power_iter_bc(B.float().contiguous(),
+                                              C.float().contiguous(),
+                                              rank, numiterations, tol,
+                                              starts)
+                nc = int(nc.item())
+                if nc == 0:
+                    return (_torch.zeros(n, 1, device=device),
+                            _torch.zeros(m, 1, device=device))
+                return Bf[:nc].t().contiguous(), Cf[:nc].t().contiguous()
     n, k = B.shape
     m = C.shape[0]
     small_k = k <= m  # work through the k x k Gram when cheaper

@@ -241,5 +241,36 @@ def linear(x, weight, bias=None, relu=False):
     return torch.nn.functional.relu(out) if relu else out
 
 
+# ---- K9/K10/K11 helpers: compression-engine math on in-tree kernels -------
+def matmul_ab(a, b):
+    """a[M,K] @ b[K,N] on the MFMA linear_dgrad kernel (GPU) or torch."""
+    if a.is_cuda and native_available():
+        return require_native().linear_dgrad(a.contiguous(), b.contiguous())
+    return a @ b
+
+
+def matmul_abT(a, b):
+    """a[M,K] @ b[N,K]^T on the MFMA linear_fwd kernel (GPU) or torch."""
+    if a.is_cuda and native_available():
+        C = require_native()
+        empty = torch.empty(0, device=a.device, dtype=a.dtype)
+        return C.linear_fwd(a.contiguous(), b.contiguous(), empty, False)
+    return a @ b.t()
+
+
+def matmul_aTb(a, b):
+    """a[M,N]^T @ b[M,K] (fp32 out) on the MFMA linear_wgrad kernel."""
+    if a.is_cuda and native_available():
+        return require_native().linear_wgrad(a.contiguous(), b.contiguous())
+    return a.t() @ b
+
+
+def row_sum(m):
+    """Per-row sum of a 2D tensor (bias grad from a [out, r] factor)."""
+    if m.is_cuda and native_available():
+        return require_native().rowsum(m.contiguous())
+    return m.sum(1)
+
+
 # inference-time conv+BN folding (host-side; see ops/fuse.py)
 from .fuse import fold_bn, fuse_conv_bn_eval  # noqa: E402

@@ -53,6 +53,11 @@ torch::Tensor conv3d_dgrad_spatial(torch::Tensor go, torch::Tensor w,
                                    std::vector<int64_t> in_shape);
 torch::Tensor conv3d_dgrad_s2_spatial(torch::Tensor go, torch::Tensor w,
                                       std::vector<int64_t> in_shape);
+// rankdad.hip
+std::vector<torch::Tensor> power_iter_bc(torch::Tensor B, torch::Tensor C,
+                                         int64_t rank, int64_t iters,
+                                         double tol, torch::Tensor starts);
+torch::Tensor rowsum(torch::Tensor m);
 // pointwise.hip
 torch::Tensor conv3d_pw_fwd(torch::Tensor x, torch::Tensor w,
                             torch::Tensor bias);
@@ -96,6 +101,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("ctile_opt") = 0);
   m.def("conv3d_dgrad_spatial", &conv3d_dgrad_spatial);
   m.def("conv3d_dgrad_s2_spatial", &conv3d_dgrad_s2_spatial);
+  m.def("power_iter_bc", &power_iter_bc);
+  m.def("rowsum", &rowsum);
   m.def("conv3d_pw_fwd", &conv3d_pw_fwd);
   m.def("conv3d_pw_dgrad", &conv3d_pw_dgrad);
   m.def("conv3d_pw_wgrad", &conv3d_pw_wgrad);

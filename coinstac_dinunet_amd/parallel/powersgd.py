@@ -15,6 +15,7 @@ communication drops from full-gradient to (n+m)*rank per matrix.
 import torch
 import torch.distributed as dist
 
+from .. import ops as _ops
 from ..distrib.learner import COINNLearner
 from ..distrib.powersgd import orthogonalize
 from ..distrib.reducer import COINNReducer
@@ -79,7 +80,7 @@ class RcclPowerSGDLearner(COINNLearner):
             Q0 = torch.randn(M.shape[1], self.rank_approx,
                              generator=gen).to(M.device)
             orthogonalize(Q0)
-            P = M @ Q0
+            P = _ops.matmul_ab(M, Q0)  # K9: skinny GEMM on MFMA kernel
             Ms[i], Ps[i] = M, P
         # round 1: average Ps
         if self.world > 1 and Ps:
@@ -95,7 +96,7 @@ class RcclPowerSGDLearner(COINNLearner):
         Qs = {}
         for i in Ps:
             orthogonalize(Ps[i])
-            Qs[i] = Ms[i].t() @ Ps[i]
+            Qs[i] = _ops.matmul_aTb(Ms[i], Ps[i])  # K9
         # round 2: average Qs + rank-1 grads
         r1 = [p.grad.detach().float().reshape(-1) for _, p in rank1]
         pieces = [Q.reshape(-1) for _, Q in sorted(Qs.items())] + r1
@@ -114,7 +115,7 @@ class RcclPowerSGDLearner(COINNLearner):
                 off += n
         # reconstruct + error feedback
         for i, p in mats:
-            recon = Ps[i] @ Qs[i].t()
+            recon = _ops.matmul_abT(Ps[i], Qs[i])  # K9
             if self.use_error_feedback:
                 error[i] = Ms[i] - recon
             p.grad.copy_(recon.view_as(p))

@@ -22,6 +22,7 @@ aggregator round trip disappears.
 import torch
 import torch.distributed as dist
 
+from .. import ops as _ops
 from ..distrib.rankdad import (DADLearner, _mm_flatten, power_iteration_BC)
 from ..distrib.reducer import COINNReducer
 
@@ -68,14 +69,16 @@ class RcclDADLearner(DADLearner):
             params = dict(mod.named_parameters(recurse=False))
             w = params.get('weight')
             if w is not None:
-                g = gf.mm(af.t()).view_as(w)
+                # K11: factor reconstruction gf[out,r] @ af[in,r]^T on the
+                # in-tree MFMA kernel
+                g = _ops.matmul_abT(gf, af).view_as(w)
                 if w.grad is None:
                     w.grad = g
                 else:
                     w.grad.copy_(g)
             b = params.get('bias')
             if b is not None:
-                gb = gf.sum(1)
+                gb = _ops.row_sum(gf)
                 if b.grad is None:
                     b.grad = gb
                 else:
