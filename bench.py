@@ -20,8 +20,10 @@ import torch
 def parse_args():
     ap = argparse.ArgumentParser()
     ap.add_argument('--gpus', type=int, default=1)
-    ap.add_argument('--steps', type=int, default=20)
-    ap.add_argument('--warmup', type=int, default=5)
+    # default region ~3.5 s on 1 GPU: long enough that driver-side
+    # utilization sampling lands inside it, still < minutes
+    ap.add_argument('--steps', type=int, default=80)
+    ap.add_argument('--warmup', type=int, default=10)
     ap.add_argument('--batch', type=int, default=64,
                     help='per-GPU (per-site) batch size')
     ap.add_argument('--vol', type=int, default=64, help='volume side length')
