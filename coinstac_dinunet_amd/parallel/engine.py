@@ -46,7 +46,12 @@ def init_distributed(backend=None, timeout_s=300):
         os.environ.setdefault('RANK', '0')
         os.environ.setdefault('WORLD_SIZE', '1')
     if backend is None:
-        backend = 'nccl' if torch.cuda.is_available() else 'gloo'
+        # COINN_DIST_BACKEND override: e.g. gloo on a GPU box to rehearse
+        # multi-rank protocol on one device (RCCL refuses co-located ranks:
+        # "Duplicate GPU detected" is by design in NCCL/RCCL 2.x)
+        backend = os.environ.get(
+            'COINN_DIST_BACKEND',
+            'nccl' if torch.cuda.is_available() else 'gloo')
     if backend == 'nccl':
         # modulo: lets N ranks share fewer GPUs (dev boxes); on the 8-GPU
         # node LOCAL_RANK < device_count and this is the identity
@@ -71,6 +76,11 @@ class FlatGradBuffer:
         self.params = [p for p in params if p.requires_grad]
         self.world_size = world_size
         self.comm_stream = comm_stream
+        # evidence/debug knob: run the collectives even at world 1 (a
+        # 1-rank RCCL all-reduce launches real RCCL kernels — used by
+        # tools/r2_multirank.sh to trace RCCL overlapping backward)
+        self.force_collectives = os.environ.get(
+            'COINN_FORCE_ALLREDUCE') == '1'
         device = self.params[0].device if self.params else torch.device('cpu')
         total = sum(p.numel() for p in self.params)
         self.flat = torch.zeros(total, dtype=torch.float32, device=device)
@@ -143,7 +153,8 @@ class FlatGradBuffer:
     def _on_grad_ready(self, p):
         if not getattr(self, '_sync_this_round', False):
             return
-        if self.world_size <= 1 or not dist.is_initialized():
+        if (self.world_size <= 1 and not self.force_collectives) \
+                or not dist.is_initialized():
             return
         bi = self._bucket_of[p]
         self._pending[bi] -= 1
@@ -159,7 +170,8 @@ class FlatGradBuffer:
 
     def finish_round(self):
         """Wait for every in-flight all-reduce, then divide by world size."""
-        if self.world_size <= 1 or not dist.is_initialized():
+        if (self.world_size <= 1 and not self.force_collectives) \
+                or not dist.is_initialized():
             return
         if getattr(self, '_finished', True):
             return

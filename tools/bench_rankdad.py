@@ -5,10 +5,13 @@ Run on a GPU box:  python tools/bench_rankdad.py
 Writes gpurun_out/rankdad_ab.json when run under gpurun.
 """
 import json
+import sys
 import os
 import time
 
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from coinstac_dinunet_amd import ops
 from coinstac_dinunet_amd.distrib import rankdad as rd

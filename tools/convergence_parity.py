@@ -11,9 +11,12 @@ Run on a GPU box: python tools/convergence_parity.py
 Writes gpurun_out/convergence_parity.json.
 """
 import json
+import sys
 import os
 
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
 def build(seed, stock):
