@@ -9,10 +9,18 @@ MI355X notes: NCDHW layout (w-contiguous rows feed the spatial-slab conv
 kernels' 16-byte staging loads); conv runs on OpsConv3d (implicit-GEMM
 MFMA) and BN+ReLU is one fused pass (OpsBatchNorm3d).
 """
+import os
+
 import torch.nn as nn
 
 from ..ops.bnorm import OpsBatchNorm3d
-from ..ops.conv import OpsConv3d
+from ..ops.conv import OpsConv3d, can_fuse_bn_conv, conv_bn3d
+
+# Cross-block BN fusion (normalize-on-load): block i's BN+ReLU is applied
+# inside block i+1's conv kernels during slab staging, so the normalized
+# activation never round-trips HBM. COINN_FUSE_BN=0 restores the separate
+# fused-BN-pass path.
+_FUSE_BN = os.environ.get('COINN_FUSE_BN', '1') == '1'
 
 
 class _ConvBlock(nn.Module):
@@ -41,6 +49,17 @@ class VBMNet(nn.Module):
         self.head = nn.Linear(cin, num_class)
 
     def forward(self, x):
-        x = self.features(x)
+        blocks = list(self.features)
+        if (_FUSE_BN and len(blocks) >= 2
+                and all(can_fuse_bn_conv(blocks[i].bn, blocks[i + 1].conv, x)
+                        for i in range(len(blocks) - 1))):
+            # fused chain: conv_0 -> [bn_i folded into conv_{i+1}'s load]
+            # -> final bn materialized once before the pool
+            x = blocks[0].conv(x)
+            for prev, blk in zip(blocks[:-1], blocks[1:]):
+                x = conv_bn3d(x, prev.bn, blk.conv)
+            x = blocks[-1].bn(x)
+        else:
+            x = self.features(x)
         x = self.pool(x).flatten(1)
         return self.head(x)

@@ -90,6 +90,112 @@ class _Conv3dFn(torch.autograd.Function):
         return gx, gw, gb, None
 
 
+class _ConvBNFn(torch.autograd.Function):
+    """Previous block's BN(+ReLU) folded into this conv's input load.
+
+    The normalized activation z = relu(gamma*xhat + beta) NEVER exists in
+    HBM: the fwd kernels apply the per-channel affine during slab staging
+    (bn_ab = [a, b] with a = gamma*rstd, b = beta - mean*a) and the wgrad
+    kernels re-apply it on load in backward. Per BN layer this deletes one
+    full activation write + one full read vs the separate normalize pass
+    (VERDICT r1 item 3). mean_rstd enters as a non-differentiable constant:
+    bn3d_bwd's dx formula accounts for the statistic dependence.
+    """
+
+    @staticmethod
+    def forward(ctx, x_raw, gamma, beta, mean_rstd, weight, bias, stride):
+        C = require_native()
+        xb = x_raw.to(torch.bfloat16)
+        wb = weight.to(torch.bfloat16)
+        with torch.no_grad():
+            a = gamma.float() * mean_rstd[:, 1]
+            b = beta.float() - mean_rstd[:, 0] * a
+            ab = torch.stack([a, b], 1).contiguous()
+        ow = (xb.size(4) + 2 - 3) // stride + 1
+        oh = (xb.size(3) + 2 - 3) // stride + 1
+        if ow % 8 == 0 and xb.size(1) >= 16 and oh * ow >= 64:
+            out = C.conv3d_fwd_spatial(xb, wb, stride, 0, ab)
+        else:
+            out = C.conv3d_fwd(xb, wb, stride, ab)
+        if bias is not None:
+            out = out + bias.to(out.dtype).view(1, -1, 1, 1, 1)
+        ctx.save_for_backward(xb, wb, gamma, beta, mean_rstd, ab)
+        ctx.stride = stride
+        ctx.has_bias = bias is not None
+        ctx.in_dtype = x_raw.dtype
+        ctx.w_dtype = weight.dtype
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        C = require_native()
+        xb, wb, gamma, beta, mean_rstd, ab = ctx.saved_tensors
+        go = grad_out.to(torch.bfloat16).contiguous()
+        stride = ctx.stride
+        # dz = conv dgrad wrt the (virtual) normalized input — same
+        # routing as the unfused path
+        wsub = (xb.size(4) + 1) // 2
+        hsub = (xb.size(3) + 1) // 2
+        if (stride == 1 and xb.size(4) % 8 == 0 and go.size(1) >= 16
+                and xb.size(3) * xb.size(4) >= 64):
+            dz = C.conv3d_dgrad_spatial(go, wb, list(xb.shape))
+        elif (stride == 2 and wsub % 8 == 0 and go.size(1) >= 32
+                and hsub * wsub >= 128):
+            dz = C.conv3d_dgrad_s2_spatial(go, wb, list(xb.shape))
+        else:
+            dz = C.conv3d_dgrad(go, wb, list(xb.shape), stride)
+        gw = C.conv3d_wgrad(xb, go, stride, 0, ab).to(ctx.w_dtype) \
+            if ctx.needs_input_grad[4] else None
+        gb = C.channel_sum(go) if (ctx.has_bias
+                                   and ctx.needs_input_grad[5]) else None
+        dx, dgamma, dbeta = C.bn3d_bwd(dz, xb, mean_rstd, gamma, beta, True)
+        return (dx.to(ctx.in_dtype), dgamma.to(gamma.dtype),
+                dbeta.to(beta.dtype), None, gw, gb, None)
+
+
+def conv_bn3d(x_raw, bn, conv):
+    """Fused BN(of x_raw, from `bn`)->ReLU->conv(`conv`) on the HIP path.
+
+    `bn` must be an OpsBatchNorm3d with relu=True; `conv` an OpsConv3d in
+    the 3x3x3/pad-1/stride-{1,2} family. Handles running-stat updates with
+    the exact semantics of the unfused module.
+    """
+    C = require_native()
+    xb = x_raw if x_raw.dtype == torch.bfloat16 \
+        else x_raw.to(torch.bfloat16)
+    if bn.training:
+        mean, var, mean_rstd = C.bn3d_stats(xb.detach(), bn.eps)
+        if bn.num_batches_tracked is not None:
+            bn.num_batches_tracked.add_(1)
+        if bn.running_mean is not None:
+            with torch.no_grad():
+                n = xb.numel() // xb.size(1)
+                unbiased = var * (n / max(n - 1, 1))
+                bn.running_mean.mul_(1 - bn.momentum).add_(
+                    mean, alpha=bn.momentum)
+                bn.running_var.mul_(1 - bn.momentum).add_(
+                    unbiased, alpha=bn.momentum)
+    else:
+        mean = bn.running_mean.float()
+        rstd = torch.rsqrt(bn.running_var.float() + bn.eps)
+        mean_rstd = torch.stack([mean, rstd], 1).contiguous()
+    return _ConvBNFn.apply(x_raw, bn.weight, bn.bias, mean_rstd,
+                           conv.weight, conv.bias, int(conv.stride[0]))
+
+
+def can_fuse_bn_conv(bn, conv, x):
+    """True when the (bn -> conv) pair routes onto the fused kernels."""
+    from .bnorm import OpsBatchNorm3d
+    return (x.is_cuda and native_available()
+            and isinstance(bn, OpsBatchNorm3d) and bn.relu
+            and isinstance(conv, OpsConv3d)
+            and conv.kernel_size == (3, 3, 3)
+            and conv.padding == (1, 1, 1)
+            and conv.stride[0] in (1, 2)
+            and conv.stride[0] == conv.stride[1] == conv.stride[2]
+            and conv.dilation == (1, 1, 1) and conv.groups == 1)
+
+
 class _ConvPw3dFn(torch.autograd.Function):
     """1x1x1 (pointwise) conv — a per-position channel GEMM on the
     bandwidth-shaped pointwise kernels (UNet3D segmentation head)."""

@@ -40,15 +40,17 @@ torch::Tensor colsum(torch::Tensor g);
 void gram_schmidt(torch::Tensor m, double eps);
 // conv3d.hip
 torch::Tensor mfma_probe_gemm(torch::Tensor A, torch::Tensor B);
-torch::Tensor conv3d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride);
+torch::Tensor conv3d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
+                         torch::Tensor bn_ab);
 torch::Tensor conv3d_dgrad(torch::Tensor go, torch::Tensor w,
                            std::vector<int64_t> in_shape, int64_t stride);
 torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go, int64_t stride,
-                           int64_t variant);
+                           int64_t variant, torch::Tensor bn_ab);
 torch::Tensor channel_sum(torch::Tensor go);
 // conv3d_spatial.hip
 torch::Tensor conv3d_fwd_spatial(torch::Tensor x, torch::Tensor w,
-                                 int64_t stride, int64_t ctile_opt);
+                                 int64_t stride, int64_t ctile_opt,
+                                 torch::Tensor bn_ab);
 torch::Tensor conv3d_dgrad_spatial(torch::Tensor go, torch::Tensor w,
                                    std::vector<int64_t> in_shape);
 torch::Tensor conv3d_dgrad_s2_spatial(torch::Tensor go, torch::Tensor w,
@@ -66,6 +68,7 @@ torch::Tensor conv3d_pw_wgrad(torch::Tensor x, torch::Tensor go);
 // bnorm.hip
 std::vector<torch::Tensor> bn3d_fwd(torch::Tensor x, torch::Tensor gamma,
                                     torch::Tensor beta, double eps, bool relu);
+std::vector<torch::Tensor> bn3d_stats(torch::Tensor x, double eps);
 torch::Tensor bn3d_infer(torch::Tensor x, torch::Tensor gamma,
                          torch::Tensor beta, torch::Tensor running_mean,
                          torch::Tensor running_var, double eps, bool relu);
@@ -91,14 +94,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum", &colsum);
   m.def("gram_schmidt", &gram_schmidt);
   m.def("mfma_probe_gemm", &mfma_probe_gemm);
-  m.def("conv3d_fwd", &conv3d_fwd);
+  m.def("conv3d_fwd", &conv3d_fwd, py::arg("x"), py::arg("w"),
+        py::arg("stride"), py::arg("bn_ab") = torch::Tensor());
   m.def("conv3d_dgrad", &conv3d_dgrad);
   m.def("conv3d_wgrad", &conv3d_wgrad, py::arg("x"), py::arg("go"),
-        py::arg("stride"), py::arg("variant") = 0);
+        py::arg("stride"), py::arg("variant") = 0,
+        py::arg("bn_ab") = torch::Tensor());
   m.def("channel_sum", &channel_sum);
   m.def("conv3d_fwd_spatial", &conv3d_fwd_spatial,
         py::arg("x"), py::arg("w"), py::arg("stride"),
-        py::arg("ctile_opt") = 0);
+        py::arg("ctile_opt") = 0, py::arg("bn_ab") = torch::Tensor());
   m.def("conv3d_dgrad_spatial", &conv3d_dgrad_spatial);
   m.def("conv3d_dgrad_s2_spatial", &conv3d_dgrad_s2_spatial);
   m.def("power_iter_bc", &power_iter_bc);
@@ -107,6 +112,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3d_pw_dgrad", &conv3d_pw_dgrad);
   m.def("conv3d_pw_wgrad", &conv3d_pw_wgrad);
   m.def("bn3d_fwd", &bn3d_fwd);
+  m.def("bn3d_stats", &bn3d_stats);
   m.def("bn3d_infer", &bn3d_infer);
   m.def("bn3d_bwd", &bn3d_bwd);
 }

@@ -225,6 +225,28 @@ std::vector<torch::Tensor> bn3d_fwd(torch::Tensor x, torch::Tensor gamma,
   return {y, mean, var, mean_rstd};
 }
 
+// stats only — for the fused conv+BN path (normalize-on-load): the
+// normalize pass never runs, the next conv applies the affine on load.
+std::vector<torch::Tensor> bn3d_stats(torch::Tensor x, double eps) {
+  CHECK_GPU(x);
+  auto xc = x.contiguous();
+  TORCH_CHECK(xc.scalar_type() == torch::kBFloat16, "bn3d_stats wants bf16");
+  int N = (int)xc.size(0), C = (int)xc.size(1);
+  int64_t spatial = xc.numel() / ((int64_t)N * C);
+  auto sums = torch::zeros({C, 2}, xc.options().dtype(torch::kFloat32));
+  dim3 grid, blk;
+  bn_grid(spatial, N * C, grid, blk);
+  hipLaunchKernelGGL(bn_reduce_kernel, grid, blk, 0, current_stream(),
+                     reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                     sums.data_ptr<float>(), C, spatial);
+  int64_t per_ch = (int64_t)N * spatial;
+  auto mean = sums.select(1, 0) / (double)per_ch;
+  auto var = sums.select(1, 1) / (double)per_ch - mean * mean;
+  auto rstd = torch::rsqrt(var.clamp_min(0) + eps);
+  auto mean_rstd = torch::stack({mean, rstd}, 1).contiguous();
+  return {mean, var, mean_rstd};
+}
+
 torch::Tensor bn3d_infer(torch::Tensor x, torch::Tensor gamma,
                          torch::Tensor beta, torch::Tensor running_mean,
                          torch::Tensor running_var, double eps, bool relu) {

@@ -135,10 +135,11 @@ __device__ inline __bf16 gather_w(const __bf16* __restrict__ w,
   return w[((int64_t)co * cd.Cin + col) * 27 + r];
 }
 
-template <bool DGRAD, int STRIDE>
+template <bool DGRAD, int STRIDE, bool FUSE_BN = false>
 __global__ __launch_bounds__(256) void conv3d_igemm_kernel(
     const __bf16* __restrict__ Ain, const __bf16* __restrict__ w,
-    __bf16* __restrict__ out, ConvDims cd, int64_t M, int Ncol, int K) {
+    __bf16* __restrict__ out, ConvDims cd, int64_t M, int Ncol, int K,
+    const float* __restrict__ bn_ab = nullptr) {
   __shared__ __bf16 sA[CBM][CBK + LDA_PAD];
   __shared__ __bf16 sBT[CBN][CBK + LDA_PAD];  // [col][k]: lanes read 8
                                               // consecutive k => b128
@@ -191,6 +192,8 @@ __global__ __launch_bounds__(256) void conv3d_igemm_kernel(
       const int r = k - ci * 27;
       const int kd = r / 9, kh = (r / 3) % 3, kw = r % 3;
       const bool k_ok = ci < cd.Cin;
+      float a_c = 1.f, b_c = 0.f;
+      if (FUSE_BN && k_ok) { a_c = bn_ab[ci * 2]; b_c = bn_ab[ci * 2 + 1]; }
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         __bf16 v = (__bf16)0.f;
@@ -199,9 +202,11 @@ __global__ __launch_bounds__(256) void conv3d_igemm_kernel(
           const int ih = ph[j] * STRIDE - 1 + kh;
           const int iw = pw[j] * STRIDE - 1 + kw;
           if ((unsigned)id < (unsigned)cd.D && (unsigned)ih < (unsigned)cd.H &&
-              (unsigned)iw < (unsigned)cd.W)
+              (unsigned)iw < (unsigned)cd.W) {
             v = Ain[(((int64_t)pn[j] * cd.Cin + ci) * cd.D + id) * HW +
                     (int64_t)ih * cd.W + iw];
+            if (FUSE_BN) v = (__bf16)fmaxf(a_c * (float)v + b_c, 0.f);
+          }
         }
         sA[mbase + j][kk_t] = v;
       }
@@ -296,9 +301,11 @@ __global__ __launch_bounds__(256) void conv3d_igemm_kernel(
 // ---------------------------------------------------------------------------
 #define WMB 128  // m positions staged per iteration (4 MFMA/wave/barrier)
 
+template <bool FUSE_BN = false>
 __global__ __launch_bounds__(256) void conv3d_wgrad_kernel(
     const __bf16* __restrict__ x, const __bf16* __restrict__ go,
-    float* __restrict__ dw, ConvDims cd, int64_t M, int K, int64_t chunk) {
+    float* __restrict__ dw, ConvDims cd, int64_t M, int K, int64_t chunk,
+    const float* __restrict__ bn_ab = nullptr) {
   __shared__ __bf16 sGoT[32][WMB + LDA_PAD];  // [co][m]: b128 frag reads
   __shared__ __bf16 sXT[32][WMB + LDA_PAD];   // [k][m]
 
@@ -321,6 +328,8 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_kernel(
   const int kd = rr / 9, kh = (rr / 3) % 3, kw = rr % 3;
   const bool k_ok = ci < cd.Cin && kq < K;
   const bool c_ok = (co0 + ct) < cd.Cout;
+  float a_c = 1.f, b_c = 0.f;
+  if (FUSE_BN && k_ok) { a_c = bn_ab[ci * 2]; b_c = bn_ab[ci * 2 + 1]; }
 
   // incremental output-position decode for m = mb + mi0 (advances by 32)
   int nn, od, oh, ow;
@@ -359,9 +368,11 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_kernel(
           const int iw = jw * cd.stride - 1 + kw;
           if ((unsigned)id < (unsigned)cd.D &&
               (unsigned)ih < (unsigned)cd.H &&
-              (unsigned)iw < (unsigned)cd.W)
+              (unsigned)iw < (unsigned)cd.W) {
             xv = x[(((int64_t)jn * cd.Cin + ci) * cd.D + id) * HW +
                    (int64_t)ih * cd.W + iw];
+            if (FUSE_BN) xv = (__bf16)fmaxf(a_c * (float)xv + b_c, 0.f);
+          }
         }
         sXT[ct][mloc] = xv;
         if (++jw == cd.OW) { jw = 0; if (++jh == cd.OH) { jh = 0;
@@ -587,10 +598,11 @@ static torch::Tensor conv3d_dgrad_s2(torch::Tensor g, torch::Tensor wc,
 // Blocks grid-stride over spatial chunks; partials fold into dw by fp32
 // atomics once at the end.
 // ---------------------------------------------------------------------------
-template <int OWT, int STRIDE, int CHUNK = 128>
+template <int OWT, int STRIDE, int CHUNK = 128, bool FUSE_BN = false>
 __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
     const __bf16* __restrict__ x, const __bf16* __restrict__ go,
-    float* __restrict__ dw, ConvDims cd, int64_t nchunks, int64_t zstride) {
+    float* __restrict__ dw, ConvDims cd, int64_t nchunks, int64_t zstride,
+    const float* __restrict__ bn_ab = nullptr) {
   constexpr int OHT = CHUNK / OWT;
   constexpr int IW = STRIDE * OWT;
   constexpr int W2 = IW + (STRIDE == 1 ? 4 : 2);
@@ -649,18 +661,27 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
       }
       const __bf16* src = xn + ((int64_t)(ci0 + ci) * cd.D + id) * HW +
                           (int64_t)ih * cd.W;
+      float a_c = 1.f, b_c = 0.f;
+      if (FUSE_BN) {
+        a_c = bn_ab[(ci0 + ci) * 2];
+        b_c = bn_ab[(ci0 + ci) * 2 + 1];
+      }
+      auto tx = [&](__bf16 v) -> __bf16 {
+        if (!FUSE_BN) return v;
+        return (__bf16)fmaxf(a_c * (float)v + b_c, 0.f);
+      };
       const int iw0 = STRIDE * ow0;
-      dst[0] = (iw0 > 0) ? src[iw0 - 1] : (__bf16)0.f;
+      dst[0] = (iw0 > 0) ? tx(src[iw0 - 1]) : (__bf16)0.f;
 #pragma unroll
       for (int v = 0; v < IW / 8; ++v) {
         bf16x8 vec = *reinterpret_cast<const bf16x8*>(src + iw0 + v * 8);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) dst[1 + v * 8 + j] = vec[j];
+        for (int j = 0; j < 8; ++j) dst[1 + v * 8 + j] = tx(vec[j]);
       }
 #pragma unroll
       for (int e = 0; e < W2 - IW - 1; ++e) {
         const int iw = iw0 + IW + e;
-        dst[1 + IW + e] = (iw < cd.W) ? src[iw] : (__bf16)0.f;
+        dst[1 + IW + e] = (iw < cd.W) ? tx(src[iw]) : (__bf16)0.f;
       }
     }
     // ---- stage go tile row-wise: [co][m over (OHT x OWT)] --------------
@@ -928,7 +949,8 @@ static ConvDims make_dims(const torch::Tensor& x, const torch::Tensor& w,
   return cd;
 }
 
-torch::Tensor conv3d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride) {
+torch::Tensor conv3d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
+                         torch::Tensor bn_ab) {
   CHECK_GPU(x);
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "conv3d_fwd wants bf16");
   auto xc = x.contiguous();
@@ -936,6 +958,14 @@ torch::Tensor conv3d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride) {
   auto cd = make_dims(xc, wc, (int)stride);
   TORCH_CHECK(wc.size(2) == 3 && wc.size(3) == 3 && wc.size(4) == 3 &&
               wc.size(1) == cd.Cin, "3x3x3 kernels only");
+  const bool fuse = bn_ab.defined() && bn_ab.numel() > 0;
+  torch::Tensor ab;
+  const float* abp = nullptr;
+  if (fuse) {
+    ab = bn_ab.to(torch::kFloat32).contiguous();
+    TORCH_CHECK(ab.numel() == 2 * cd.Cin, "bn_ab must be [Cin,2]");
+    abp = ab.data_ptr<float>();
+  }
   auto out = torch::empty({cd.N, cd.Cout, cd.OD, cd.OH, cd.OW}, xc.options());
   int64_t M = (int64_t)cd.N * cd.OD * cd.OH * cd.OW;
   int K = cd.Cin * 27;
@@ -945,11 +975,17 @@ torch::Tensor conv3d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride) {
                        reinterpret_cast<const __bf16*>(xc.data_ptr()),
                        reinterpret_cast<const __bf16*>(wc.data_ptr()),
                        reinterpret_cast<__bf16*>(out.data_ptr()), cd, M,
-                       cd.Cout, K);
+                       cd.Cout, K, abp);
   };
   TORCH_CHECK(stride == 1 || stride == 2, "stride must be 1 or 2");
-  if (stride == 1) launch(conv3d_igemm_kernel<false, 1>);
-  else launch(conv3d_igemm_kernel<false, 2>);
+  if (fuse) {
+    if (stride == 1) launch(conv3d_igemm_kernel<false, 1, true>);
+    else launch(conv3d_igemm_kernel<false, 2, true>);
+  } else if (stride == 1) {
+    launch(conv3d_igemm_kernel<false, 1>);
+  } else {
+    launch(conv3d_igemm_kernel<false, 2>);
+  }
   return out;
 }
 
@@ -975,12 +1011,13 @@ torch::Tensor conv3d_dgrad(torch::Tensor go, torch::Tensor w,
                      reinterpret_cast<const __bf16*>(g.data_ptr()),
                      reinterpret_cast<const __bf16*>(wc.data_ptr()),
                      reinterpret_cast<__bf16*>(dx.data_ptr()), cd, M, cd.Cin,
-                     K);
+                     K, (const float*)nullptr);
   return dx;
 }
 
 torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
-                           int64_t stride, int64_t variant) {
+                           int64_t stride, int64_t variant,
+                           torch::Tensor bn_ab) {
   CHECK_GPU(x);
   auto xc = x.to(torch::kBFloat16).contiguous();
   auto g = go.to(torch::kBFloat16).contiguous();
@@ -994,6 +1031,15 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
   int64_t M = (int64_t)cd.N * cd.OD * cd.OH * cd.OW;
   auto dw = torch::zeros({cd.Cout, (int64_t)K},
                          xc.options().dtype(torch::kFloat32));
+  const bool fuse = bn_ab.defined() && bn_ab.numel() > 0;
+  torch::Tensor ab;
+  const float* abp = nullptr;
+  if (fuse) {
+    TORCH_CHECK(variant == 0, "fused BN wgrad: default variant only");
+    ab = bn_ab.to(torch::kFloat32).contiguous();
+    TORCH_CHECK(ab.numel() == 2 * cd.Cin, "bn_ab must be [Cin,2]");
+    abp = ab.data_ptr<float>();
+  }
 
   if ((cd.OW % 8) == 0 && cd.Cin >= 16 && cd.OH * cd.OW >= 64 &&
       (stride == 1 || stride == 2)) {
@@ -1017,14 +1063,34 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
       hipLaunchKernelGGL(kern, grid, dim3(256), 0, current_stream(),
                          reinterpret_cast<const __bf16*>(xc.data_ptr()),
                          reinterpret_cast<const __bf16*>(g.data_ptr()),
+                         dw.data_ptr<float>(), cd, nchunks, zstride, abp);
+    };
+    if (fuse) {
+      if (stride == 1) {
+        if (chunk == 64) L(conv3d_wgrad_s1_kernel<8, 1, 64, true>);
+        else if (OWT == 32) L(conv3d_wgrad_s1_kernel<32, 1, 128, true>);
+        else if (OWT == 16) L(conv3d_wgrad_s1_kernel<16, 1, 128, true>);
+        else L(conv3d_wgrad_s1_kernel<8, 1, 128, true>);
+      } else {
+        if (chunk == 64) L(conv3d_wgrad_s1_kernel<8, 2, 64, true>);
+        else if (OWT == 32) L(conv3d_wgrad_s1_kernel<32, 2, 128, true>);
+        else if (OWT == 16) L(conv3d_wgrad_s1_kernel<16, 2, 128, true>);
+        else L(conv3d_wgrad_s1_kernel<8, 2, 128, true>);
+      }
+      return dw.view({cd.Cout, cd.Cin, 3, 3, 3});
+    }
+    auto LDB = [&](auto kern) {
+      hipLaunchKernelGGL(kern, grid, dim3(256), 0, current_stream(),
+                         reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                         reinterpret_cast<const __bf16*>(g.data_ptr()),
                          dw.data_ptr<float>(), cd, nchunks, zstride);
     };
     if (stride == 1 && variant == 1) {
       // experimental double-buffered instances
-      if (chunk == 64) L(conv3d_wgrad_s1_db_kernel<8, 1, 64>);
-      else if (OWT == 32) L(conv3d_wgrad_s1_db_kernel<32, 1>);
-      else if (OWT == 16) L(conv3d_wgrad_s1_db_kernel<16, 1>);
-      else L(conv3d_wgrad_s1_db_kernel<8, 1>);
+      if (chunk == 64) LDB(conv3d_wgrad_s1_db_kernel<8, 1, 64>);
+      else if (OWT == 32) LDB(conv3d_wgrad_s1_db_kernel<32, 1>);
+      else if (OWT == 16) LDB(conv3d_wgrad_s1_db_kernel<16, 1>);
+      else LDB(conv3d_wgrad_s1_db_kernel<8, 1>);
     } else if (stride == 1) {
       if (chunk == 64) L(conv3d_wgrad_s1_kernel<8, 1, 64>);
       else if (OWT == 32) L(conv3d_wgrad_s1_kernel<32, 1>);
@@ -1047,11 +1113,14 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
   chunk = ((chunk + 127) / 128) * 128;
   int64_t nchunks = (M + chunk - 1) / chunk;
   dim3 grid((cd.Cout + 31) / 32, (K + 31) / 32, (unsigned)nchunks);
-  hipLaunchKernelGGL(conv3d_wgrad_kernel, grid, dim3(256), 0,
-                     current_stream(),
-                     reinterpret_cast<const __bf16*>(xc.data_ptr()),
-                     reinterpret_cast<const __bf16*>(g.data_ptr()),
-                     dw.data_ptr<float>(), cd, M, K, chunk);
+  auto LSK = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, current_stream(),
+                       reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                       reinterpret_cast<const __bf16*>(g.data_ptr()),
+                       dw.data_ptr<float>(), cd, M, K, chunk, abp);
+  };
+  if (fuse) LSK(conv3d_wgrad_kernel<true>);
+  else LSK(conv3d_wgrad_kernel<false>);
   return dw.view({cd.Cout, cd.Cin, 3, 3, 3});
 }
 

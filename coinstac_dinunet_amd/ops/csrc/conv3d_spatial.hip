@@ -33,11 +33,17 @@ struct SpDims {
 #define LDA_PAD 8
 #endif
 
+// FUSE_BN: normalize-on-load — the staged input is the PREVIOUS block's
+// raw conv output; z = relu(a[c]*x + b[c]) is applied per element during
+// staging (a = gamma*rstd, b = beta - mean*a, exactly bn_normalize's
+// folding) so the normalized tensor never exists in HBM. Padding halo
+// stays 0 (conv pads the BN OUTPUT with zeros).
 template <int OWT, int STRIDE, int CTILE,
-          int CHUNK = (STRIDE == 1 ? 256 : 128)>
+          int CHUNK = (STRIDE == 1 ? 256 : 128), bool FUSE_BN = false>
 __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
     const __bf16* __restrict__ in, const __bf16* __restrict__ wb,
-    __bf16* __restrict__ out, SpDims sd, int64_t nchunks) {
+    __bf16* __restrict__ out, SpDims sd, int64_t nchunks,
+    const float* __restrict__ bn_ab = nullptr) {
   constexpr int OHT = CHUNK / OWT;
   constexpr int IW = STRIDE * OWT;                  // staged interior width
   constexpr int W2 = IW + (STRIDE == 1 ? 4 : 2);
@@ -108,18 +114,24 @@ __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
       }
       const __bf16* src = in + in_n + ((int64_t)ch * sd.D + id) * HW +
                           (int64_t)ih * sd.W;
+      float a_c = 1.f, b_c = 0.f;
+      if (FUSE_BN) { a_c = bn_ab[ch * 2]; b_c = bn_ab[ch * 2 + 1]; }
+      auto tx = [&](__bf16 v) -> __bf16 {
+        if (!FUSE_BN) return v;
+        return (__bf16)fmaxf(a_c * (float)v + b_c, 0.f);
+      };
       const int iw0 = STRIDE * ow0;
-      dst[0] = (iw0 > 0) ? src[iw0 - 1] : (__bf16)0.f;
+      dst[0] = (iw0 > 0) ? tx(src[iw0 - 1]) : (__bf16)0.f;
 #pragma unroll
       for (int v = 0; v < IW / 8; ++v) {
         bf16x8 vec = *reinterpret_cast<const bf16x8*>(src + iw0 + v * 8);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) dst[1 + v * 8 + j] = vec[j];
+        for (int j = 0; j < 8; ++j) dst[1 + v * 8 + j] = tx(vec[j]);
       }
 #pragma unroll
       for (int e = 0; e < W2 - IW - 1; ++e) {
         const int iw = iw0 + IW + e;
-        dst[1 + IW + e] = (iw < sd.W) ? src[iw] : (__bf16)0.f;
+        dst[1 + IW + e] = (iw < sd.W) ? tx(src[iw]) : (__bf16)0.f;
       }
     }
     __syncthreads();
@@ -366,7 +378,7 @@ static torch::Tensor prep_wb(torch::Tensor w_flat2d, int KCH, int ctile) {
 
 static void launch_spatial(torch::Tensor in, torch::Tensor wb,
                            torch::Tensor out, SpDims sd, int stride,
-                           int ctile = 0) {
+                           int ctile = 0, const float* bn_ab = nullptr) {
   int OWT = sd.TW % 32 == 0 ? 32 : (sd.TW % 16 == 0 ? 16 : 8);
   int chunk = stride == 1 ? 256 : 128;
   // small images (d8-class): 64-position chunks keep the grid dense
@@ -383,15 +395,40 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
   const __bf16* ip = reinterpret_cast<const __bf16*>(in.data_ptr());
   const __bf16* wp = reinterpret_cast<const __bf16*>(wb.data_ptr());
   __bf16* op = reinterpret_cast<__bf16*>(out.data_ptr());
-  auto L = [&](auto kern) {
+  // db kernel has no bn_ab parameter; the regular kernel always takes one
+  // (hipLaunchKernelGGL cannot use default arguments)
+  auto LDB = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, dim3(256), 0, s, ip, wp, op, sd, nchunks);
   };
+  auto L = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, s, ip, wp, op, sd, nchunks,
+                       (const float*)nullptr);
+  };
+  auto LF = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, s, ip, wp, op, sd, nchunks,
+                       bn_ab);
+  };
+  if (bn_ab != nullptr) {
+    // fused normalize-on-load instances (default tilings only)
+    if (stride == 1) {
+      if (chunk == 64) LF(conv3d_spatial_kernel<8, 1, 32, 64, true>);
+      else if (OWT == 32) LF(conv3d_spatial_kernel<32, 1, 32, 256, true>);
+      else if (OWT == 16) LF(conv3d_spatial_kernel<16, 1, 32, 256, true>);
+      else LF(conv3d_spatial_kernel<8, 1, 32, 256, true>);
+    } else {
+      if (chunk == 64) LF(conv3d_spatial_kernel<8, 2, 16, 64, true>);
+      else if (OWT == 32) LF(conv3d_spatial_kernel<32, 2, 16, 128, true>);
+      else if (OWT == 16) LF(conv3d_spatial_kernel<16, 2, 16, 128, true>);
+      else LF(conv3d_spatial_kernel<8, 2, 16, 128, true>);
+    }
+    return;
+  }
   if (stride == 1 && ctile == 16) {
     // experimental double-buffered CTILE=16 instances (ctile_opt=16)
-    if (chunk == 64) L(conv3d_spatial_db_kernel<8, 1, 16, 64>);
-    else if (OWT == 32) L(conv3d_spatial_db_kernel<32, 1, 16>);
-    else if (OWT == 16) L(conv3d_spatial_db_kernel<16, 1, 16>);
-    else L(conv3d_spatial_db_kernel<8, 1, 16>);
+    if (chunk == 64) LDB(conv3d_spatial_db_kernel<8, 1, 16, 64>);
+    else if (OWT == 32) LDB(conv3d_spatial_db_kernel<32, 1, 16>);
+    else if (OWT == 16) LDB(conv3d_spatial_db_kernel<16, 1, 16>);
+    else LDB(conv3d_spatial_db_kernel<8, 1, 16>);
   } else if (stride == 1 && ctile == 1) {
     // single-channel (first-layer) instances: one 32-k-step covers the
     // whole 27-tap K, slab is [1][3][H2][W2]
@@ -413,12 +450,20 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
 }
 
 torch::Tensor conv3d_fwd_spatial(torch::Tensor x, torch::Tensor w,
-                                 int64_t stride, int64_t ctile_opt) {
+                                 int64_t stride, int64_t ctile_opt,
+                                 torch::Tensor bn_ab) {
   CHECK_GPU(x);
   auto xc = x.contiguous();
   auto wc = w.to(torch::kBFloat16).contiguous();
   TORCH_CHECK(xc.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(stride == 1 || stride == 2);
+  const bool fuse = bn_ab.defined() && bn_ab.numel() > 0;
+  torch::Tensor ab;
+  if (fuse) {
+    TORCH_CHECK(ctile_opt == 0, "fused BN requires default tiling");
+    ab = bn_ab.to(torch::kFloat32).contiguous();
+    TORCH_CHECK(ab.numel() == 2 * x.size(1), "bn_ab must be [Cin,2]");
+  }
   SpDims sd;
   sd.N = (int)xc.size(0); sd.KCH = (int)xc.size(1);
   sd.D = (int)xc.size(2); sd.H = (int)xc.size(3); sd.W = (int)xc.size(4);
@@ -437,7 +482,8 @@ torch::Tensor conv3d_fwd_spatial(torch::Tensor x, torch::Tensor w,
   sd.Kpad = (int)wb.size(1);
   auto out = torch::empty({sd.N, sd.NCOL, sd.TD, sd.TH, sd.TW},
                           xc.options());
-  launch_spatial(xc, wb, out, sd, (int)stride, ctile);
+  launch_spatial(xc, wb, out, sd, (int)stride, ctile,
+                 fuse ? ab.data_ptr<float>() : nullptr);
   return out;
 }
 

@@ -1,0 +1,129 @@
+"""Fused BN->conv (normalize-on-load) vs the unfused kernel path and the
+torch fp32 reference (MI355X).
+
+The fused path computes the SAME fp32 affine and the same bf16 rounding
+as bn_normalize_kernel, just inside the next conv's staging loop — so
+fused and unfused forward should agree to bf16 rounding noise, and both
+should track torch fp32 within accumulated-conv tolerance.
+"""
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from coinstac_dinunet_amd import ops
+    from coinstac_dinunet_amd.ops.bnorm import OpsBatchNorm3d
+    from coinstac_dinunet_amd.ops.conv import (OpsConv3d, conv_bn3d)
+    C = ops.require_native()
+
+
+@pytest.fixture(scope='module')
+def dev():
+    assert torch.cuda.is_available()
+    return torch.device('cuda:0')
+
+
+CASES = [
+    # (N, Cin, Cout, D, H, W, stride) — Cin = channels of the RAW input
+    (2, 32, 32, 16, 16, 16, 1),    # spatial fwd + spatial wgrad
+    (2, 32, 64, 16, 16, 16, 2),    # stride-2 spatial
+    (1, 32, 32, 6, 6, 6, 1),       # small => igemm fwd + splitK wgrad
+    (1, 64, 32, 8, 8, 8, 2),       # stride-2 small
+]
+
+
+def _pair(Cin, Cout, stride, dev, seed):
+    torch.manual_seed(seed)
+    bn = OpsBatchNorm3d(Cin, relu=True).to(dev)
+    with torch.no_grad():
+        bn.weight.mul_(0.0).add_(torch.rand(Cin, device=dev) + 0.5)
+        bn.bias.add_(torch.randn(Cin, device=dev) * 0.1)
+    conv = OpsConv3d(Cin, Cout, 3, stride=stride, padding=1,
+                     bias=False).to(dev)
+    return bn, conv
+
+
+@pytest.mark.parametrize('case', CASES)
+def test_fused_forward_matches_unfused(dev, case):
+    N, Ci, Co, D, H, W, s = case
+    bn, conv = _pair(Ci, Co, s, dev, 51)
+    x = torch.randn(N, Ci, D, H, W, device=dev, dtype=torch.bfloat16)
+    bn.train(), conv.train()
+    fused = conv_bn3d(x, bn, conv)
+    # reset running stats so both paths update from the same state
+    bn.running_mean.zero_(), bn.running_var.fill_(1.0)
+    bn.num_batches_tracked.zero_()
+    unfused = conv(bn(x))
+    torch.testing.assert_close(fused.float(), unfused.float(),
+                               rtol=2e-2, atol=2e-2 * (Ci * 27) ** 0.5 * 0.3)
+
+
+@pytest.mark.parametrize('case', CASES)
+def test_fused_backward_matches_unfused(dev, case):
+    N, Ci, Co, D, H, W, s = case
+    bn, conv = _pair(Ci, Co, s, dev, 52)
+    x0 = torch.randn(N, Ci, D, H, W, device=dev, dtype=torch.bfloat16)
+    go = torch.randn(N, Co, (D + 2 - 3) // s + 1, (H + 2 - 3) // s + 1,
+                     (W + 2 - 3) // s + 1, device=dev,
+                     dtype=torch.bfloat16) * 0.1
+
+    def run(fused):
+        bn.running_mean.zero_(), bn.running_var.fill_(1.0)
+        x = x0.clone().requires_grad_(True)
+        for p in list(bn.parameters()) + list(conv.parameters()):
+            p.grad = None
+        out = conv_bn3d(x, bn, conv) if fused else conv(bn(x))
+        out.backward(go)
+        return (x.grad.float().clone(), bn.weight.grad.clone(),
+                bn.bias.grad.clone(), conv.weight.grad.clone())
+
+    gx_f, gg_f, gb_f, gw_f = run(True)
+    gx_u, gg_u, gb_u, gw_u = run(False)
+    m = (N * D * H * W) ** 0.5
+    torch.testing.assert_close(gx_f, gx_u, rtol=5e-2, atol=2e-2)
+    torch.testing.assert_close(gg_f, gg_u, rtol=5e-2, atol=5e-2 * m * 0.1)
+    torch.testing.assert_close(gb_f, gb_u, rtol=5e-2, atol=5e-2 * m * 0.1)
+    torch.testing.assert_close(gw_f, gw_u, rtol=5e-2, atol=5e-2 * m * 0.1)
+
+
+def test_fused_forward_vs_torch_fp32(dev):
+    """End-to-end check against a pure fp32 torch bn+relu+conv."""
+    N, Ci, Co, D, H, W, s = 2, 32, 32, 16, 16, 16, 1
+    bn, conv = _pair(Ci, Co, s, dev, 53)
+    bn.train(), conv.train()
+    x = torch.randn(N, Ci, D, H, W, device=dev)
+    fused = conv_bn3d(x.bfloat16(), bn, conv)
+    xf = x.float()
+    mu = xf.mean(dim=(0, 2, 3, 4), keepdim=True)
+    var = xf.var(dim=(0, 2, 3, 4), unbiased=False, keepdim=True)
+    z = F.relu((xf - mu) * torch.rsqrt(var + bn.eps)
+               * bn.weight.view(1, -1, 1, 1, 1)
+               + bn.bias.view(1, -1, 1, 1, 1))
+    ref = F.conv3d(z, conv.weight.float(), stride=s, padding=1)
+    torch.testing.assert_close(fused.float(), ref, rtol=5e-2,
+                               atol=5e-2 * (Ci * 27) ** 0.5 * 0.3)
+
+
+def test_vbmnet_fused_chain_trains(dev):
+    """Whole-model fused chain: loss decreases and all grads flow."""
+    from coinstac_dinunet_amd.models.vbm import VBMNet
+    torch.manual_seed(9)
+    net = VBMNet(in_channels=1, num_class=2, widths=(16, 32)).to(dev)
+    x = torch.randn(8, 1, 16, 16, 16, device=dev)
+    y = (torch.arange(8) % 2).to(dev)
+    x[y == 1] += 0.4
+    opt = torch.optim.Adam(net.parameters(), lr=2e-3)
+    losses = []
+    for _ in range(30):
+        opt.zero_grad(set_to_none=True)
+        with torch.autocast('cuda', dtype=torch.bfloat16):
+            out = net(x)
+        loss = ops.cross_entropy(out.float(), y)
+        loss.backward()
+        opt.step()
+        losses.append(float(loss.detach()))
+    assert losses[-1] < losses[0] * 0.7, losses[::6]
+    for n, p in net.named_parameters():
+        assert p.grad is not None, n
