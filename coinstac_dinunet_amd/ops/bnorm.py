@@ -40,17 +40,13 @@ class _BN3dFn(torch.autograd.Function):
                 dbeta.to(beta.dtype), None, None, None, None, None)
 
 
-class OpsBatchNorm3d(nn.BatchNorm3d):
-    """BatchNorm3d with optional fused ReLU; HIP kernels on GPU."""
+class _OpsBNMixin:
+    """Shared HIP-kernel forward for the 2D/3D BatchNorm modules — the
+    streaming kernels only see [N, C, spatial] and are rank-agnostic."""
 
-    def __init__(self, num_features, eps=1e-5, momentum=0.1, relu=False,
-                 **kw):
-        super().__init__(num_features, eps=eps, momentum=momentum, **kw)
-        self.relu = relu
-
-    def forward(self, x):
+    def _ops_forward(self, x, super_forward):
         if not (x.is_cuda and native_available()):
-            y = super().forward(x)
+            y = super_forward(x)
             return F.relu(y, inplace=True) if self.relu else y
         if self.training:
             if self.num_batches_tracked is not None:
@@ -62,3 +58,28 @@ class OpsBatchNorm3d(nn.BatchNorm3d):
         return C.bn3d_infer(x.to(torch.bfloat16), self.weight, self.bias,
                             self.running_mean, self.running_var, self.eps,
                             self.relu)
+
+
+class OpsBatchNorm3d(_OpsBNMixin, nn.BatchNorm3d):
+    """BatchNorm3d with optional fused ReLU; HIP kernels on GPU."""
+
+    def __init__(self, num_features, eps=1e-5, momentum=0.1, relu=False,
+                 **kw):
+        super().__init__(num_features, eps=eps, momentum=momentum, **kw)
+        self.relu = relu
+
+    def forward(self, x):
+        return self._ops_forward(x, super().forward)
+
+
+class OpsBatchNorm2d(_OpsBNMixin, nn.BatchNorm2d):
+    """BatchNorm2d with optional fused ReLU on the same streaming kernels
+    (ResNet-18 config)."""
+
+    def __init__(self, num_features, eps=1e-5, momentum=0.1, relu=False,
+                 **kw):
+        super().__init__(num_features, eps=eps, momentum=momentum, **kw)
+        self.relu = relu
+
+    def forward(self, x):
+        return self._ops_forward(x, super().forward)

@@ -236,6 +236,68 @@ def conv3d(x, weight, bias=None, stride=1):
     return F.conv3d(x, weight, bias, stride=stride, padding=1)
 
 
+class _Conv2dFn(torch.autograd.Function):
+    """3x3/pad-1/stride-{1,2} Conv2d on the 2D implicit-GEMM MFMA kernels
+    (conv2d.hip) — the ResNet-18 hot path."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, stride):
+        C = require_native()
+        xb = x.to(torch.bfloat16)
+        wb = weight.to(torch.bfloat16)
+        out = C.conv2d_fwd(xb, wb, stride)
+        if bias is not None:
+            out = out + bias.to(out.dtype).view(1, -1, 1, 1)
+        ctx.save_for_backward(xb, wb)
+        ctx.stride = stride
+        ctx.has_bias = bias is not None
+        ctx.in_dtype = x.dtype
+        ctx.w_dtype = weight.dtype
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        C = require_native()
+        xb, wb = ctx.saved_tensors
+        go = grad_out.to(torch.bfloat16).contiguous()
+        gx = gw = gb = None
+        if ctx.needs_input_grad[0]:
+            gx = C.conv2d_dgrad(go, wb, list(xb.shape),
+                                ctx.stride).to(ctx.in_dtype)
+        if ctx.needs_input_grad[1]:
+            gw = C.conv2d_wgrad(xb, go, ctx.stride).to(ctx.w_dtype)
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            gb = C.channel_sum(go)
+        return gx, gw, gb, None
+
+
+class OpsConv2d(nn.Conv2d):
+    """nn.Conv2d on the in-tree MFMA kernels where the family allows.
+
+    3x3/pad-1/stride-{1,2} -> 2D implicit-GEMM kernels; 1x1 -> the
+    (rank-agnostic) pointwise kernels, with stride-2 1x1 downsamples
+    handled by a strided view ahead of the pointwise GEMM. Anything else
+    (e.g. the 7x7 stem) falls through to the library conv.
+    """
+
+    def forward(self, x):
+        if x.is_cuda and native_available() and self.dilation == (1, 1) \
+                and self.groups == 1:
+            if (self.kernel_size == (3, 3) and self.padding == (1, 1)
+                    and self.stride[0] in (1, 2)
+                    and self.stride[0] == self.stride[1]):
+                return _Conv2dFn.apply(x, self.weight, self.bias,
+                                       int(self.stride[0]))
+            if self.kernel_size == (1, 1) and self.padding == (0, 0):
+                if self.stride != (1, 1):
+                    x = x[:, :, ::self.stride[0], ::self.stride[1]] \
+                        .contiguous()
+                return _ConvPw3dFn.apply(x, self.weight, self.bias)
+        if x.is_cuda and x.dtype != self.weight.dtype:
+            x = x.to(self.weight.dtype)
+        return super().forward(x)
+
+
 class OpsConv3d(nn.Conv3d):
     """nn.Conv3d that runs the MFMA implicit-GEMM kernels on GPU.
 

@@ -55,6 +55,13 @@ torch::Tensor conv3d_dgrad_spatial(torch::Tensor go, torch::Tensor w,
                                    std::vector<int64_t> in_shape);
 torch::Tensor conv3d_dgrad_s2_spatial(torch::Tensor go, torch::Tensor w,
                                       std::vector<int64_t> in_shape);
+// conv2d.hip
+torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
+                         torch::Tensor bn_ab);
+torch::Tensor conv2d_dgrad(torch::Tensor go, torch::Tensor w,
+                           std::vector<int64_t> in_shape, int64_t stride);
+torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor go,
+                           int64_t stride, torch::Tensor bn_ab);
 // rankdad.hip
 std::vector<torch::Tensor> power_iter_bc(torch::Tensor B, torch::Tensor C,
                                          int64_t rank, int64_t iters,
@@ -106,6 +113,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("ctile_opt") = 0, py::arg("bn_ab") = torch::Tensor());
   m.def("conv3d_dgrad_spatial", &conv3d_dgrad_spatial);
   m.def("conv3d_dgrad_s2_spatial", &conv3d_dgrad_s2_spatial);
+  m.def("conv2d_fwd", &conv2d_fwd, py::arg("x"), py::arg("w"),
+        py::arg("stride"), py::arg("bn_ab") = torch::Tensor());
+  m.def("conv2d_dgrad", &conv2d_dgrad);
+  m.def("conv2d_wgrad", &conv2d_wgrad, py::arg("x"), py::arg("go"),
+        py::arg("stride"), py::arg("bn_ab") = torch::Tensor());
   m.def("power_iter_bc", &power_iter_bc);
   m.def("rowsum", &rowsum);
   m.def("conv3d_pw_fwd", &conv3d_pw_fwd);
