@@ -64,7 +64,28 @@ def build_model(args, device):
         net = FreeSurferMLP(in_features=66, num_class=2)
         data = torch.randn(args.batch, 66)
     else:
-        net = ResNet18(in_channels=3, num_class=10)
+        if args.stock:
+            # library-op ResNet-18 (MIOpen convs, eager BN) for comparison
+            import torch.nn as _nn
+            import coinstac_dinunet_amd.models.resnet as _rn
+
+            class _StockBN(_nn.BatchNorm2d):
+                def __init__(self, c, relu=False, **kw):
+                    super().__init__(c, **kw)
+                    self._relu = relu
+
+                def forward(self, x):
+                    y = super().forward(x)
+                    return _nn.functional.relu(y) if self._relu else y
+
+            o_conv, o_bn = _rn.OpsConv2d, _rn.OpsBatchNorm2d
+            _rn.OpsConv2d, _rn.OpsBatchNorm2d = _nn.Conv2d, _StockBN
+            try:
+                net = ResNet18(in_channels=3, num_class=10)
+            finally:
+                _rn.OpsConv2d, _rn.OpsBatchNorm2d = o_conv, o_bn
+        else:
+            net = ResNet18(in_channels=3, num_class=10)
         data = torch.randn(args.batch, 3, 224, 224)
     labels = torch.randint(0, 2, (args.batch,))
     return net.to(device), data.to(device), labels.to(device)

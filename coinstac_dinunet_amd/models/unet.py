@@ -12,17 +12,24 @@ conv — both stay inside the supported kernel family, no transposed-conv
 kernel needed. Skip connections go through safe_concat (center-crop) so
 odd input sizes work.
 """
+import os
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops.bnorm import OpsBatchNorm3d
-from ..ops.conv import OpsConv3d
+from ..ops.conv import OpsConv3d, can_fuse_bn_conv, conv_bn3d
 from ..utils.tensorutils import safe_concat
+
+_FUSE_BN = os.environ.get('COINN_FUSE_BN', '1') == '1'
 
 
 class _Block(nn.Module):
-    """conv(3x3x3) -> fused BN+ReLU, twice; optional stride-2 entry."""
+    """conv(3x3x3) -> fused BN+ReLU, twice; optional stride-2 entry.
+
+    On GPU the inner (b1 -> c2) pair runs as one normalize-on-load fused
+    conv (conv_bn3d), so b1's normalized activation never hits HBM."""
 
     def __init__(self, cin, cout, stride=1):
         super().__init__()
@@ -33,6 +40,8 @@ class _Block(nn.Module):
         self.b2 = OpsBatchNorm3d(cout, relu=True)
 
     def forward(self, x):
+        if _FUSE_BN and can_fuse_bn_conv(self.b1, self.c2, x):
+            return self.b2(conv_bn3d(self.c1(x), self.b1, self.c2))
         return self.b2(self.c2(self.b1(self.c1(x))))
 
 
