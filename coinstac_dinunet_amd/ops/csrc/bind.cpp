@@ -41,27 +41,27 @@ void gram_schmidt(torch::Tensor m, double eps);
 // conv3d.hip
 torch::Tensor mfma_probe_gemm(torch::Tensor A, torch::Tensor B);
 torch::Tensor conv3d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
-                         torch::Tensor bn_ab);
+                         c10::optional<torch::Tensor> bn_ab);
 torch::Tensor conv3d_dgrad(torch::Tensor go, torch::Tensor w,
                            std::vector<int64_t> in_shape, int64_t stride);
 torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go, int64_t stride,
-                           int64_t variant, torch::Tensor bn_ab);
+                           int64_t variant, c10::optional<torch::Tensor> bn_ab);
 torch::Tensor channel_sum(torch::Tensor go);
 // conv3d_spatial.hip
 torch::Tensor conv3d_fwd_spatial(torch::Tensor x, torch::Tensor w,
                                  int64_t stride, int64_t ctile_opt,
-                                 torch::Tensor bn_ab);
+                                 c10::optional<torch::Tensor> bn_ab);
 torch::Tensor conv3d_dgrad_spatial(torch::Tensor go, torch::Tensor w,
                                    std::vector<int64_t> in_shape);
 torch::Tensor conv3d_dgrad_s2_spatial(torch::Tensor go, torch::Tensor w,
                                       std::vector<int64_t> in_shape);
 // conv2d.hip
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
-                         torch::Tensor bn_ab);
+                         c10::optional<torch::Tensor> bn_ab);
 torch::Tensor conv2d_dgrad(torch::Tensor go, torch::Tensor w,
                            std::vector<int64_t> in_shape, int64_t stride);
 torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor go,
-                           int64_t stride, torch::Tensor bn_ab);
+                           int64_t stride, c10::optional<torch::Tensor> bn_ab);
 // rankdad.hip
 std::vector<torch::Tensor> power_iter_bc(torch::Tensor B, torch::Tensor C,
                                          int64_t rank, int64_t iters,
@@ -102,22 +102,22 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gram_schmidt", &gram_schmidt);
   m.def("mfma_probe_gemm", &mfma_probe_gemm);
   m.def("conv3d_fwd", &conv3d_fwd, py::arg("x"), py::arg("w"),
-        py::arg("stride"), py::arg("bn_ab") = torch::Tensor());
+        py::arg("stride"), py::arg("bn_ab") = py::none());
   m.def("conv3d_dgrad", &conv3d_dgrad);
   m.def("conv3d_wgrad", &conv3d_wgrad, py::arg("x"), py::arg("go"),
         py::arg("stride"), py::arg("variant") = 0,
-        py::arg("bn_ab") = torch::Tensor());
+        py::arg("bn_ab") = py::none());
   m.def("channel_sum", &channel_sum);
   m.def("conv3d_fwd_spatial", &conv3d_fwd_spatial,
         py::arg("x"), py::arg("w"), py::arg("stride"),
-        py::arg("ctile_opt") = 0, py::arg("bn_ab") = torch::Tensor());
+        py::arg("ctile_opt") = 0, py::arg("bn_ab") = py::none());
   m.def("conv3d_dgrad_spatial", &conv3d_dgrad_spatial);
   m.def("conv3d_dgrad_s2_spatial", &conv3d_dgrad_s2_spatial);
   m.def("conv2d_fwd", &conv2d_fwd, py::arg("x"), py::arg("w"),
-        py::arg("stride"), py::arg("bn_ab") = torch::Tensor());
+        py::arg("stride"), py::arg("bn_ab") = py::none());
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad, py::arg("x"), py::arg("go"),
-        py::arg("stride"), py::arg("bn_ab") = torch::Tensor());
+        py::arg("stride"), py::arg("bn_ab") = py::none());
   m.def("power_iter_bc", &power_iter_bc);
   m.def("rowsum", &rowsum);
   m.def("conv3d_pw_fwd", &conv3d_pw_fwd);

@@ -308,7 +308,8 @@ static Conv2dDims make_dims2(const torch::Tensor& x, const torch::Tensor& w,
 }
 
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
-                         torch::Tensor bn_ab) {
+                         c10::optional<torch::Tensor> bn_ab_opt) {
+  torch::Tensor bn_ab = bn_ab_opt.value_or(torch::Tensor());
   CHECK_GPU(x);
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "conv2d_fwd wants bf16");
   auto xc = x.contiguous();
@@ -376,7 +377,8 @@ torch::Tensor conv2d_dgrad(torch::Tensor go, torch::Tensor w,
 }
 
 torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor go,
-                           int64_t stride, torch::Tensor bn_ab) {
+                           int64_t stride, c10::optional<torch::Tensor> bn_ab_opt) {
+  torch::Tensor bn_ab = bn_ab_opt.value_or(torch::Tensor());
   CHECK_GPU(x);
   auto xc = x.to(torch::kBFloat16).contiguous();
   auto g = go.to(torch::kBFloat16).contiguous();

@@ -950,7 +950,8 @@ static ConvDims make_dims(const torch::Tensor& x, const torch::Tensor& w,
 }
 
 torch::Tensor conv3d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
-                         torch::Tensor bn_ab) {
+                         c10::optional<torch::Tensor> bn_ab_opt) {
+  torch::Tensor bn_ab = bn_ab_opt.value_or(torch::Tensor());
   CHECK_GPU(x);
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "conv3d_fwd wants bf16");
   auto xc = x.contiguous();
@@ -1017,7 +1018,8 @@ torch::Tensor conv3d_dgrad(torch::Tensor go, torch::Tensor w,
 
 torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
                            int64_t stride, int64_t variant,
-                           torch::Tensor bn_ab) {
+                           c10::optional<torch::Tensor> bn_ab_opt) {
+  torch::Tensor bn_ab = bn_ab_opt.value_or(torch::Tensor());
   CHECK_GPU(x);
   auto xc = x.to(torch::kBFloat16).contiguous();
   auto g = go.to(torch::kBFloat16).contiguous();

@@ -451,7 +451,8 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
 
 torch::Tensor conv3d_fwd_spatial(torch::Tensor x, torch::Tensor w,
                                  int64_t stride, int64_t ctile_opt,
-                                 torch::Tensor bn_ab) {
+                                 c10::optional<torch::Tensor> bn_ab_opt) {
+  torch::Tensor bn_ab = bn_ab_opt.value_or(torch::Tensor());
   CHECK_GPU(x);
   auto xc = x.contiguous();
   auto wc = w.to(torch::kBFloat16).contiguous();
