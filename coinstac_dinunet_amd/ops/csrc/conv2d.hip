@@ -185,6 +185,341 @@ __global__ __launch_bounds__(256) void conv2d_igemm_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Spatial-slab tap-reuse kernels (the conv3d_spatial.hip design with the
+// depth axis dropped): a block stages one input slab [CTILE][H2][W2] per
+// channel tile and computes the whole 9-tap K loop from it through a
+// precomputed u16 k->offset table. Unlike the 3D kernels, staging is
+// CLAMPED per row (vector loads only when the full interior is in
+// bounds), so OWT need not divide the output width — ResNet's 28/14/7
+// grids route here with partial edge tiles instead of falling back.
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(8))) unsigned short u16x8_2;
+
+struct Sp2Dims {
+  int N, KCH;        // input-side channels (fwd: Cin; dgrad: Cout)
+  int H, W;          // input-side spatial
+  int NCOL;          // output-side channels
+  int TH, TW;        // output-side spatial
+  int Kpad;          // row length of WB = kts * KT_PAD
+};
+
+template <int OWT, int STRIDE, int CTILE, int CHUNK, bool FUSE_BN = false>
+__global__ __launch_bounds__(256) void conv2d_spatial_kernel(
+    const __bf16* __restrict__ in, const __bf16* __restrict__ wb,
+    __bf16* __restrict__ out, Sp2Dims sd, int64_t nchunks,
+    const float* __restrict__ bn_ab) {
+  constexpr int OHT = CHUNK / OWT;
+  constexpr int IW = STRIDE * OWT;
+  constexpr int W2 = IW + (STRIDE == 1 ? 4 : 2);
+  constexpr int H2 = STRIDE * (OHT - 1) + 3;
+  constexpr int MPW = CHUNK / 64;
+  static_assert(MPW >= 1, "chunk too small");
+  constexpr int KT_PAD = ((CTILE * 9 + 31) / 32) * 32;
+  __shared__ __bf16 sX[CTILE][H2][W2];
+  __shared__ unsigned short sKtab[KT_PAD + 8];
+
+  const int ncol0 = blockIdx.y * 32;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int row = lane & 15, kg = lane >> 4;
+
+  const int wtiles = (sd.TW + OWT - 1) / OWT;
+  const int htiles = (sd.TH + OHT - 1) / OHT;
+
+  int64_t t = blockIdx.x;
+  const int wt = (int)(t % wtiles);
+  t /= wtiles;
+  const int ht = (int)(t % htiles);
+  const int n = (int)(t / htiles);
+  const int oh0 = ht * OHT, ow0 = wt * OWT;
+
+  for (int k = tid; k < KT_PAD; k += 256) {
+    unsigned short off = 0;
+    if (k < CTILE * 9) {
+      const int cl = k / 9;
+      const int r9 = k - cl * 9;
+      const int b = r9 / 3, c = r9 % 3;
+      off = (unsigned short)((cl * H2 + b) * W2 + c);
+    }
+    sKtab[k] = off;
+  }
+
+  f32x4 acc[MPW][2];
+#pragma unroll
+  for (int i = 0; i < MPW; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int64_t HW = (int64_t)sd.H * sd.W;
+  const int64_t in_n = (int64_t)n * sd.KCH * HW;
+  const int kts = (sd.KCH + CTILE - 1) / CTILE;
+
+  for (int kt = 0; kt < kts; ++kt) {
+    constexpr int NROWS = CTILE * H2;
+    if (kt) __syncthreads();
+    for (int r = tid; r < NROWS; r += 256) {
+      const int hrow = r % H2;
+      const int c = r / H2;
+      const int ih = STRIDE * oh0 - 1 + hrow;
+      const int ch = kt * CTILE + c;
+      __bf16* dst = &sX[c][hrow][0];
+      const bool row_ok = (unsigned)ih < (unsigned)sd.H && ch < sd.KCH;
+      if (!row_ok) {
+#pragma unroll
+        for (int col = 0; col < W2; ++col) dst[col] = (__bf16)0.f;
+        continue;
+      }
+      const __bf16* src = in + in_n + (int64_t)ch * HW + (int64_t)ih * sd.W;
+      float a_c = 1.f, b_c = 0.f;
+      if (FUSE_BN) { a_c = bn_ab[ch * 2]; b_c = bn_ab[ch * 2 + 1]; }
+      auto tx = [&](__bf16 v) -> __bf16 {
+        if (!FUSE_BN) return v;
+        return (__bf16)fmaxf(a_c * (float)v + b_c, 0.f);
+      };
+      const int iw0 = STRIDE * ow0;
+      if (iw0 + IW <= sd.W) {  // interior tile: vector loads
+        dst[0] = (iw0 > 0) ? tx(src[iw0 - 1]) : (__bf16)0.f;
+#pragma unroll
+        for (int v = 0; v < IW / 8; ++v) {
+          bf16x8 vec = *reinterpret_cast<const bf16x8*>(src + iw0 + v * 8);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) dst[1 + v * 8 + j] = tx(vec[j]);
+        }
+#pragma unroll
+        for (int e = 0; e < W2 - IW - 1; ++e) {
+          const int iw = iw0 + IW + e;
+          dst[1 + IW + e] = (iw < sd.W) ? tx(src[iw]) : (__bf16)0.f;
+        }
+      } else {  // edge tile: clamped scalar loads
+#pragma unroll
+        for (int col = 0; col < W2; ++col) {
+          const int iw = iw0 - 1 + col;
+          dst[col] = ((unsigned)iw < (unsigned)sd.W) ? tx(src[iw])
+                                                     : (__bf16)0.f;
+        }
+      }
+    }
+    __syncthreads();
+
+    const int kbase_g = kt * KT_PAD;
+#pragma unroll 1
+    for (int ks = 0; ks < KT_PAD / 32; ++ks) {
+      bf16x8 afrag[MPW];
+      {
+        const int kb = ks * 32 + kg * 8;
+        const u16x8_2 kt8 = *reinterpret_cast<const u16x8_2*>(&sKtab[kb]);
+        const __bf16* slab = &sX[0][0][0];
+#pragma unroll
+        for (int i = 0; i < MPW; ++i) {
+          const int m = (wave * MPW + i) * 16 + row;
+          const int base = (STRIDE * (m / OWT)) * W2 + STRIDE * (m % OWT);
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            afrag[i][j] = slab[base + kt8[j]];
+        }
+      }
+      bf16x8 bfrag[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        const int col = ncol0 + i * 16 + row;
+        const int64_t off =
+            (int64_t)col * sd.Kpad + kbase_g + ks * 32 + kg * 8;
+        bfrag[i] = (col < sd.NCOL)
+                       ? *reinterpret_cast<const bf16x8*>(wb + off)
+                       : bf16x8{};
+      }
+#pragma unroll
+      for (int i = 0; i < MPW; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    }
+  }
+
+  const int64_t THW = (int64_t)sd.TH * sd.TW;
+  const int64_t out_n = (int64_t)n * sd.NCOL * THW;
+  const int ccol = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < MPW; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int col = ncol0 + j * 16 + ccol;
+      if (col >= sd.NCOL) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = (wave * MPW + i) * 16 + crow0 + r;
+        const int oh = oh0 + m / OWT;
+        const int ow = ow0 + m % OWT;
+        if (oh < sd.TH && ow < sd.TW)
+          out[out_n + (int64_t)col * THW + (int64_t)oh * sd.TW + ow] =
+              (__bf16)(acc[i][j][r]);
+      }
+    }
+  }
+}
+
+// WGRAD with tap reuse (2D): block stages one x slab [CT][H2][W2] + one go
+// tile [co][m] once and computes all 9 tap GEMMs from it; grid-strides
+// over (n, h-tile, w-tile) chunks; fp32 atomics fold partials into dw.
+template <int OWT, int STRIDE, int CHUNK = 128, bool FUSE_BN = false>
+__global__ __launch_bounds__(256) void conv2d_wgrad_sp_kernel(
+    const __bf16* __restrict__ x, const __bf16* __restrict__ go,
+    float* __restrict__ dw, Conv2dDims cd, int64_t nchunks, int64_t zstride,
+    const float* __restrict__ bn_ab) {
+  constexpr int OHT = CHUNK / OWT;
+  constexpr int IW = STRIDE * OWT;
+  constexpr int W2 = IW + (STRIDE == 1 ? 4 : 2);
+  constexpr int H2 = STRIDE * (OHT - 1) + 3;
+  constexpr int CT = STRIDE == 1 ? 32 : 16;
+  constexpr int COT = STRIDE == 1 ? 32 : 64;
+  __shared__ __bf16 sX[CT][H2][W2];
+  __shared__ __bf16 sGo[COT][CHUNK + LDA_PAD2];
+
+  const int co0 = blockIdx.x * COT;
+  const int ci0 = blockIdx.y * CT;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wi = (STRIDE == 1) ? (wave >> 1) : wave;
+  const int wj = (STRIDE == 1) ? (wave & 1) : 0;
+  const int row = lane & 15, kg = lane >> 4;
+
+  const int wtiles = (cd.OW + OWT - 1) / OWT;
+  const int htiles = (cd.OH + OHT - 1) / OHT;
+
+  f32x4 acc[9];
+#pragma unroll
+  for (int t = 0; t < 9; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
+
+  const int64_t HW = (int64_t)cd.H * cd.W;
+  const int64_t OHW = (int64_t)cd.OH * cd.OW;
+
+  for (int64_t z = blockIdx.z; z < nchunks; z += zstride) {
+    int64_t t = z;
+    const int wt = (int)(t % wtiles);
+    t /= wtiles;
+    const int ht = (int)(t % htiles);
+    const int n = (int)(t / htiles);
+    const int oh0 = ht * OHT, ow0 = wt * OWT;
+
+    constexpr int NXROWS = CT * H2;
+    const __bf16* xn = x + (int64_t)n * cd.Cin * HW;
+    for (int r = tid; r < NXROWS; r += 256) {
+      const int hrow = r % H2;
+      const int ci = r / H2;
+      const int ih = STRIDE * oh0 - 1 + hrow;
+      __bf16* dst = &sX[ci][hrow][0];
+      const bool row_ok = (unsigned)ih < (unsigned)cd.H &&
+                          (ci0 + ci) < cd.Cin;
+      if (!row_ok) {
+#pragma unroll
+        for (int col = 0; col < W2; ++col) dst[col] = (__bf16)0.f;
+        continue;
+      }
+      const __bf16* src = xn + (int64_t)(ci0 + ci) * HW +
+                          (int64_t)ih * cd.W;
+      float a_c = 1.f, b_c = 0.f;
+      if (FUSE_BN) {
+        a_c = bn_ab[(ci0 + ci) * 2];
+        b_c = bn_ab[(ci0 + ci) * 2 + 1];
+      }
+      auto tx = [&](__bf16 v) -> __bf16 {
+        if (!FUSE_BN) return v;
+        return (__bf16)fmaxf(a_c * (float)v + b_c, 0.f);
+      };
+      const int iw0 = STRIDE * ow0;
+      if (iw0 + IW <= cd.W) {
+        dst[0] = (iw0 > 0) ? tx(src[iw0 - 1]) : (__bf16)0.f;
+#pragma unroll
+        for (int v = 0; v < IW / 8; ++v) {
+          bf16x8 vec = *reinterpret_cast<const bf16x8*>(src + iw0 + v * 8);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) dst[1 + v * 8 + j] = tx(vec[j]);
+        }
+#pragma unroll
+        for (int e = 0; e < W2 - IW - 1; ++e) {
+          const int iw = iw0 + IW + e;
+          dst[1 + IW + e] = (iw < cd.W) ? tx(src[iw]) : (__bf16)0.f;
+        }
+      } else {
+#pragma unroll
+        for (int col = 0; col < W2; ++col) {
+          const int iw = iw0 - 1 + col;
+          dst[col] = ((unsigned)iw < (unsigned)cd.W) ? tx(src[iw])
+                                                     : (__bf16)0.f;
+        }
+      }
+    }
+    const __bf16* gon = go + (int64_t)n * cd.Cout * OHW;
+    for (int r = tid; r < COT * OHT; r += 256) {
+      const int oh_off = r % OHT;
+      const int co = r / OHT;
+      __bf16* dst = &sGo[co][oh_off * OWT];
+      const int oh = oh0 + oh_off;
+      if ((co0 + co) >= cd.Cout || oh >= cd.OH) {
+#pragma unroll
+        for (int j = 0; j < OWT; ++j) dst[j] = (__bf16)0.f;
+        continue;
+      }
+      const __bf16* src = gon + (int64_t)(co0 + co) * OHW +
+                          (int64_t)oh * cd.OW + ow0;
+      if (ow0 + OWT <= cd.OW) {
+#pragma unroll
+        for (int v = 0; v < OWT / 8; ++v)
+          *reinterpret_cast<bf16x8*>(dst + v * 8) =
+              *reinterpret_cast<const bf16x8*>(src + v * 8);
+      } else {
+#pragma unroll
+        for (int j = 0; j < OWT; ++j)
+          dst[j] = (ow0 + j < cd.OW) ? src[j] : (__bf16)0.f;
+      }
+    }
+    __syncthreads();
+
+#pragma unroll 1
+    for (int ms = 0; ms < CHUNK / 32; ++ms) {
+      bf16x8 afrag;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        afrag[j] = sGo[wi * 16 + row][ms * 32 + kg * 8 + j];
+      const int mbase = ms * 32 + kg * 8;
+      const int oh_off = mbase / OWT;
+      const int ow_off = mbase % OWT;
+#pragma unroll
+      for (int kh = 0; kh < 3; ++kh) {
+#pragma unroll
+        for (int kw = 0; kw < 3; ++kw) {
+          bf16x8 bfrag;
+          const __bf16* src = &sX[wj * 16 + row]
+                                 [STRIDE * oh_off + kh]
+                                 [STRIDE * ow_off + kw];
+#pragma unroll
+          for (int j = 0; j < 8; ++j) bfrag[j] = src[STRIDE * j];
+          acc[kh * 3 + kw] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag, bfrag, acc[kh * 3 + kw], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  const int K = cd.Cin * 9;
+  const int ccol = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+#pragma unroll 1
+  for (int tp = 0; tp < 9; ++tp) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int co = co0 + wi * 16 + crow0 + r;
+      const int ci = ci0 + wj * 16 + ccol;
+      if (co < cd.Cout && ci < cd.Cin)
+        atomicAdd(&dw[(int64_t)co * K + ci * 9 + tp], acc[tp][r]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // WGRAD: split-K implicit GEMM with fp32 atomics (structure of
 // conv3d.hip's conv3d_wgrad_kernel with the depth axis dropped).
 // ---------------------------------------------------------------------------
@@ -307,6 +642,72 @@ static Conv2dDims make_dims2(const torch::Tensor& x, const torch::Tensor& w,
   return cd;
 }
 
+// WB layout: [NCOL][kts * KT_PAD]; channel-tile block kt holds the CTILE*9
+// weights, zero-padded to a 32-multiple (CTILE=32 -> 288, no padding).
+static torch::Tensor prep_wb2(torch::Tensor w_flat2d, int KCH, int ctile) {
+  int ncol = (int)w_flat2d.size(0);
+  int kt_pad = (ctile * 9 + 31) / 32 * 32;
+  int kts = (KCH + ctile - 1) / ctile;
+  auto wb = torch::zeros({ncol, (int64_t)kts * kt_pad}, w_flat2d.options());
+  for (int kt = 0; kt < kts; ++kt) {
+    int64_t k0 = (int64_t)kt * ctile * 9;
+    int64_t klen = (int64_t)std::min(KCH - kt * ctile, ctile) * 9;
+    wb.narrow(1, (int64_t)kt * kt_pad, klen).copy_(
+        w_flat2d.narrow(1, k0, klen));
+  }
+  return wb;
+}
+
+static int pick_owt2(int tw) {
+  if (tw % 32 == 0) return 32;
+  if (tw % 16 == 0) return 16;
+  return 8;  // clamped staging: divisibility not required
+}
+
+static void launch_spatial2(torch::Tensor in, torch::Tensor wb,
+                            torch::Tensor out, Sp2Dims sd, int stride,
+                            const float* bn_ab) {
+  int OWT = pick_owt2(sd.TW);
+  int chunk = stride == 1 ? 256 : 128;
+  if (sd.TH * sd.TW < chunk) { chunk = 64; OWT = 8; }
+  int OHT = chunk / OWT;
+  int wtiles = (sd.TW + OWT - 1) / OWT;
+  int htiles = (sd.TH + OHT - 1) / OHT;
+  int64_t nchunks = (int64_t)sd.N * htiles * wtiles;
+  dim3 grid((unsigned)nchunks, (sd.NCOL + 31) / 32);
+  auto s = current_stream();
+  const __bf16* ip = reinterpret_cast<const __bf16*>(in.data_ptr());
+  const __bf16* wp = reinterpret_cast<const __bf16*>(wb.data_ptr());
+  __bf16* op = reinterpret_cast<__bf16*>(out.data_ptr());
+  auto L = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, s, ip, wp, op, sd, nchunks,
+                       bn_ab);
+  };
+  if (bn_ab != nullptr) {
+    if (stride == 1) {
+      if (chunk == 64) L(conv2d_spatial_kernel<8, 1, 32, 64, true>);
+      else if (OWT == 32) L(conv2d_spatial_kernel<32, 1, 32, 256, true>);
+      else if (OWT == 16) L(conv2d_spatial_kernel<16, 1, 32, 256, true>);
+      else L(conv2d_spatial_kernel<8, 1, 32, 256, true>);
+    } else {
+      if (chunk == 64) L(conv2d_spatial_kernel<8, 2, 16, 64, true>);
+      else if (OWT == 32) L(conv2d_spatial_kernel<32, 2, 16, 128, true>);
+      else if (OWT == 16) L(conv2d_spatial_kernel<16, 2, 16, 128, true>);
+      else L(conv2d_spatial_kernel<8, 2, 16, 128, true>);
+    }
+  } else if (stride == 1) {
+    if (chunk == 64) L(conv2d_spatial_kernel<8, 1, 32, 64>);
+    else if (OWT == 32) L(conv2d_spatial_kernel<32, 1, 32, 256>);
+    else if (OWT == 16) L(conv2d_spatial_kernel<16, 1, 32, 256>);
+    else L(conv2d_spatial_kernel<8, 1, 32, 256>);
+  } else {
+    if (chunk == 64) L(conv2d_spatial_kernel<8, 2, 16, 64>);
+    else if (OWT == 32) L(conv2d_spatial_kernel<32, 2, 16, 128>);
+    else if (OWT == 16) L(conv2d_spatial_kernel<16, 2, 16, 128>);
+    else L(conv2d_spatial_kernel<8, 2, 16, 128>);
+  }
+}
+
 torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
                          c10::optional<torch::Tensor> bn_ab_opt) {
   torch::Tensor bn_ab = bn_ab_opt.value_or(torch::Tensor());
@@ -327,6 +728,20 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
     abp = ab.data_ptr<float>();
   }
   auto out = torch::empty({cd.N, cd.Cout, cd.OH, cd.OW}, xc.options());
+
+  if (cd.Cin >= 16 && cd.OH * cd.OW >= 32) {
+    // spatial-slab tap-reuse path (clamped staging: any width)
+    Sp2Dims sd;
+    sd.N = cd.N; sd.KCH = cd.Cin; sd.H = cd.H; sd.W = cd.W;
+    sd.NCOL = cd.Cout; sd.TH = cd.OH; sd.TW = cd.OW;
+    int ctile = stride == 1 ? 32 : 16;
+    auto wb = prep_wb2(wc.reshape({sd.NCOL, (int64_t)sd.KCH * 9}), sd.KCH,
+                       ctile);
+    sd.Kpad = (int)wb.size(1);
+    launch_spatial2(xc, wb, out, sd, (int)stride, abp);
+    return out;
+  }
+
   int64_t M = (int64_t)cd.N * cd.OH * cd.OW;
   int K = cd.Cin * 9;
   dim3 grid((unsigned)((M + CBM2 - 1) / CBM2), (cd.Cout + CBN2 - 1) / CBN2);
@@ -361,6 +776,20 @@ torch::Tensor conv2d_dgrad(torch::Tensor go, torch::Tensor w,
   cd.OH = (int)g.size(2); cd.OW = (int)g.size(3);
   TORCH_CHECK(stride == 1 || stride == 2, "stride must be 1 or 2");
   auto dx = torch::empty(in_shape, g.options());
+
+  if (stride == 1 && cd.Cout >= 16 && cd.H * cd.W >= 32) {
+    // dgrad as spatial conv of go with tap-flipped transposed weights
+    Sp2Dims sd;
+    sd.N = cd.N; sd.KCH = cd.Cout; sd.H = cd.OH; sd.W = cd.OW;
+    sd.NCOL = cd.Cin; sd.TH = cd.H; sd.TW = cd.W;
+    auto wf = wc.reshape({cd.Cout, cd.Cin, 9}).flip(-1).permute({1, 0, 2})
+                  .reshape({cd.Cin, (int64_t)cd.Cout * 9}).contiguous();
+    auto wb = prep_wb2(wf, cd.Cout, 32);
+    sd.Kpad = (int)wb.size(1);
+    launch_spatial2(g, wb, dx, sd, 1, nullptr);
+    return dx;
+  }
+
   int64_t M = (int64_t)cd.N * cd.H * cd.W;
   int K = cd.Cout * 9;
   dim3 grid((unsigned)((M + CBM2 - 1) / CBM2), (cd.Cin + CBN2 - 1) / CBN2);
@@ -400,6 +829,53 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor go,
     TORCH_CHECK(ab.numel() == 2 * cd.Cin, "bn_ab must be [Cin,2]");
     abp = ab.data_ptr<float>();
   }
+
+  if (cd.Cin >= 16 && cd.OH * cd.OW >= 32 &&
+      (stride == 1 || stride == 2)) {
+    int OWT = pick_owt2(cd.OW);
+    int chunk = 128;
+    if (cd.OH * cd.OW < 128) { chunk = 64; OWT = 8; }
+    int wtiles = (cd.OW + OWT - 1) / OWT;
+    int OHT = chunk / OWT;
+    int htiles = (cd.OH + OHT - 1) / OHT;
+    int COT = stride == 1 ? 32 : 64, CT = stride == 1 ? 32 : 16;
+    int co_t = (cd.Cout + COT - 1) / COT, ci_t = (cd.Cin + CT - 1) / CT;
+    int64_t nchunks = (int64_t)cd.N * htiles * wtiles;
+    int64_t zstride = std::max<int64_t>(
+        1, std::min<int64_t>(nchunks, 768 / std::max(co_t * ci_t, 1)));
+    dim3 grid(co_t, ci_t, (unsigned)zstride);
+    auto L = [&](auto kern) {
+      hipLaunchKernelGGL(kern, grid, dim3(256), 0, current_stream(),
+                         reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                         reinterpret_cast<const __bf16*>(g.data_ptr()),
+                         dw.data_ptr<float>(), cd, nchunks, zstride, abp);
+    };
+    if (fuse) {
+      if (stride == 1) {
+        if (chunk == 64) L(conv2d_wgrad_sp_kernel<8, 1, 64, true>);
+        else if (OWT == 32) L(conv2d_wgrad_sp_kernel<32, 1, 128, true>);
+        else if (OWT == 16) L(conv2d_wgrad_sp_kernel<16, 1, 128, true>);
+        else L(conv2d_wgrad_sp_kernel<8, 1, 128, true>);
+      } else {
+        if (chunk == 64) L(conv2d_wgrad_sp_kernel<8, 2, 64, true>);
+        else if (OWT == 32) L(conv2d_wgrad_sp_kernel<32, 2, 128, true>);
+        else if (OWT == 16) L(conv2d_wgrad_sp_kernel<16, 2, 128, true>);
+        else L(conv2d_wgrad_sp_kernel<8, 2, 128, true>);
+      }
+    } else if (stride == 1) {
+      if (chunk == 64) L(conv2d_wgrad_sp_kernel<8, 1, 64>);
+      else if (OWT == 32) L(conv2d_wgrad_sp_kernel<32, 1, 128>);
+      else if (OWT == 16) L(conv2d_wgrad_sp_kernel<16, 1, 128>);
+      else L(conv2d_wgrad_sp_kernel<8, 1, 128>);
+    } else {
+      if (chunk == 64) L(conv2d_wgrad_sp_kernel<8, 2, 64>);
+      else if (OWT == 32) L(conv2d_wgrad_sp_kernel<32, 2, 128>);
+      else if (OWT == 16) L(conv2d_wgrad_sp_kernel<16, 2, 128>);
+      else L(conv2d_wgrad_sp_kernel<8, 2, 128>);
+    }
+    return dw.view({cd.Cout, cd.Cin, 3, 3});
+  }
+
   int planes = ((cd.Cout + 31) / 32) * ((K + 31) / 32);
   int64_t target_chunks = std::max<int64_t>(1, 2048 / std::max(planes, 1));
   int64_t chunk = std::max<int64_t>(128, (M + target_chunks - 1) /
