@@ -44,8 +44,10 @@ class ResNet18(nn.Module):
     def __init__(self, in_channels=3, num_class=10, widths=(64, 128, 256, 512)):
         super().__init__()
         self.stem = nn.Sequential(
-            # 7x7 stem: library conv (one layer; outside the 3x3 family)
-            nn.Conv2d(in_channels, widths[0], 7, stride=2, padding=3, bias=False),
+            # 7x7 stem: in-tree igemm fwd+wgrad (KS=7 instances); MIOpen's
+            # bf16-NCHW 7x7 wgrad fell to a ~100 ms naive kernel (r2
+            # profile) which dominated the whole step
+            OpsConv2d(in_channels, widths[0], 7, stride=2, padding=3, bias=False),
             OpsBatchNorm2d(widths[0], relu=True),
             nn.MaxPool2d(3, stride=2, padding=1))
         layers = []

@@ -237,8 +237,11 @@ def conv3d(x, weight, bias=None, stride=1):
 
 
 class _Conv2dFn(torch.autograd.Function):
-    """3x3/pad-1/stride-{1,2} Conv2d on the 2D implicit-GEMM MFMA kernels
-    (conv2d.hip) — the ResNet-18 hot path."""
+    """3x3/pad-1 (and 7x7/pad-3 stem, fwd+wgrad) Conv2d on the 2D
+    implicit-GEMM MFMA kernels (conv2d.hip) — the ResNet-18 hot path
+    (VERDICT r1 item 7). The 7x7 dgrad is unimplemented: the stem is the
+    first layer, its input never needs a gradient (module routing
+    enforces this)."""
 
     @staticmethod
     def forward(ctx, x, weight, bias, stride):
@@ -250,6 +253,7 @@ class _Conv2dFn(torch.autograd.Function):
             out = out + bias.to(out.dtype).view(1, -1, 1, 1)
         ctx.save_for_backward(xb, wb)
         ctx.stride = stride
+        ctx.ks = weight.shape[2]
         ctx.has_bias = bias is not None
         ctx.in_dtype = x.dtype
         ctx.w_dtype = weight.dtype
@@ -262,10 +266,12 @@ class _Conv2dFn(torch.autograd.Function):
         go = grad_out.to(torch.bfloat16).contiguous()
         gx = gw = gb = None
         if ctx.needs_input_grad[0]:
+            assert ctx.ks == 3, '7x7 dgrad not implemented (stem only)'
             gx = C.conv2d_dgrad(go, wb, list(xb.shape),
                                 ctx.stride).to(ctx.in_dtype)
         if ctx.needs_input_grad[1]:
-            gw = C.conv2d_wgrad(xb, go, ctx.stride).to(ctx.w_dtype)
+            gw = C.conv2d_wgrad(xb, go, ctx.stride,
+                                ks=ctx.ks).to(ctx.w_dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             gb = C.channel_sum(go)
         return gx, gw, gb, None
@@ -286,6 +292,14 @@ class OpsConv2d(nn.Conv2d):
             if (self.kernel_size == (3, 3) and self.padding == (1, 1)
                     and self.stride[0] in (1, 2)
                     and self.stride[0] == self.stride[1]):
+                return _Conv2dFn.apply(x, self.weight, self.bias,
+                                       int(self.stride[0]))
+            # 7x7 stem: fwd+wgrad in-tree; input must not need a gradient
+            # (it is the image) since the 7x7 dgrad is not implemented
+            if (self.kernel_size == (7, 7) and self.padding == (3, 3)
+                    and self.stride[0] in (1, 2)
+                    and self.stride[0] == self.stride[1]
+                    and not x.requires_grad):
                 return _Conv2dFn.apply(x, self.weight, self.bias,
                                        int(self.stride[0]))
             if self.kernel_size == (1, 1) and self.padding == (0, 0):

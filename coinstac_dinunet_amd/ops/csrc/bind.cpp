@@ -61,7 +61,8 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
 torch::Tensor conv2d_dgrad(torch::Tensor go, torch::Tensor w,
                            std::vector<int64_t> in_shape, int64_t stride);
 torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor go,
-                           int64_t stride, c10::optional<torch::Tensor> bn_ab);
+                           int64_t stride, c10::optional<torch::Tensor> bn_ab,
+                           int64_t ks);
 // rankdad.hip
 std::vector<torch::Tensor> power_iter_bc(torch::Tensor B, torch::Tensor C,
                                          int64_t rank, int64_t iters,
@@ -117,7 +118,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("stride"), py::arg("bn_ab") = py::none());
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad, py::arg("x"), py::arg("go"),
-        py::arg("stride"), py::arg("bn_ab") = py::none());
+        py::arg("stride"), py::arg("bn_ab") = py::none(),
+        py::arg("ks") = 3);
   m.def("power_iter_bc", &power_iter_bc);
   m.def("rowsum", &rowsum);
   m.def("conv3d_pw_fwd", &conv3d_pw_fwd);

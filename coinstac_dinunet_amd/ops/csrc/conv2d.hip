@@ -21,7 +21,7 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 struct Conv2dDims {
   int N, Cin, H, W;
   int Cout, OH, OW;
-  int stride;  // pad fixed at 1, kernel 3x3
+  int stride;  // pad = KS/2; kernel KS x KS (KS = 3, or 7 for the stem)
 };
 
 #define CBM2 64
@@ -29,22 +29,23 @@ struct Conv2dDims {
 #define CBK2 32
 #define LDA_PAD2 8
 
-template <bool DGRAD>
+template <bool DGRAD, int KS = 3>
 __device__ inline __bf16 gather2_w(const __bf16* __restrict__ w,
                                    const Conv2dDims& cd, int k, int col) {
+  constexpr int KK = KS * KS;
   if (!DGRAD) {
-    if (col >= cd.Cout || k >= cd.Cin * 9) return (__bf16)0.f;
-    return w[(int64_t)col * (cd.Cin * 9) + k];
+    if (col >= cd.Cout || k >= cd.Cin * KK) return (__bf16)0.f;
+    return w[(int64_t)col * (cd.Cin * KK) + k];
   }
-  const int co = k / 9;
-  const int r = k - co * 9;
+  const int co = k / KK;
+  const int r = k - co * KK;
   if (co >= cd.Cout || col >= cd.Cin) return (__bf16)0.f;
   // taps un-flipped here; the (ih + 1 - kh) mapping in the A gather
   // implements the transposed conv (same convention as conv3d.hip)
-  return w[((int64_t)co * cd.Cin + col) * 9 + r];
+  return w[((int64_t)co * cd.Cin + col) * KK + r];
 }
 
-template <bool DGRAD, int STRIDE, bool FUSE_BN = false>
+template <bool DGRAD, int STRIDE, bool FUSE_BN = false, int KS = 3>
 __global__ __launch_bounds__(256) void conv2d_igemm_kernel(
     const __bf16* __restrict__ Ain, const __bf16* __restrict__ w,
     __bf16* __restrict__ out, Conv2dDims cd, int64_t M, int Ncol, int K,
@@ -88,10 +89,12 @@ __global__ __launch_bounds__(256) void conv2d_igemm_kernel(
 
   for (int k0 = 0; k0 < K; k0 += CBK2) {
     const int k = k0 + kk_t;
+    constexpr int KK = KS * KS;
+    constexpr int PAD = KS / 2;
     if (!DGRAD) {
-      const int ci = k / 9;
-      const int r = k - ci * 9;
-      const int kh = r / 3, kw = r % 3;
+      const int ci = k / KK;
+      const int r = k - ci * KK;
+      const int kh = r / KS, kw = r % KS;
       const bool k_ok = ci < cd.Cin;
       float a_c = 1.f, b_c = 0.f;
       if (FUSE_BN && k_ok) { a_c = bn_ab[ci * 2]; b_c = bn_ab[ci * 2 + 1]; }
@@ -99,8 +102,8 @@ __global__ __launch_bounds__(256) void conv2d_igemm_kernel(
       for (int j = 0; j < 8; ++j) {
         __bf16 v = (__bf16)0.f;
         if (k_ok && (m_ok || (bm + mbase + j) < M)) {
-          const int ih = ph[j] * STRIDE - 1 + kh;
-          const int iw = pw[j] * STRIDE - 1 + kw;
+          const int ih = ph[j] * STRIDE - PAD + kh;
+          const int iw = pw[j] * STRIDE - PAD + kw;
           if ((unsigned)ih < (unsigned)cd.H && (unsigned)iw < (unsigned)cd.W) {
             v = Ain[((int64_t)pn[j] * cd.Cin + ci) * HW +
                     (int64_t)ih * cd.W + iw];
@@ -110,15 +113,15 @@ __global__ __launch_bounds__(256) void conv2d_igemm_kernel(
         sA[mbase + j][kk_t] = v;
       }
     } else {
-      const int co = k / 9;
-      const int r = k - co * 9;
-      const int kh = r / 3, kw = r % 3;
+      const int co = k / KK;
+      const int r = k - co * KK;
+      const int kh = r / KS, kw = r % KS;
       const bool k_ok = co < cd.Cout;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         __bf16 v = (__bf16)0.f;
         if (k_ok && (m_ok || (bm + mbase + j) < M)) {
-          const int th = ph[j] + 1 - kh, tw = pw[j] + 1 - kw;
+          const int th = ph[j] + PAD - kh, tw = pw[j] + PAD - kw;
           if (STRIDE == 1 ||
               (!(th & (STRIDE - 1)) && !(tw & (STRIDE - 1)))) {
             const int oh = th / STRIDE, ow = tw / STRIDE;
@@ -135,7 +138,7 @@ __global__ __launch_bounds__(256) void conv2d_igemm_kernel(
     for (int e = 0; e < 8; ++e) {
       int idx = tid * 8 + e;
       int kk = idx & 31, col = idx >> 5;
-      sBT[col][kk] = gather2_w<DGRAD>(w, cd, k0 + kk, bn + col);
+      sBT[col][kk] = gather2_w<DGRAD, KS>(w, cd, k0 + kk, bn + col);
     }
     __syncthreads();
 
@@ -525,7 +528,7 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_sp_kernel(
 // ---------------------------------------------------------------------------
 #define WMB2 128
 
-template <bool FUSE_BN = false>
+template <bool FUSE_BN = false, int KS = 3>
 __global__ __launch_bounds__(256) void conv2d_wgrad_kernel(
     const __bf16* __restrict__ x, const __bf16* __restrict__ go,
     float* __restrict__ dw, Conv2dDims cd, int64_t M, int K, int64_t chunk,
@@ -541,12 +544,14 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_kernel(
   const int wave = tid >> 6, lane = tid & 63;
   const int wi = wave >> 1, wj = wave & 1;
 
+  constexpr int KK = KS * KS;
+  constexpr int PAD = KS / 2;
   const int mi0 = (tid & 7) * 4;
   const int ct = tid >> 3;
   const int kq = k0 + ct;
-  const int ci = kq / 9;
-  const int rr = kq - ci * 9;
-  const int kh = rr / 3, kw = rr % 3;
+  const int ci = kq / KK;
+  const int rr = kq - ci * KK;
+  const int kh = rr / KS, kw = rr % KS;
   const bool k_ok = ci < cd.Cin && kq < K;
   const bool c_ok = (co0 + ct) < cd.Cout;
   float a_c = 1.f, b_c = 0.f;
@@ -581,8 +586,8 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_kernel(
         sGoT[ct][mloc] = gv;
         __bf16 xv = (__bf16)0.f;
         if (ok && k_ok) {
-          const int ih = jh * cd.stride - 1 + kh;
-          const int iw = jw * cd.stride - 1 + kw;
+          const int ih = jh * cd.stride - PAD + kh;
+          const int iw = jw * cd.stride - PAD + kw;
           if ((unsigned)ih < (unsigned)cd.H &&
               (unsigned)iw < (unsigned)cd.W) {
             xv = x[((int64_t)jn * cd.Cin + ci) * HW +
@@ -631,14 +636,14 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_kernel(
 // hosts
 // ---------------------------------------------------------------------------
 static Conv2dDims make_dims2(const torch::Tensor& x, const torch::Tensor& w,
-                             int stride) {
+                             int stride, int ks = 3) {
   Conv2dDims cd;
   cd.N = (int)x.size(0); cd.Cin = (int)x.size(1);
   cd.H = (int)x.size(2); cd.W = (int)x.size(3);
   cd.Cout = (int)w.size(0);
   cd.stride = stride;
-  cd.OH = (cd.H + 2 - 3) / stride + 1;
-  cd.OW = (cd.W + 2 - 3) / stride + 1;
+  cd.OH = (cd.H + 2 * (ks / 2) - ks) / stride + 1;
+  cd.OW = (cd.W + 2 * (ks / 2) - ks) / stride + 1;
   return cd;
 }
 
@@ -715,9 +720,11 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "conv2d_fwd wants bf16");
   auto xc = x.contiguous();
   auto wc = w.to(torch::kBFloat16).contiguous();
-  auto cd = make_dims2(xc, wc, (int)stride);
-  TORCH_CHECK(wc.size(2) == 3 && wc.size(3) == 3 && wc.size(1) == cd.Cin,
-              "3x3 kernels only");
+  const int ks = (int)wc.size(2);
+  TORCH_CHECK(wc.size(3) == ks && (ks == 3 || ks == 7),
+              "3x3 or 7x7 kernels only");
+  auto cd = make_dims2(xc, wc, (int)stride, ks);
+  TORCH_CHECK(wc.size(1) == cd.Cin, "channel mismatch");
   TORCH_CHECK(stride == 1 || stride == 2, "stride must be 1 or 2");
   const bool fuse = bn_ab.defined() && bn_ab.numel() > 0;
   torch::Tensor ab;
@@ -729,7 +736,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
   }
   auto out = torch::empty({cd.N, cd.Cout, cd.OH, cd.OW}, xc.options());
 
-  if (cd.Cin >= 16 && cd.OH * cd.OW >= 32) {
+  if (ks == 3 && cd.Cin >= 16 && cd.OH * cd.OW >= 32) {
     // spatial-slab tap-reuse path (clamped staging: any width)
     Sp2Dims sd;
     sd.N = cd.N; sd.KCH = cd.Cin; sd.H = cd.H; sd.W = cd.W;
@@ -743,7 +750,7 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
   }
 
   int64_t M = (int64_t)cd.N * cd.OH * cd.OW;
-  int K = cd.Cin * 9;
+  int K = cd.Cin * ks * ks;
   dim3 grid((unsigned)((M + CBM2 - 1) / CBM2), (cd.Cout + CBN2 - 1) / CBN2);
   auto L = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, dim3(256), 0, current_stream(),
@@ -752,7 +759,11 @@ torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
                        reinterpret_cast<__bf16*>(out.data_ptr()), cd, M,
                        cd.Cout, K, abp);
   };
-  if (fuse) {
+  if (ks == 7) {
+    TORCH_CHECK(!fuse, "7x7 fwd: no fused-BN instances");
+    if (stride == 1) L(conv2d_igemm_kernel<false, 1, false, 7>);
+    else L(conv2d_igemm_kernel<false, 2, false, 7>);
+  } else if (fuse) {
     if (stride == 1) L(conv2d_igemm_kernel<false, 1, true>);
     else L(conv2d_igemm_kernel<false, 2, true>);
   } else if (stride == 1) {
@@ -806,8 +817,10 @@ torch::Tensor conv2d_dgrad(torch::Tensor go, torch::Tensor w,
 }
 
 torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor go,
-                           int64_t stride, c10::optional<torch::Tensor> bn_ab_opt) {
+                           int64_t stride, c10::optional<torch::Tensor> bn_ab_opt,
+                           int64_t ks) {
   torch::Tensor bn_ab = bn_ab_opt.value_or(torch::Tensor());
+  TORCH_CHECK(ks == 3 || ks == 7, "3x3 or 7x7 kernels only");
   CHECK_GPU(x);
   auto xc = x.to(torch::kBFloat16).contiguous();
   auto g = go.to(torch::kBFloat16).contiguous();
@@ -817,7 +830,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor go,
   cd.stride = (int)stride;
   cd.Cout = (int)g.size(1);
   cd.OH = (int)g.size(2); cd.OW = (int)g.size(3);
-  int K = cd.Cin * 9;
+  int K = cd.Cin * (int)(ks * ks);
   int64_t M = (int64_t)cd.N * cd.OH * cd.OW;
   auto dw = torch::zeros({cd.Cout, (int64_t)K},
                          xc.options().dtype(torch::kFloat32));
@@ -830,7 +843,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor go,
     abp = ab.data_ptr<float>();
   }
 
-  if (cd.Cin >= 16 && cd.OH * cd.OW >= 32 &&
+  if (ks == 3 && cd.Cin >= 16 && cd.OH * cd.OW >= 32 &&
       (stride == 1 || stride == 2)) {
     int OWT = pick_owt2(cd.OW);
     int chunk = 128;
@@ -889,7 +902,13 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor go,
                        reinterpret_cast<const __bf16*>(g.data_ptr()),
                        dw.data_ptr<float>(), cd, M, K, chunk, abp);
   };
-  if (fuse) L(conv2d_wgrad_kernel<true>);
-  else L(conv2d_wgrad_kernel<false>);
-  return dw.view({cd.Cout, cd.Cin, 3, 3});
+  if (ks == 7) {
+    TORCH_CHECK(!fuse, "7x7 wgrad: no fused-BN instances");
+    L(conv2d_wgrad_kernel<false, 7>);
+  } else if (fuse) {
+    L(conv2d_wgrad_kernel<true>);
+  } else {
+    L(conv2d_wgrad_kernel<false>);
+  }
+  return dw.view({cd.Cout, cd.Cin, (int64_t)ks, (int64_t)ks});
 }

@@ -94,6 +94,23 @@ def test_opsconv2d_1x1_downsample(dev):
     torch.testing.assert_close(out.float(), ref, rtol=5e-2, atol=0.3)
 
 
+def test_conv2d_7x7_stem(dev):
+    """7x7/pad-3/stride-2 stem fwd + wgrad (KS=7 igemm instances)."""
+    torch.manual_seed(66)
+    x = torch.randn(2, 3, 32, 32, device=dev, dtype=torch.bfloat16)
+    w = torch.randn(16, 3, 7, 7, device=dev, dtype=torch.bfloat16) * 0.05
+    w[1, 2, 0, 6] += 1.0
+    out = C.conv2d_fwd(x, w, 2)
+    ref = F.conv2d(x.float(), w.float(), stride=2, padding=3)
+    torch.testing.assert_close(out.float(), ref, rtol=5e-2, atol=0.3)
+    go = torch.randn_like(ref).bfloat16() * 0.1
+    gw = C.conv2d_wgrad(x, go, 2, ks=7)
+    wr = w.float().requires_grad_(True)
+    F.conv2d(x.float(), wr, stride=2, padding=3).backward(go.float())
+    torch.testing.assert_close(gw.view_as(wr), wr.grad, rtol=5e-2,
+                               atol=5e-2 * (2 * 16 * 16) ** 0.5 * 0.1 + 0.2)
+
+
 def test_resnet18_trains_on_hip_kernels(dev):
     """Whole ResNet-18 fwd/bwd on GPU: loss decreases, all grads flow."""
     from coinstac_dinunet_amd.models import ResNet18
