@@ -231,6 +231,84 @@ __global__ __launch_bounds__(256) void pw_gemm_batched_kernel(
 // TRANS_A=false); dgrad uses the SAME W but needs opA(i=ci, c=co) =
 // W[co*Ci+ci] = A[c*M + i] — the TRANS_A=true indexing above.
 
+// wgrad as batched MFMA: gw[Co, Ci] = sum_n go_n[Co, S] @ x_n[Ci, S]^T.
+// Per z-block NZ batches are accumulated in registers, then one fp32
+// atomicAdd per tile element folds partials across z-blocks (replaces
+// the streaming pw_wgrad whose per-(n,s) atomics serialized at ~1 ms per
+// call on the 512x256 downsample shapes).
+template <int NZ>
+__global__ __launch_bounds__(256) void pw_wgrad_batched_kernel(
+    const __bf16* __restrict__ go, const __bf16* __restrict__ x,
+    float* __restrict__ gw, int M, int N, int K, int batches,
+    int64_t strideA, int64_t strideB) {
+  __shared__ float sA[PWBM][PWLDK];
+  __shared__ float sB[PWBN][PWLDK];
+
+  const int bm = blockIdx.x * PWBM;
+  const int bn = blockIdx.y * PWBN;
+  const int n0 = blockIdx.z * NZ;
+  const int tid = threadIdx.x;
+  const int wave = tid / WAVE_SIZE;
+  const int lane = tid % WAVE_SIZE;
+  const int wm = (wave >> 1) * 32;
+  const int wn = (wave & 1) * 32;
+
+  pwf32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int nz = 0; nz < NZ && n0 + nz < batches; ++nz) {
+    const __bf16* An = go + (int64_t)(n0 + nz) * strideA;
+    const __bf16* Bn = x + (int64_t)(n0 + nz) * strideB;
+    for (int k0 = 0; k0 < K; k0 += PWBK) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int idx = tid + r * 256;
+        int row = idx / PWBK, col = idx % PWBK;
+        int gm = bm + row, gk = k0 + col;
+        sA[row][col] = (gm < M && gk < K)
+                           ? __bfloat162float(An[(int64_t)gm * K + gk])
+                           : 0.f;
+        int gn = bn + row;
+        sB[row][col] = (gn < N && gk < K)
+                           ? __bfloat162float(Bn[(int64_t)gn * K + gk])
+                           : 0.f;
+      }
+      __syncthreads();
+#pragma unroll
+      for (int kk = 0; kk < PWBK; kk += 4) {
+        const int ar = lane & 15, ak = lane >> 4;
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+          for (int j = 0; j < 2; ++j) {
+            float a = sA[wm + i * 16 + ar][kk + ak];
+            float b = sB[wn + j * 16 + ar][kk + ak];
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                a, b, acc[i][j], 0, 0, 0);
+          }
+      }
+      __syncthreads();
+    }
+  }
+
+  const int ccol = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int gm = bm + wm + i * 16 + crow0 + r;
+        int gn = bn + wn + j * 16 + ccol;
+        if (gm < M && gn < N)
+          atomicAdd(&gw[(int64_t)gm * N + gn], acc[i][j][r]);
+      }
+}
+
 // ---------------------------------------------------------------------------
 // host wrappers
 // ---------------------------------------------------------------------------
@@ -310,14 +388,15 @@ torch::Tensor conv3d_pw_wgrad(torch::Tensor x, torch::Tensor go) {
   const int64_t S = xc.numel() / ((int64_t)N * Cin);
   auto gw = torch::zeros({Cout, Cin},
                          xc.options().dtype(torch::kFloat32));
-  constexpr int CO_T = 4, CI_T = 16;
-  const int nco = (Cout + CO_T - 1) / CO_T;
-  const int nci = (Cin + CI_T - 1) / CI_T;
-  hipLaunchKernelGGL((pw_wgrad_kernel<CO_T, CI_T>),
-                     dim3(pw_grid((int64_t)N * S), nco * nci),
-                     dim3(ELEM_BLOCK), 0, current_stream(),
-                     reinterpret_cast<const __bf16*>(xc.data_ptr()),
+  // batched MFMA: per n, go_n[Co,S] @ x_n[Ci,S]^T, NZ batches per block
+  constexpr int NZ = 8;
+  dim3 grid((Cout + PWBM - 1) / PWBM, (Cin + PWBN - 1) / PWBN,
+            (N + NZ - 1) / NZ);
+  hipLaunchKernelGGL((pw_wgrad_batched_kernel<NZ>), grid, dim3(256), 0,
+                     current_stream(),
                      reinterpret_cast<const __bf16*>(g.data_ptr()),
-                     gw.data_ptr<float>(), N, Cin, Cout, S);
+                     reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                     gw.data_ptr<float>(), Cout, Cin, (int)S, N,
+                     (int64_t)Cout * S, (int64_t)Cin * S);
   return gw;
 }
