@@ -206,7 +206,12 @@ struct Sp2Dims {
   int Kpad;          // row length of WB = kts * KT_PAD
 };
 
-template <int OWT, int STRIDE, int CTILE, int CHUNK, bool FUSE_BN = false>
+// NCOLT: output-channel tile per block. The deep small-image layers
+// (7^2/14^2, 256-512 channels) are slab-staging bound — every 32-column
+// block re-stages the same input slab; NCOLT=64 halves that redundancy
+// and doubles the MFMA work per barrier.
+template <int OWT, int STRIDE, int CTILE, int CHUNK, bool FUSE_BN = false,
+          int NCOLT = 32>
 __global__ __launch_bounds__(256) void conv2d_spatial_kernel(
     const __bf16* __restrict__ in, const __bf16* __restrict__ wb,
     __bf16* __restrict__ out, Sp2Dims sd, int64_t nchunks,
@@ -216,12 +221,13 @@ __global__ __launch_bounds__(256) void conv2d_spatial_kernel(
   constexpr int W2 = IW + (STRIDE == 1 ? 4 : 2);
   constexpr int H2 = STRIDE * (OHT - 1) + 3;
   constexpr int MPW = CHUNK / 64;
+  constexpr int NCF = NCOLT / 16;  // ncol fragments per wave
   static_assert(MPW >= 1, "chunk too small");
   constexpr int KT_PAD = ((CTILE * 9 + 31) / 32) * 32;
   __shared__ __bf16 sX[CTILE][H2][W2];
   __shared__ unsigned short sKtab[KT_PAD + 8];
 
-  const int ncol0 = blockIdx.y * 32;
+  const int ncol0 = blockIdx.y * NCOLT;
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
   const int row = lane & 15, kg = lane >> 4;
@@ -247,11 +253,11 @@ __global__ __launch_bounds__(256) void conv2d_spatial_kernel(
     sKtab[k] = off;
   }
 
-  f32x4 acc[MPW][2];
+  f32x4 acc[MPW][NCF];
 #pragma unroll
   for (int i = 0; i < MPW; ++i)
 #pragma unroll
-    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < NCF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const int64_t HW = (int64_t)sd.H * sd.W;
   const int64_t in_n = (int64_t)n * sd.KCH * HW;
@@ -321,9 +327,9 @@ __global__ __launch_bounds__(256) void conv2d_spatial_kernel(
             afrag[i][j] = slab[base + kt8[j]];
         }
       }
-      bf16x8 bfrag[2];
+      bf16x8 bfrag[NCF];
 #pragma unroll
-      for (int i = 0; i < 2; ++i) {
+      for (int i = 0; i < NCF; ++i) {
         const int col = ncol0 + i * 16 + row;
         const int64_t off =
             (int64_t)col * sd.Kpad + kbase_g + ks * 32 + kg * 8;
@@ -334,7 +340,7 @@ __global__ __launch_bounds__(256) void conv2d_spatial_kernel(
 #pragma unroll
       for (int i = 0; i < MPW; ++i)
 #pragma unroll
-        for (int j = 0; j < 2; ++j)
+        for (int j = 0; j < NCF; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     }
@@ -347,7 +353,7 @@ __global__ __launch_bounds__(256) void conv2d_spatial_kernel(
 #pragma unroll
   for (int i = 0; i < MPW; ++i) {
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
+    for (int j = 0; j < NCF; ++j) {
       const int col = ncol0 + j * 16 + ccol;
       if (col >= sd.NCOL) continue;
 #pragma unroll
@@ -674,12 +680,17 @@ static void launch_spatial2(torch::Tensor in, torch::Tensor wb,
                             const float* bn_ab) {
   int OWT = pick_owt2(sd.TW);
   int chunk = stride == 1 ? 256 : 128;
-  if (sd.TH * sd.TW < chunk) { chunk = 64; OWT = 8; }
+  int ncolt = 32;
+  if (sd.TH * sd.TW < chunk) {
+    chunk = 64;
+    OWT = 8;
+    if (sd.NCOL >= 64) ncolt = 64;  // small-image wide-column instances
+  }
   int OHT = chunk / OWT;
   int wtiles = (sd.TW + OWT - 1) / OWT;
   int htiles = (sd.TH + OHT - 1) / OHT;
   int64_t nchunks = (int64_t)sd.N * htiles * wtiles;
-  dim3 grid((unsigned)nchunks, (sd.NCOL + 31) / 32);
+  dim3 grid((unsigned)nchunks, (sd.NCOL + ncolt - 1) / ncolt);
   auto s = current_stream();
   const __bf16* ip = reinterpret_cast<const __bf16*>(in.data_ptr());
   const __bf16* wp = reinterpret_cast<const __bf16*>(wb.data_ptr());
@@ -690,23 +701,31 @@ static void launch_spatial2(torch::Tensor in, torch::Tensor wb,
   };
   if (bn_ab != nullptr) {
     if (stride == 1) {
-      if (chunk == 64) L(conv2d_spatial_kernel<8, 1, 32, 64, true>);
+      if (chunk == 64 && ncolt == 64)
+        L(conv2d_spatial_kernel<8, 1, 32, 64, true, 64>);
+      else if (chunk == 64) L(conv2d_spatial_kernel<8, 1, 32, 64, true>);
       else if (OWT == 32) L(conv2d_spatial_kernel<32, 1, 32, 256, true>);
       else if (OWT == 16) L(conv2d_spatial_kernel<16, 1, 32, 256, true>);
       else L(conv2d_spatial_kernel<8, 1, 32, 256, true>);
     } else {
-      if (chunk == 64) L(conv2d_spatial_kernel<8, 2, 16, 64, true>);
+      if (chunk == 64 && ncolt == 64)
+        L(conv2d_spatial_kernel<8, 2, 16, 64, true, 64>);
+      else if (chunk == 64) L(conv2d_spatial_kernel<8, 2, 16, 64, true>);
       else if (OWT == 32) L(conv2d_spatial_kernel<32, 2, 16, 128, true>);
       else if (OWT == 16) L(conv2d_spatial_kernel<16, 2, 16, 128, true>);
       else L(conv2d_spatial_kernel<8, 2, 16, 128, true>);
     }
   } else if (stride == 1) {
-    if (chunk == 64) L(conv2d_spatial_kernel<8, 1, 32, 64>);
+    if (chunk == 64 && ncolt == 64)
+      L(conv2d_spatial_kernel<8, 1, 32, 64, false, 64>);
+    else if (chunk == 64) L(conv2d_spatial_kernel<8, 1, 32, 64>);
     else if (OWT == 32) L(conv2d_spatial_kernel<32, 1, 32, 256>);
     else if (OWT == 16) L(conv2d_spatial_kernel<16, 1, 32, 256>);
     else L(conv2d_spatial_kernel<8, 1, 32, 256>);
   } else {
-    if (chunk == 64) L(conv2d_spatial_kernel<8, 2, 16, 64>);
+    if (chunk == 64 && ncolt == 64)
+      L(conv2d_spatial_kernel<8, 2, 16, 64, false, 64>);
+    else if (chunk == 64) L(conv2d_spatial_kernel<8, 2, 16, 64>);
     else if (OWT == 32) L(conv2d_spatial_kernel<32, 2, 16, 128>);
     else if (OWT == 16) L(conv2d_spatial_kernel<16, 2, 16, 128>);
     else L(conv2d_spatial_kernel<8, 2, 16, 128>);
