@@ -92,6 +92,34 @@ static AdamHyper make_hyper(double lr, double beta1, double beta2, double eps,
   return h;
 }
 
+
+// Multi-tensor form: ONE launch for the whole parameter list. Blocks map
+// to (tensor, chunk) pairs through a small device table (the per-tensor
+// launch form cost ~60 launches/step on ResNet-18 — pure overhead).
+__global__ void adam_mt_kernel(const int64_t* __restrict__ ptrs,
+                               const int* __restrict__ bmap,
+                               int64_t chunk, AdamHyper h) {
+  const int t = bmap[blockIdx.x * 2];
+  const int64_t off = (int64_t)bmap[blockIdx.x * 2 + 1] * chunk;
+  float* p = reinterpret_cast<float*>(ptrs[t * 5 + 0]);
+  const float* g = reinterpret_cast<const float*>(ptrs[t * 5 + 1]);
+  float* m = reinterpret_cast<float*>(ptrs[t * 5 + 2]);
+  float* v = reinterpret_cast<float*>(ptrs[t * 5 + 3]);
+  const int64_t n = ptrs[t * 5 + 4];
+  const int64_t end = min(off + chunk, n);
+  const float inv_b1 = 1.0f / h.bias1;
+  const float inv_b2 = 1.0f / h.bias2;
+  for (int64_t i = off + threadIdx.x; i < end; i += blockDim.x) {
+    float pp = p[i], gg = g[i];
+    gg += h.weight_decay * pp;
+    float mm = h.beta1 * m[i] + (1.0f - h.beta1) * gg;
+    float vv = h.beta2 * v[i] + (1.0f - h.beta2) * gg * gg;
+    p[i] = pp - h.lr * (mm * inv_b1) / (sqrtf(vv * inv_b2) + h.eps);
+    m[i] = mm;
+    v[i] = vv;
+  }
+}
+
 void fused_adam_flat(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                      torch::Tensor v, double lr, double beta1, double beta2,
                      double eps, double weight_decay, int64_t step) {
@@ -112,22 +140,44 @@ void fused_adam(std::vector<torch::Tensor> params,
                 std::vector<torch::Tensor> exp_avg_sqs, double lr,
                 double beta1, double beta2, double eps, double weight_decay,
                 int64_t step) {
-  auto h = make_hyper(lr, beta1, beta2, eps, weight_decay, step);
-  hipStream_t stream = current_stream();
-  for (size_t i = 0; i < params.size(); ++i) {
-    auto p = params[i];
-    auto g = grads[i].to(torch::kFloat32).contiguous();
-    int64_t n = p.numel();
-    TORCH_CHECK(p.scalar_type() == torch::kFloat32,
-                "fused_adam expects fp32 master params");
-    hipLaunchKernelGGL(adam_kernel, dim3(elem_grid(n)), dim3(ELEM_BLOCK), 0,
-                       stream, p.data_ptr<float>(), g.data_ptr<float>(),
-                       exp_avgs[i].data_ptr<float>(),
-                       exp_avg_sqs[i].data_ptr<float>(), n,
-                       vec4_count(n, p.data_ptr(), g.data_ptr(),
-                                  exp_avgs[i].data_ptr(),
-                                  exp_avg_sqs[i].data_ptr()), h);
+  const int T = (int)params.size();
+  if (T == 0) return;
+  AdamHyper h;
+  h.lr = (float)lr; h.beta1 = (float)beta1; h.beta2 = (float)beta2;
+  h.eps = (float)eps; h.weight_decay = (float)weight_decay;
+  h.bias1 = 1.0f - powf((float)beta1, (float)step);
+  h.bias2 = 1.0f - powf((float)beta2, (float)step);
+
+  const int64_t chunk = 16384;
+  auto ptrs_cpu = torch::empty({T, 5}, torch::dtype(torch::kInt64));
+  auto* pc = ptrs_cpu.data_ptr<int64_t>();
+  std::vector<int> bmap_v;
+  for (int t = 0; t < T; ++t) {
+    TORCH_CHECK(params[t].scalar_type() == torch::kFloat32 &&
+                grads[t].scalar_type() == torch::kFloat32,
+                "fused_adam wants fp32 params/grads");
+    pc[t * 5 + 0] = (int64_t)params[t].data_ptr();
+    pc[t * 5 + 1] = (int64_t)grads[t].data_ptr();
+    pc[t * 5 + 2] = (int64_t)exp_avgs[t].data_ptr();
+    pc[t * 5 + 3] = (int64_t)exp_avg_sqs[t].data_ptr();
+    const int64_t n = params[t].numel();
+    pc[t * 5 + 4] = n;
+    const int64_t nch = (n + chunk - 1) / chunk;
+    for (int64_t c = 0; c < nch; ++c) {
+      bmap_v.push_back(t);
+      bmap_v.push_back((int)c);
+    }
   }
+  auto dev = params[0].device();
+  auto ptrs = ptrs_cpu.to(dev, /*non_blocking=*/true);
+  auto bmap_cpu = torch::from_blob(bmap_v.data(),
+                                   {(int64_t)bmap_v.size()},
+                                   torch::dtype(torch::kInt32)).clone();
+  auto bmap = bmap_cpu.to(dev, /*non_blocking=*/true);
+  const unsigned G = (unsigned)(bmap_v.size() / 2);
+  hipLaunchKernelGGL(adam_mt_kernel, dim3(G), dim3(ELEM_BLOCK), 0,
+                     current_stream(), ptrs.data_ptr<int64_t>(),
+                     bmap.data_ptr<int>(), chunk, h);
 }
 
 void fused_sgd(std::vector<torch::Tensor> params,

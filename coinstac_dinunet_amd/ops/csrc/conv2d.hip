@@ -187,6 +187,150 @@ __global__ __launch_bounds__(256) void conv2d_igemm_kernel(
       }
 }
 
+
+// ---------------------------------------------------------------------------
+// Stride-2 DGRAD, parity-decomposed (2D port of conv3d_dgrad_s2_kernel):
+// the masked formulation wastes 3/4 of the MFMA work; decompose dx by
+// (ih, iw) mod 2 into 4 classes, each a DENSE implicit GEMM with
+// K = Cout * (1 or 2)^2 taps. blockIdx.z = class.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void conv2d_dgrad_s2_kernel(
+    const __bf16* __restrict__ go, const __bf16* __restrict__ w,
+    __bf16* __restrict__ dx, Conv2dDims cd) {
+  __shared__ __bf16 sA[CBM2][CBK2 + LDA_PAD2];
+  __shared__ __bf16 sBT[CBN2][CBK2 + LDA_PAD2];
+
+  const int cls = blockIdx.z;
+  const int b = (cls >> 1) & 1, c = cls & 1;
+  const int Hb = (cd.H - b + 1) >> 1;
+  const int Wc = (cd.W - c + 1) >> 1;
+  const int nh = b ? 2 : 1, nw = c ? 2 : 1;
+  const int l2w = c;
+  const int T = nh * nw;
+  const int l2T = b + c;
+  const int K = cd.Cout << l2T;
+  const int64_t M = (int64_t)cd.N * Hb * Wc;
+
+  const int64_t bm = (int64_t)blockIdx.x * CBM2;
+  if (bm >= M) return;
+  const int bn = blockIdx.y * CBN2;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wm = (wave >> 1) * 32, wn = (wave & 1) * 32;
+
+  const int kk_t = tid >> 3;
+  const int mbase = (tid * 8) & 63;
+  int pn[8], ph[8], pw[8];
+  {
+    int64_t m = bm + mbase;
+    int ww = (int)(m % Wc);
+    int64_t t = m / Wc;
+    int hh = (int)(t % Hb);
+    int nn = (int)(t / Hb);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      pn[j] = nn; ph[j] = hh; pw[j] = ww;
+      if (++ww == Wc) { ww = 0; if (++hh == Hb) { hh = 0; ++nn; } }
+    }
+  }
+  const bool m_ok = (bm + mbase + 7) < M;
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int64_t OHW = (int64_t)cd.OH * cd.OW;
+
+  for (int k0 = 0; k0 < K; k0 += CBK2) {
+    {
+      const int k = k0 + kk_t;
+      const int co = k >> l2T;
+      const int r = k & (T - 1);
+      const int tw_i = r & (nw - 1);
+      const int th_i = r >> l2w;
+      const int doh = b ? (1 - th_i) : 0;  // oh = ih' + doh
+      const int dow = c ? (1 - tw_i) : 0;
+      const bool k_ok = co < cd.Cout;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        __bf16 v = (__bf16)0.f;
+        if (k_ok && (m_ok || (bm + mbase + j) < M)) {
+          const int oh = ph[j] + doh, ow = pw[j] + dow;
+          if ((unsigned)oh < (unsigned)cd.OH &&
+              (unsigned)ow < (unsigned)cd.OW)
+            v = go[((int64_t)pn[j] * cd.Cout + co) * OHW +
+                   (int64_t)oh * cd.OW + ow];
+        }
+        sA[mbase + j][kk_t] = v;
+      }
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      int idx = tid * 8 + e;
+      int kk = idx & 31, col = idx >> 5;
+      const int k = k0 + kk;
+      const int co = k >> l2T;
+      const int r = k & (T - 1);
+      const int tw_i = r & (nw - 1);
+      const int th_i = r >> l2w;
+      const int kh = b ? (th_i * 2) : 1;
+      const int kw = c ? (tw_i * 2) : 1;
+      __bf16 v = (__bf16)0.f;
+      if (co < cd.Cout && (bn + col) < cd.Cin)
+        v = w[((int64_t)co * cd.Cin + (bn + col)) * 9 + kh * 3 + kw];
+      sBT[col][kk] = v;
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int ks = 0; ks < CBK2; ks += 32) {
+      const int row = lane & 15, kg = lane >> 4;
+      bf16x8 afrag[2], bfrag[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          afrag[i][j] = sA[wm + i * 16 + row][ks + kg * 8 + j];
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          bfrag[i][j] = sBT[wn + i * 16 + row][ks + kg * 8 + j];
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int ccol = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+  const int64_t HW = (int64_t)cd.H * cd.W;
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int64_t m = bm + wm + i * 16 + crow0 + r;
+        int ci = bn + wn + j * 16 + ccol;
+        if (m < M && ci < cd.Cin) {
+          const int iw = (int)(m % Wc);
+          int64_t t = m / Wc;
+          const int ih = (int)(t % Hb);
+          const int n = (int)(t / Hb);
+          dx[((int64_t)n * cd.Cin + ci) * HW +
+             (int64_t)(2 * ih + b) * cd.W + (2 * iw + c)] =
+              (__bf16)(acc[i][j][r]);
+        }
+      }
+}
+
 // ---------------------------------------------------------------------------
 // Spatial-slab tap-reuse kernels (the conv3d_spatial.hip design with the
 // depth axis dropped): a block stages one input slab [CTILE][H2][W2] per
@@ -820,18 +964,27 @@ torch::Tensor conv2d_dgrad(torch::Tensor go, torch::Tensor w,
     return dx;
   }
 
+  if (stride == 2) {
+    // parity classes: dense sub-GEMMs, M = worst-case class size
+    int64_t Mmax = (int64_t)cd.N * ((cd.H + 1) / 2) * ((cd.W + 1) / 2);
+    dim3 grid((unsigned)((Mmax + CBM2 - 1) / CBM2),
+              (cd.Cin + CBN2 - 1) / CBN2, 4);
+    hipLaunchKernelGGL(conv2d_dgrad_s2_kernel, grid, dim3(256), 0,
+                       current_stream(),
+                       reinterpret_cast<const __bf16*>(g.data_ptr()),
+                       reinterpret_cast<const __bf16*>(wc.data_ptr()),
+                       reinterpret_cast<__bf16*>(dx.data_ptr()), cd);
+    return dx;
+  }
   int64_t M = (int64_t)cd.N * cd.H * cd.W;
   int K = cd.Cout * 9;
   dim3 grid((unsigned)((M + CBM2 - 1) / CBM2), (cd.Cin + CBN2 - 1) / CBN2);
-  auto L = [&](auto kern) {
-    hipLaunchKernelGGL(kern, grid, dim3(256), 0, current_stream(),
-                       reinterpret_cast<const __bf16*>(g.data_ptr()),
-                       reinterpret_cast<const __bf16*>(wc.data_ptr()),
-                       reinterpret_cast<__bf16*>(dx.data_ptr()), cd, M,
-                       cd.Cin, K, (const float*)nullptr);
-  };
-  if (stride == 1) L(conv2d_igemm_kernel<true, 1>);
-  else L(conv2d_igemm_kernel<true, 2>);
+  hipLaunchKernelGGL((conv2d_igemm_kernel<true, 1>), grid, dim3(256), 0,
+                     current_stream(),
+                     reinterpret_cast<const __bf16*>(g.data_ptr()),
+                     reinterpret_cast<const __bf16*>(wc.data_ptr()),
+                     reinterpret_cast<__bf16*>(dx.data_ptr()), cd, M,
+                     cd.Cin, K, (const float*)nullptr);
   return dx;
 }
 
