@@ -180,11 +180,15 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor weight,
 // contracted along its rows. Staging differs from the fwd kernel: B tiles
 // are read coalesced along j and written transposed into LDS.
 // ---------------------------------------------------------------------------
-template <typename TA, typename TB, typename TC, bool TRANS_A>
+template <typename TA, typename TB, typename TC, bool TRANS_A,
+          bool SPLIT_K = false>
 __global__ __launch_bounds__(256) void linear_bwd_kernel(
     const TA* __restrict__ A, const TB* __restrict__ B, TC* __restrict__ C,
-    int M, int N, int K, int lda, int ldb, int ldc) {
-  // M = rows of C, N = cols of C, K = contraction length
+    int M, int N, int K, int lda, int ldb, int ldc, int kchunk = 0) {
+  // M = rows of C, N = cols of C, K = contraction length. SPLIT_K:
+  // grid.z splits the contraction (skinny GEMMs whose big dimension is
+  // K — e.g. MLP wgrad [256,66] over batch 1024 — otherwise collapse to
+  // a handful of blocks); partials fold by fp32 atomics, C pre-zeroed.
   __shared__ float sA[BM][LDK];
   __shared__ float sB[BN][LDK];
 
@@ -202,7 +206,12 @@ __global__ __launch_bounds__(256) void linear_bwd_kernel(
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
+  int kbeg = 0, kend = K;
+  if (SPLIT_K) {
+    kbeg = blockIdx.z * kchunk;
+    kend = min(K, kbeg + kchunk);
+  }
+  for (int k0 = kbeg; k0 < kend; k0 += BK) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int idx = tid + r * 256;  // 0..1023 = BM*BK
@@ -252,8 +261,13 @@ __global__ __launch_bounds__(256) void linear_bwd_kernel(
       for (int r = 0; r < 4; ++r) {
         int gm = bm + wm + i * 16 + crow0 + r;
         int gn = bn + wn + j * 16 + ccol;
-        if (gm < M && gn < N)
-          C[(int64_t)gm * ldc + gn] = stf<TC>(acc[i][j][r]);
+        if (gm < M && gn < N) {
+          if (SPLIT_K)
+            atomicAdd(reinterpret_cast<float*>(&C[(int64_t)gm * ldc + gn]),
+                      acc[i][j][r]);
+          else
+            C[(int64_t)gm * ldc + gn] = stf<TC>(acc[i][j][r]);
+        }
       }
 }
 
@@ -292,38 +306,72 @@ torch::Tensor linear_dgrad(torch::Tensor go, torch::Tensor weight) {
 }
 
 // gw[N,K] = go[M,N]^T @ x[M,K]; fp32 out (master-weight gradient).
+// Skinny output over a long contraction (the MLP shape) uses split-K.
 torch::Tensor linear_wgrad(torch::Tensor go, torch::Tensor x) {
   CHECK_GPU(go);
   auto g = go.contiguous();
   auto x2 = x.to(g.scalar_type()).contiguous();
   int64_t M = g.size(0), N = g.size(1), K = x2.size(1);
   TORCH_CHECK(x2.size(0) == M, "x/grad M mismatch");
-  auto gw = torch::empty({N, K}, g.options().dtype(torch::kFloat32));
   auto stream = current_stream();
+  int64_t tiles = ((N + BM - 1) / BM) * ((K + BN - 1) / BN);
+  int nsplit = 1;
+  if (tiles < 256 && M > 4 * BK) {
+    nsplit = (int)std::min<int64_t>((M + BK - 1) / BK,
+                                    std::max<int64_t>(1, 512 / tiles));
+  }
+  const bool split = nsplit > 1;
+  auto gw = split
+      ? torch::zeros({N, K}, g.options().dtype(torch::kFloat32))
+      : torch::empty({N, K}, g.options().dtype(torch::kFloat32));
+  int kchunk = (int)(((M + (int64_t)nsplit * BK - 1) /
+                      ((int64_t)nsplit * BK)) * BK);
+  dim3 grid((unsigned)((N + BM - 1) / BM), (unsigned)((K + BN - 1) / BN),
+            (unsigned)((M + kchunk - 1) / kchunk));
+  auto L = [&](auto AB, auto kern) {
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, stream, AB.first,
+                       AB.second, gw.data_ptr<float>(), (int)N, (int)K,
+                       (int)M, (int)N, (int)K, (int)K, kchunk);
+  };
   if (g.scalar_type() == torch::kFloat32) {
-    launch_linear_bwd<float, float, float, true>(
-        g.data_ptr<float>(), x2.data_ptr<float>(), gw.data_ptr<float>(),
-        (int)N, (int)K, (int)M, (int)N, (int)K, (int)K, stream);
+    auto ab = std::make_pair(g.data_ptr<float>(), x2.data_ptr<float>());
+    if (split) L(ab, (linear_bwd_kernel<float, float, float, true, true>));
+    else L(ab, (linear_bwd_kernel<float, float, float, true, false>));
   } else if (g.scalar_type() == torch::kBFloat16) {
-    launch_linear_bwd<__hip_bfloat16, __hip_bfloat16, float, true>(
+    auto ab = std::make_pair(
         reinterpret_cast<const __hip_bfloat16*>(g.data_ptr()),
-        reinterpret_cast<const __hip_bfloat16*>(x2.data_ptr()),
-        gw.data_ptr<float>(), (int)N, (int)K, (int)M, (int)N, (int)K,
-        (int)K, stream);
+        reinterpret_cast<const __hip_bfloat16*>(x2.data_ptr()));
+    if (split)
+      L(ab, (linear_bwd_kernel<__hip_bfloat16, __hip_bfloat16, float, true,
+                               true>));
+    else
+      L(ab, (linear_bwd_kernel<__hip_bfloat16, __hip_bfloat16, float, true,
+                               false>));
   } else {
     TORCH_CHECK(false, "linear_wgrad: fp32 or bf16 only");
   }
   return gw;
 }
 
-// bias gradient: column sum of grad_out [M,N] -> [N]
+// bias gradient: column sum of grad_out [M,N] -> [N]. One 256-thread
+// block per column, threads stride the rows, wave+LDS reduce (the old
+// one-thread-per-column loop serialized the whole tensor into a single
+// block at MLP shapes — 144 us for [1024, 256]).
 __global__ void colsum_kernel(const float* __restrict__ g,
                               float* __restrict__ out, int64_t M, int64_t N) {
-  for (int64_t n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
-       n += (int64_t)gridDim.x * blockDim.x) {
-    float s = 0.f;
-    for (int64_t m = 0; m < M; ++m) s += g[m * N + n];
-    out[n] = s;
+  const int64_t n = blockIdx.x;
+  float s = 0.f;
+  for (int64_t m = threadIdx.x; m < M; m += blockDim.x)
+    s += g[m * N + n];
+  for (int off = WAVE_SIZE / 2; off > 0; off >>= 1)
+    s += __shfl_down(s, off);
+  __shared__ float part[4];
+  if ((threadIdx.x & 63) == 0) part[threadIdx.x >> 6] = s;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float t = 0.f;
+    for (int w = 0; w < (int)(blockDim.x >> 6); ++w) t += part[w];
+    out[n] = t;
   }
 }
 
@@ -332,7 +380,7 @@ torch::Tensor colsum(torch::Tensor g) {
   auto gc = g.to(torch::kFloat32).contiguous();
   int64_t M = gc.size(0), N = gc.size(1);
   auto out = torch::empty({N}, gc.options());
-  hipLaunchKernelGGL(colsum_kernel, dim3(elem_grid(N, 1)), dim3(ELEM_BLOCK),
+  hipLaunchKernelGGL(colsum_kernel, dim3((unsigned)N), dim3(ELEM_BLOCK),
                      0, current_stream(), gc.data_ptr<float>(),
                      out.data_ptr<float>(), M, N);
   return out;
