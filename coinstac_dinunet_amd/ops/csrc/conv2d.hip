@@ -516,7 +516,12 @@ __global__ __launch_bounds__(256) void conv2d_spatial_kernel(
 // WGRAD with tap reuse (2D): block stages one x slab [CT][H2][W2] + one go
 // tile [co][m] once and computes all 9 tap GEMMs from it; grid-strides
 // over (n, h-tile, w-tile) chunks; fp32 atomics fold partials into dw.
-template <int OWT, int STRIDE, int CHUNK = 128, bool FUSE_BN = false>
+// WIDE=true (stride 1 only): waves span 64 output channels (4x16) and
+// each wave computes BOTH 16-wide ci fragments — halves the go-staging
+// redundancy at the 256/512-channel stages and doubles the MFMA work
+// per barrier (acc[9][2] = 72 VGPRs).
+template <int OWT, int STRIDE, int CHUNK = 128, bool FUSE_BN = false,
+          bool WIDE = false>
 __global__ __launch_bounds__(256) void conv2d_wgrad_sp_kernel(
     const __bf16* __restrict__ x, const __bf16* __restrict__ go,
     float* __restrict__ dw, Conv2dDims cd, int64_t nchunks, int64_t zstride,
@@ -526,7 +531,8 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_sp_kernel(
   constexpr int W2 = IW + (STRIDE == 1 ? 4 : 2);
   constexpr int H2 = STRIDE * (OHT - 1) + 3;
   constexpr int CT = STRIDE == 1 ? 32 : 16;
-  constexpr int COT = STRIDE == 1 ? 32 : 64;
+  constexpr int COT = (STRIDE == 1 && !WIDE) ? 32 : 64;
+  constexpr int CIF = (STRIDE == 1 && WIDE) ? 2 : 1;  // ci frags per wave
   __shared__ __bf16 sX[CT][H2][W2];
   __shared__ __bf16 sGo[COT][CHUNK + LDA_PAD2];
 
@@ -534,16 +540,18 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_sp_kernel(
   const int ci0 = blockIdx.y * CT;
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
-  const int wi = (STRIDE == 1) ? (wave >> 1) : wave;
-  const int wj = (STRIDE == 1) ? (wave & 1) : 0;
+  const int wi = (STRIDE == 1 && !WIDE) ? (wave >> 1) : wave;
+  const int wj = (STRIDE == 1 && !WIDE) ? (wave & 1) : 0;
   const int row = lane & 15, kg = lane >> 4;
 
   const int wtiles = (cd.OW + OWT - 1) / OWT;
   const int htiles = (cd.OH + OHT - 1) / OHT;
 
-  f32x4 acc[9];
+  f32x4 acc[9][CIF];
 #pragma unroll
-  for (int t = 0; t < 9; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
+  for (int t = 0; t < 9; ++t)
+#pragma unroll
+    for (int f = 0; f < CIF; ++f) acc[t][f] = {0.f, 0.f, 0.f, 0.f};
 
   const int64_t HW = (int64_t)cd.H * cd.W;
   const int64_t OHW = (int64_t)cd.OH * cd.OW;
@@ -643,14 +651,17 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_sp_kernel(
       for (int kh = 0; kh < 3; ++kh) {
 #pragma unroll
         for (int kw = 0; kw < 3; ++kw) {
-          bf16x8 bfrag;
-          const __bf16* src = &sX[wj * 16 + row]
-                                 [STRIDE * oh_off + kh]
-                                 [STRIDE * ow_off + kw];
 #pragma unroll
-          for (int j = 0; j < 8; ++j) bfrag[j] = src[STRIDE * j];
-          acc[kh * 3 + kw] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afrag, bfrag, acc[kh * 3 + kw], 0, 0, 0);
+          for (int f = 0; f < CIF; ++f) {
+            bf16x8 bfrag;
+            const __bf16* src = &sX[(wj + f) * 16 + row]
+                                   [STRIDE * oh_off + kh]
+                                   [STRIDE * ow_off + kw];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) bfrag[j] = src[STRIDE * j];
+            acc[kh * 3 + kw][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag, bfrag, acc[kh * 3 + kw][f], 0, 0, 0);
+          }
         }
       }
     }
@@ -663,11 +674,14 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_sp_kernel(
 #pragma unroll 1
   for (int tp = 0; tp < 9; ++tp) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int co = co0 + wi * 16 + crow0 + r;
-      const int ci = ci0 + wj * 16 + ccol;
-      if (co < cd.Cout && ci < cd.Cin)
-        atomicAdd(&dw[(int64_t)co * K + ci * 9 + tp], acc[tp][r]);
+    for (int f = 0; f < CIF; ++f) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int co = co0 + wi * 16 + crow0 + r;
+        const int ci = ci0 + (wj + f) * 16 + ccol;
+        if (co < cd.Cout && ci < cd.Cin)
+          atomicAdd(&dw[(int64_t)co * K + ci * 9 + tp], acc[tp][f][r]);
+      }
     }
   }
 }
@@ -1023,7 +1037,9 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor go,
     int wtiles = (cd.OW + OWT - 1) / OWT;
     int OHT = chunk / OWT;
     int htiles = (cd.OH + OHT - 1) / OHT;
-    int COT = stride == 1 ? 32 : 64, CT = stride == 1 ? 32 : 16;
+    const bool wide = (stride == 1 && cd.Cout >= 64 && !fuse);
+    int COT = stride == 1 ? (wide ? 64 : 32) : 64;
+    int CT = stride == 1 ? 32 : 16;
     int co_t = (cd.Cout + COT - 1) / COT, ci_t = (cd.Cin + CT - 1) / CT;
     int64_t nchunks = (int64_t)cd.N * htiles * wtiles;
     int64_t zstride = std::max<int64_t>(
@@ -1036,6 +1052,7 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor go,
                          dw.data_ptr<float>(), cd, nchunks, zstride, abp);
     };
     if (fuse) {
+      // fused-BN wgrad keeps the 32-wide form (no wide instances)
       if (stride == 1) {
         if (chunk == 64) L(conv2d_wgrad_sp_kernel<8, 1, 64, true>);
         else if (OWT == 32) L(conv2d_wgrad_sp_kernel<32, 1, 128, true>);
@@ -1047,6 +1064,11 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor go,
         else if (OWT == 16) L(conv2d_wgrad_sp_kernel<16, 2, 128, true>);
         else L(conv2d_wgrad_sp_kernel<8, 2, 128, true>);
       }
+    } else if (stride == 1 && wide) {
+      if (chunk == 64) L(conv2d_wgrad_sp_kernel<8, 1, 64, false, true>);
+      else if (OWT == 32) L(conv2d_wgrad_sp_kernel<32, 1, 128, false, true>);
+      else if (OWT == 16) L(conv2d_wgrad_sp_kernel<16, 1, 128, false, true>);
+      else L(conv2d_wgrad_sp_kernel<8, 1, 128, false, true>);
     } else if (stride == 1) {
       if (chunk == 64) L(conv2d_wgrad_sp_kernel<8, 1, 64>);
       else if (OWT == 32) L(conv2d_wgrad_sp_kernel<32, 1, 128>);
