@@ -26,9 +26,6 @@ _SPATIAL_DB = os.environ.get('COINN_SPATIAL_DB', '0') == '1'
 # (45.89 ms/step — LDS doubling costs co-residency more than intra-block
 # overlap buys). Kept compiled behind the env flag.
 _WGRAD_DB = os.environ.get('COINN_WGRAD_DB', '0') == '1'
-# chunk-256 stride-1 wgrad (fewer/larger staging rounds, 1 block/CU):
-# A/B experiment, off by default.
-_WGRAD_C256 = os.environ.get('COINN_WGRAD_CHUNK256', '0') == '1'
 
 
 class _Conv3dFn(torch.autograd.Function):
@@ -85,8 +82,7 @@ class _Conv3dFn(torch.autograd.Function):
                 gx = C.conv3d_dgrad(go, wb, list(xb.shape),
                                     ctx.stride).to(ctx.in_dtype)
         if ctx.needs_input_grad[1]:
-            variant = 1 if (_WGRAD_DB and ctx.stride == 1) else \
-                (2 if (_WGRAD_C256 and ctx.stride == 1) else 0)
+            variant = 1 if (_WGRAD_DB and ctx.stride == 1) else 0
             gw = C.conv3d_wgrad(xb, go, ctx.stride,
                                 variant).to(ctx.w_dtype)
         if ctx.has_bias and ctx.needs_input_grad[2]:
@@ -148,8 +144,7 @@ class _ConvBNFn(torch.autograd.Function):
             dz = C.conv3d_dgrad_s2_spatial(go, wb, list(xb.shape))
         else:
             dz = C.conv3d_dgrad(go, wb, list(xb.shape), stride)
-        wvar = 2 if (_WGRAD_C256 and stride == 1) else 0
-        gw = C.conv3d_wgrad(xb, go, stride, wvar, ab).to(ctx.w_dtype) \
+        gw = C.conv3d_wgrad(xb, go, stride, 0, ab).to(ctx.w_dtype) \
             if ctx.needs_input_grad[4] else None
         gb = C.channel_sum(go) if (ctx.has_bias
                                    and ctx.needs_input_grad[5]) else None

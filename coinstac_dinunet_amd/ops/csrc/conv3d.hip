@@ -1037,8 +1037,7 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
   torch::Tensor ab;
   const float* abp = nullptr;
   if (fuse) {
-    TORCH_CHECK(variant == 0 || variant == 2,
-                "fused BN wgrad: variant 0 or 2 (chunk-256) only");
+    TORCH_CHECK(variant == 0, "fused BN wgrad: default variant only");
     ab = bn_ab.to(torch::kFloat32).contiguous();
     TORCH_CHECK(ab.numel() == 2 * cd.Cin, "bn_ab must be [Cin,2]");
     abp = ab.data_ptr<float>();
@@ -1069,11 +1068,7 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
                          dw.data_ptr<float>(), cd, nchunks, zstride, abp);
     };
     if (fuse) {
-      if (stride == 1 && variant == 2 && chunk == 128 && OWT == 32) {
-        // chunk-256 experiment: fewer/larger staging rounds per barrier
-        // (1 block/CU at ~86 KB LDS) — A/B via COINN_WGRAD_CHUNK256
-        L(conv3d_wgrad_s1_kernel<32, 1, 256, true>);
-      } else if (stride == 1) {
+      if (stride == 1) {
         if (chunk == 64) L(conv3d_wgrad_s1_kernel<8, 1, 64, true>);
         else if (OWT == 32) L(conv3d_wgrad_s1_kernel<32, 1, 128, true>);
         else if (OWT == 16) L(conv3d_wgrad_s1_kernel<16, 1, 128, true>);
@@ -1098,8 +1093,6 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
       else if (OWT == 32) LDB(conv3d_wgrad_s1_db_kernel<32, 1>);
       else if (OWT == 16) LDB(conv3d_wgrad_s1_db_kernel<16, 1>);
       else LDB(conv3d_wgrad_s1_db_kernel<8, 1>);
-    } else if (stride == 1 && variant == 2 && chunk == 128 && OWT == 32) {
-      L(conv3d_wgrad_s1_kernel<32, 1, 256>);
     } else if (stride == 1) {
       if (chunk == 64) L(conv3d_wgrad_s1_kernel<8, 1, 64>);
       else if (OWT == 32) L(conv3d_wgrad_s1_kernel<32, 1>);
