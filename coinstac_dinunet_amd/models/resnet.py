@@ -11,7 +11,7 @@ remain on the library path (documented scope).
 """
 import torch.nn as nn
 
-from ..ops.bnorm import OpsBatchNorm2d
+from ..ops.bnorm import OpsBatchNorm2d, bn_add_relu
 from ..ops.conv import OpsConv2d
 
 
@@ -34,10 +34,8 @@ class BasicBlock(nn.Module):
     def forward(self, x):
         idt = x if self.down is None else self.down(x)
         y = self.bn1(self.conv1(x))
-        y = self.bn2(self.conv2(y))
-        if y.dtype != idt.dtype:
-            idt = idt.to(y.dtype)
-        return self.act(y + idt)
+        # bn2 -> +identity -> ReLU fused into one pass each way
+        return bn_add_relu(self.conv2(y), idt, self.bn2)
 
 
 class ResNet18(nn.Module):

@@ -60,6 +60,57 @@ class _OpsBNMixin:
                             self.relu)
 
 
+class _BNAddReluFn(torch.autograd.Function):
+    """BN(x) + residual -> ReLU as ONE pass each way (ResNet BasicBlock
+    tail: bn2 -> +identity -> ReLU was three elementwise round-trips).
+    Backward recomputes the mask from (affine(x) + res) and emits the
+    residual-branch gradient (masked dy) from the same kernel."""
+
+    @staticmethod
+    def forward(ctx, x, res, gamma, beta, running_mean, running_var,
+                momentum, eps):
+        C = require_native()
+        xb = x.to(torch.bfloat16)
+        rb = res.to(torch.bfloat16)
+        y, mean, var, mean_rstd = C.bn3d_fwd_res(xb, rb, gamma, beta, eps)
+        if running_mean is not None:
+            with torch.no_grad():
+                n = xb.numel() // xb.size(1)
+                unbiased = var * (n / max(n - 1, 1))
+                running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
+        ctx.save_for_backward(xb, rb, mean_rstd, gamma, beta)
+        ctx.in_dtype = x.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        C = require_native()
+        xb, rb, mean_rstd, gamma, beta = ctx.saved_tensors
+        dx, dgamma, dbeta, dres = C.bn3d_bwd_res(
+            dy.to(torch.bfloat16), xb, rb, mean_rstd, gamma, beta)
+        return (dx.to(ctx.in_dtype), dres.to(ctx.in_dtype),
+                dgamma.to(gamma.dtype), dbeta.to(beta.dtype),
+                None, None, None, None)
+
+
+def bn_add_relu(x, residual, bn):
+    """relu(bn(x) + residual) fused on the HIP path (train mode);
+    falls back to the composed ops elsewhere."""
+    if x.is_cuda and native_available() and bn.training:
+        if bn.num_batches_tracked is not None:
+            bn.num_batches_tracked.add_(1)
+        if residual.dtype != torch.bfloat16:
+            residual = residual.to(torch.bfloat16)
+        return _BNAddReluFn.apply(x, residual, bn.weight, bn.bias,
+                                  bn.running_mean, bn.running_var,
+                                  bn.momentum, bn.eps)
+    y = bn(x)
+    if y.dtype != residual.dtype:
+        residual = residual.to(y.dtype)
+    return F.relu(y + residual)
+
+
 class OpsBatchNorm3d(_OpsBNMixin, nn.BatchNorm3d):
     """BatchNorm3d with optional fused ReLU; HIP kernels on GPU."""
 

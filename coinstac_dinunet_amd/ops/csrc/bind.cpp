@@ -77,6 +77,14 @@ torch::Tensor conv3d_pw_wgrad(torch::Tensor x, torch::Tensor go);
 std::vector<torch::Tensor> bn3d_fwd(torch::Tensor x, torch::Tensor gamma,
                                     torch::Tensor beta, double eps, bool relu);
 std::vector<torch::Tensor> bn3d_stats(torch::Tensor x, double eps);
+std::vector<torch::Tensor> bn3d_fwd_res(torch::Tensor x, torch::Tensor res,
+                                        torch::Tensor gamma,
+                                        torch::Tensor beta, double eps);
+std::vector<torch::Tensor> bn3d_bwd_res(torch::Tensor dy, torch::Tensor x,
+                                        torch::Tensor res,
+                                        torch::Tensor mean_rstd,
+                                        torch::Tensor gamma,
+                                        torch::Tensor beta);
 torch::Tensor bn3d_infer(torch::Tensor x, torch::Tensor gamma,
                          torch::Tensor beta, torch::Tensor running_mean,
                          torch::Tensor running_var, double eps, bool relu);
@@ -127,6 +135,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3d_pw_wgrad", &conv3d_pw_wgrad);
   m.def("bn3d_fwd", &bn3d_fwd);
   m.def("bn3d_stats", &bn3d_stats);
+  m.def("bn3d_fwd_res", &bn3d_fwd_res);
+  m.def("bn3d_bwd_res", &bn3d_bwd_res);
   m.def("bn3d_infer", &bn3d_infer);
   m.def("bn3d_bwd", &bn3d_bwd);
 }

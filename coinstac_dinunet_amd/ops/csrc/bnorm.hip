@@ -60,28 +60,35 @@ __global__ void bn_reduce_kernel(const __bf16* __restrict__ x,
   block_reduce_2(s, ss, &sums[c * 2], &sums[c * 2 + 1]);
 }
 
-// ---- fwd pass 2: y = relu((x - mean) * rstd * gamma + beta) ---------------
-template <bool RELU>
+// ---- fwd pass 2: y = relu((x - mean) * rstd * gamma + beta [+ res]) -------
+// RES: residual branch added before the ReLU — fuses ResNet's
+// bn2 -> (+identity) -> ReLU into the one normalize pass.
+template <bool RELU, bool RES = false>
 __global__ void bn_normalize_kernel(const __bf16* __restrict__ x,
                                     __bf16* __restrict__ y,
                                     const float* __restrict__ mean_rstd,
                                     const float* __restrict__ gamma,
                                     const float* __restrict__ beta, int C,
-                                    int64_t spatial) {
+                                    int64_t spatial,
+                                    const __bf16* __restrict__ res = nullptr) {
   const int nc = blockIdx.y;
   const int c = nc % C;
   const float mean = mean_rstd[c * 2], rstd = mean_rstd[c * 2 + 1];
   const float g = gamma[c] * rstd, b = beta[c] - mean * g;
   const __bf16* xp = x + (int64_t)nc * spatial;
+  const __bf16* rp = RES ? res + (int64_t)nc * spatial : nullptr;
   __bf16* yp = y + (int64_t)nc * spatial;
   int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
   for (; i + 8 <= spatial; i += stride) {
     bf16x8v v = *reinterpret_cast<const bf16x8v*>(xp + i);
+    bf16x8v rv;
+    if (RES) rv = *reinterpret_cast<const bf16x8v*>(rp + i);
     bf16x8v o;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float f = (float)v[j] * g + b;
+      if (RES) f += (float)rv[j];
       if (RELU) f = fmaxf(f, 0.f);
       o[j] = (__bf16)f;
     }
@@ -90,37 +97,45 @@ __global__ void bn_normalize_kernel(const __bf16* __restrict__ x,
   if (i < spatial)
     for (int64_t j = i; j < spatial; ++j) {
       float f = (float)xp[j] * g + b;
+      if (RES) f += (float)rp[j];
       if (RELU) f = fmaxf(f, 0.f);
       yp[j] = (__bf16)f;
     }
 }
 
 // ---- bwd pass 1: per-channel sum(dz), sum(dz * xhat) ----------------------
-template <bool RELU>
+// RES: the fused residual path — the ReLU mask is (affine(x)+res) > 0.
+template <bool RELU, bool RES = false>
 __global__ void bn_bwd_reduce_kernel(const __bf16* __restrict__ dy,
                                      const __bf16* __restrict__ x,
                                      float* __restrict__ sums,  // [C][2]
                                      const float* __restrict__ mean_rstd,
                                      const float* __restrict__ gamma,
                                      const float* __restrict__ beta, int C,
-                                     int64_t spatial) {
+                                     int64_t spatial,
+                                     const __bf16* __restrict__ res = nullptr) {
   const int nc = blockIdx.y;
   const int c = nc % C;
   const float mean = mean_rstd[c * 2], rstd = mean_rstd[c * 2 + 1];
   const float g = gamma[c], bt = beta[c];
   const __bf16* dyp = dy + (int64_t)nc * spatial;
   const __bf16* xp = x + (int64_t)nc * spatial;
+  const __bf16* rp = RES ? res + (int64_t)nc * spatial : nullptr;
   float s = 0.f, sx = 0.f;
   int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
   for (; i + 8 <= spatial; i += stride) {
     bf16x8v dv = *reinterpret_cast<const bf16x8v*>(dyp + i);
     bf16x8v xv = *reinterpret_cast<const bf16x8v*>(xp + i);
+    bf16x8v rv;
+    if (RES) rv = *reinterpret_cast<const bf16x8v*>(rp + i);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float xh = ((float)xv[j] - mean) * rstd;
       float d = (float)dv[j];
-      if (RELU && (g * xh + bt) <= 0.f) continue;
+      float z = g * xh + bt;
+      if (RES) z += (float)rv[j];
+      if (RELU && z <= 0.f) continue;
       s += d;
       sx += d * xh;
     }
@@ -129,7 +144,9 @@ __global__ void bn_bwd_reduce_kernel(const __bf16* __restrict__ dy,
     for (int64_t j = i; j < spatial; ++j) {
       float xh = ((float)xp[j] - mean) * rstd;
       float d = (float)dyp[j];
-      if (RELU && (g * xh + bt) <= 0.f) continue;
+      float z = g * xh + bt;
+      if (RES) z += (float)rp[j];
+      if (RELU && z <= 0.f) continue;
       s += d;
       sx += d * xh;
     }
@@ -137,7 +154,8 @@ __global__ void bn_bwd_reduce_kernel(const __bf16* __restrict__ dy,
 }
 
 // ---- bwd pass 2: dx = g*rstd*(dz - mean(dz) - xhat*mean(dz*xhat)) ---------
-template <bool RELU>
+// RES: also writes dres = relu-masked dy (the residual branch gradient).
+template <bool RELU, bool RES = false>
 __global__ void bn_bwd_dx_kernel(const __bf16* __restrict__ dy,
                                  const __bf16* __restrict__ x,
                                  __bf16* __restrict__ dx,
@@ -145,7 +163,9 @@ __global__ void bn_bwd_dx_kernel(const __bf16* __restrict__ dy,
                                  const float* __restrict__ sums,
                                  const float* __restrict__ gamma,
                                  const float* __restrict__ beta,
-                                 int64_t per_ch, int C, int64_t spatial) {
+                                 int64_t per_ch, int C, int64_t spatial,
+                                 const __bf16* __restrict__ res = nullptr,
+                                 __bf16* __restrict__ dres = nullptr) {
   const int nc = blockIdx.y;
   const int c = nc % C;
   const float mean = mean_rstd[c * 2], rstd = mean_rstd[c * 2 + 1];
@@ -155,27 +175,38 @@ __global__ void bn_bwd_dx_kernel(const __bf16* __restrict__ dy,
   const float scale = g * rstd;
   const __bf16* dyp = dy + (int64_t)nc * spatial;
   const __bf16* xp = x + (int64_t)nc * spatial;
+  const __bf16* rp = RES ? res + (int64_t)nc * spatial : nullptr;
+  __bf16* drp = RES ? dres + (int64_t)nc * spatial : nullptr;
   __bf16* dxp = dx + (int64_t)nc * spatial;
   int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
   for (; i + 8 <= spatial; i += stride) {
     bf16x8v dv = *reinterpret_cast<const bf16x8v*>(dyp + i);
     bf16x8v xv = *reinterpret_cast<const bf16x8v*>(xp + i);
-    bf16x8v o;
+    bf16x8v rv;
+    if (RES) rv = *reinterpret_cast<const bf16x8v*>(rp + i);
+    bf16x8v o, dr;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float xh = ((float)xv[j] - mean) * rstd;
       float d = (float)dv[j];
-      if (RELU && (g * xh + bt) <= 0.f) d = 0.f;
+      float z = g * xh + bt;
+      if (RES) z += (float)rv[j];
+      if (RELU && z <= 0.f) d = 0.f;
+      if (RES) dr[j] = (__bf16)d;
       o[j] = (__bf16)(scale * (d - m_dy - xh * m_dyxh));
     }
     *reinterpret_cast<bf16x8v*>(dxp + i) = o;
+    if (RES) *reinterpret_cast<bf16x8v*>(drp + i) = dr;
   }
   if (i < spatial)
     for (int64_t j = i; j < spatial; ++j) {
       float xh = ((float)xp[j] - mean) * rstd;
       float d = (float)dyp[j];
-      if (RELU && (g * xh + bt) <= 0.f) d = 0.f;
+      float z = g * xh + bt;
+      if (RES) z += (float)rp[j];
+      if (RELU && z <= 0.f) d = 0.f;
+      if (RES) drp[j] = (__bf16)d;
       dxp[j] = (__bf16)(scale * (d - m_dy - xh * m_dyxh));
     }
 }
@@ -218,10 +249,46 @@ std::vector<torch::Tensor> bn3d_fwd(torch::Tensor x, torch::Tensor gamma,
                        reinterpret_cast<const __bf16*>(xc.data_ptr()),
                        reinterpret_cast<__bf16*>(y.data_ptr()),
                        mean_rstd.data_ptr<float>(), g.data_ptr<float>(),
-                       b.data_ptr<float>(), C, spatial);
+                       b.data_ptr<float>(), C, spatial,
+                       (const __bf16*)nullptr);
   };
   if (relu) launch(bn_normalize_kernel<true>);
   else launch(bn_normalize_kernel<false>);
+  return {y, mean, var, mean_rstd};
+}
+
+// fused residual form: y = relu(affine(x) + res); returns stats too.
+std::vector<torch::Tensor> bn3d_fwd_res(torch::Tensor x, torch::Tensor res,
+                                        torch::Tensor gamma,
+                                        torch::Tensor beta, double eps) {
+  CHECK_GPU(x);
+  auto xc = x.contiguous();
+  auto rc = res.to(torch::kBFloat16).contiguous();
+  TORCH_CHECK(xc.scalar_type() == torch::kBFloat16, "bf16 only");
+  TORCH_CHECK(rc.sizes() == xc.sizes(), "residual shape mismatch");
+  int N = (int)xc.size(0), C = (int)xc.size(1);
+  int64_t spatial = xc.numel() / ((int64_t)N * C);
+  auto g = gamma.to(torch::kFloat32).contiguous();
+  auto b = beta.to(torch::kFloat32).contiguous();
+  auto sums = torch::zeros({C, 2}, xc.options().dtype(torch::kFloat32));
+  dim3 grid, blk;
+  bn_grid(spatial, N * C, grid, blk);
+  hipLaunchKernelGGL(bn_reduce_kernel, grid, blk, 0, current_stream(),
+                     reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                     sums.data_ptr<float>(), C, spatial);
+  int64_t per_ch = (int64_t)N * spatial;
+  auto mean = sums.select(1, 0) / (double)per_ch;
+  auto var = sums.select(1, 1) / (double)per_ch - mean * mean;
+  auto rstd = torch::rsqrt(var.clamp_min(0) + eps);
+  auto mean_rstd = torch::stack({mean, rstd}, 1).contiguous();
+  auto y = torch::empty_like(xc);
+  hipLaunchKernelGGL((bn_normalize_kernel<true, true>), grid, blk, 0,
+                     current_stream(),
+                     reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                     reinterpret_cast<__bf16*>(y.data_ptr()),
+                     mean_rstd.data_ptr<float>(), g.data_ptr<float>(),
+                     b.data_ptr<float>(), C, spatial,
+                     reinterpret_cast<const __bf16*>(rc.data_ptr()));
   return {y, mean, var, mean_rstd};
 }
 
@@ -267,7 +334,8 @@ torch::Tensor bn3d_infer(torch::Tensor x, torch::Tensor gamma,
                        reinterpret_cast<const __bf16*>(xc.data_ptr()),
                        reinterpret_cast<__bf16*>(y.data_ptr()),
                        mean_rstd.data_ptr<float>(), g.data_ptr<float>(),
-                       b.data_ptr<float>(), C, spatial);
+                       b.data_ptr<float>(), C, spatial,
+                       (const __bf16*)nullptr);
   };
   if (relu) launch(bn_normalize_kernel<true>);
   else launch(bn_normalize_kernel<false>);
@@ -294,7 +362,8 @@ std::vector<torch::Tensor> bn3d_bwd(torch::Tensor dy, torch::Tensor x,
                        reinterpret_cast<const __bf16*>(dyc.data_ptr()),
                        reinterpret_cast<const __bf16*>(xc.data_ptr()),
                        sums.data_ptr<float>(), mean_rstd.data_ptr<float>(),
-                       g.data_ptr<float>(), b.data_ptr<float>(), C, spatial);
+                       g.data_ptr<float>(), b.data_ptr<float>(), C, spatial,
+                       (const __bf16*)nullptr);
   };
   if (relu) launch_r(bn_bwd_reduce_kernel<true>);
   else launch_r(bn_bwd_reduce_kernel<false>);
@@ -308,7 +377,7 @@ std::vector<torch::Tensor> bn3d_bwd(torch::Tensor dy, torch::Tensor x,
                        reinterpret_cast<__bf16*>(dx.data_ptr()),
                        mean_rstd.data_ptr<float>(), sums.data_ptr<float>(),
                        g.data_ptr<float>(), b.data_ptr<float>(), per_ch, C,
-                       spatial);
+                       spatial, (const __bf16*)nullptr, (__bf16*)nullptr);
   };
   if (relu) launch_d(bn_bwd_dx_kernel<true>);
   else launch_d(bn_bwd_dx_kernel<false>);
@@ -316,4 +385,46 @@ std::vector<torch::Tensor> bn3d_bwd(torch::Tensor dy, torch::Tensor x,
   auto dbeta = sums.select(1, 0).clone();
   auto dgamma = sums.select(1, 1).clone();
   return {dx, dgamma, dbeta};
+}
+
+// backward of the fused residual form; extra output dres = masked dy.
+std::vector<torch::Tensor> bn3d_bwd_res(torch::Tensor dy, torch::Tensor x,
+                                        torch::Tensor res,
+                                        torch::Tensor mean_rstd,
+                                        torch::Tensor gamma,
+                                        torch::Tensor beta) {
+  CHECK_GPU(dy);
+  auto dyc = dy.contiguous();
+  auto xc = x.contiguous();
+  auto rc = res.contiguous();
+  int N = (int)dyc.size(0), C = (int)dyc.size(1);
+  int64_t spatial = dyc.numel() / ((int64_t)N * C);
+  auto g = gamma.to(torch::kFloat32).contiguous();
+  auto b = beta.to(torch::kFloat32).contiguous();
+  auto sums = torch::zeros({C, 2}, dyc.options().dtype(torch::kFloat32));
+  dim3 grid, blk;
+  bn_grid(spatial, N * C, grid, blk);
+  hipLaunchKernelGGL((bn_bwd_reduce_kernel<true, true>), grid, blk, 0,
+                     current_stream(),
+                     reinterpret_cast<const __bf16*>(dyc.data_ptr()),
+                     reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                     sums.data_ptr<float>(), mean_rstd.data_ptr<float>(),
+                     g.data_ptr<float>(), b.data_ptr<float>(), C, spatial,
+                     reinterpret_cast<const __bf16*>(rc.data_ptr()));
+  auto dx = torch::empty_like(dyc);
+  auto dres = torch::empty_like(dyc);
+  int64_t per_ch = (int64_t)N * spatial;
+  hipLaunchKernelGGL((bn_bwd_dx_kernel<true, true>), grid, blk, 0,
+                     current_stream(),
+                     reinterpret_cast<const __bf16*>(dyc.data_ptr()),
+                     reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                     reinterpret_cast<__bf16*>(dx.data_ptr()),
+                     mean_rstd.data_ptr<float>(), sums.data_ptr<float>(),
+                     g.data_ptr<float>(), b.data_ptr<float>(), per_ch, C,
+                     spatial,
+                     reinterpret_cast<const __bf16*>(rc.data_ptr()),
+                     reinterpret_cast<__bf16*>(dres.data_ptr()));
+  auto dbeta = sums.select(1, 0).clone();
+  auto dgamma = sums.select(1, 1).clone();
+  return {dx, dgamma, dbeta, dres};
 }

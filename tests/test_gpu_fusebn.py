@@ -127,3 +127,38 @@ def test_vbmnet_fused_chain_trains(dev):
     assert losses[-1] < losses[0] * 0.7, losses[::6]
     for n, p in net.named_parameters():
         assert p.grad is not None, n
+
+
+def test_bn_add_relu_matches_composed(dev):
+    """relu(bn(x)+res) fused (one pass each way) vs the composed ops."""
+    from coinstac_dinunet_amd.ops.bnorm import OpsBatchNorm2d, bn_add_relu
+    torch.manual_seed(77)
+    bn = OpsBatchNorm2d(32).to(dev)
+    with torch.no_grad():
+        bn.weight.mul_(0).add_(torch.rand(32, device=dev) + 0.5)
+        bn.bias.add_(torch.randn(32, device=dev) * 0.1)
+    x0 = torch.randn(4, 32, 14, 14, device=dev, dtype=torch.bfloat16)
+    r0 = torch.randn(4, 32, 14, 14, device=dev, dtype=torch.bfloat16)
+    go = torch.randn(4, 32, 14, 14, device=dev, dtype=torch.bfloat16) * 0.1
+
+    def run(fused):
+        bn.running_mean.zero_(), bn.running_var.fill_(1.0)
+        bn.train()
+        x = x0.clone().requires_grad_(True)
+        r = r0.clone().requires_grad_(True)
+        bn.weight.grad = bn.bias.grad = None
+        if fused:
+            y = bn_add_relu(x, r, bn)
+        else:
+            y = torch.relu(bn(x) + r)
+        y.backward(go)
+        return (y.float(), x.grad.float(), r.grad.float(),
+                bn.weight.grad.clone(), bn.bias.grad.clone())
+
+    yf, gxf, grf, ggf, gbf = run(True)
+    yu, gxu, gru, ggu, gbu = run(False)
+    torch.testing.assert_close(yf, yu, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(gxf, gxu, rtol=5e-2, atol=2e-2)
+    torch.testing.assert_close(grf, gru, rtol=5e-2, atol=2e-2)
+    torch.testing.assert_close(ggf, ggu, rtol=5e-2, atol=0.3)
+    torch.testing.assert_close(gbf, gbu, rtol=5e-2, atol=0.3)
