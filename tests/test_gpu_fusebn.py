@@ -158,7 +158,18 @@ def test_bn_add_relu_matches_composed(dev):
     yf, gxf, grf, ggf, gbf = run(True)
     yu, gxu, gru, ggu, gbu = run(False)
     torch.testing.assert_close(yf, yu, rtol=2e-2, atol=2e-2)
-    torch.testing.assert_close(gxf, gxu, rtol=5e-2, atol=2e-2)
-    torch.testing.assert_close(grf, gru, rtol=5e-2, atol=2e-2)
+    # the ReLU mask differs where affine(x)+res ~ 0 (fused masks on the
+    # fp32 sum, composed on the bf16-rounded sum) — exclude the boundary
+    with torch.no_grad():
+        xf = x0.float()
+        mu = xf.mean((0, 2, 3), keepdim=True)
+        var = xf.var((0, 2, 3), unbiased=False, keepdim=True)
+        z = ((xf - mu) * torch.rsqrt(var + bn.eps)
+             * bn.weight.view(1, -1, 1, 1)
+             + bn.bias.view(1, -1, 1, 1) + r0.float()).abs()
+        interior = (z > 0.05).float()
+    for a, b in [(gxf, gxu), (grf, gru)]:
+        torch.testing.assert_close(a * interior, b * interior,
+                                   rtol=5e-2, atol=2e-2)
     torch.testing.assert_close(ggf, ggu, rtol=5e-2, atol=0.3)
     torch.testing.assert_close(gbf, gbu, rtol=5e-2, atol=0.3)
