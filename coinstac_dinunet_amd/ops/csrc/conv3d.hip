@@ -413,6 +413,165 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// WGRAD, Cin=1 specialization (the VBM first layer): the split-K form
+// gathers each x element once per tap column (27x traffic, 4-element
+// scalar runs). Here a block owns a group of (n, od) slices: the single-
+// channel x slab [3][OH+2][OW+2] stages ONCE per slice with 16B row
+// vectors, go tiles [co][OHT*OW] stream through LDS, and the 27 taps form
+// the MFMA B fragments read directly from the slab (taps = fragment rows,
+// m = contraction). Per-block LDS reduction + one atomicAdd set per block
+// keeps contention at blocks x Cout x 27.
+// Requires: Cin==1, stride==1, OW % 32 == 0.
+// ---------------------------------------------------------------------------
+#define CI1_OHT 8
+
+__global__ __launch_bounds__(256) void conv3d_wgrad_ci1_kernel(
+    const __bf16* __restrict__ x, const __bf16* __restrict__ go,
+    float* __restrict__ dw, ConvDims cd, int od_groups) {
+  const int W2 = cd.W + 2;
+  const int H2 = cd.H + 2;
+  extern __shared__ __bf16 smem_ci1[];
+  __bf16* sX = smem_ci1;                       // [3][H2][W2]
+  __bf16* sGo = sX + 3 * H2 * W2;              // [32][CI1_OHT * OW + 8]
+  const int GOL = CI1_OHT * cd.OW + 8;
+
+  const int n = blockIdx.x;
+  const int og = blockIdx.y;
+  const int co0 = blockIdx.z * 32;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int row = lane & 15, kg = lane >> 4;
+
+  const int64_t HW = (int64_t)cd.H * cd.W;
+  const int64_t OHW = (int64_t)cd.OH * cd.OW;
+  const int od_per = (cd.OD + od_groups - 1) / od_groups;
+  const int od_lo = og * od_per;
+  const int od_hi = min(cd.OD, od_lo + od_per);
+
+  // acc[co frag][tap frag]: (16 co x 16 taps) x 2 x 2 per wave
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int od = od_lo; od < od_hi; ++od) {
+    // ---- stage the x slab for this od: rows (kd, ih) -------------------
+    __syncthreads();
+    for (int r = tid; r < 3 * H2; r += 256) {
+      const int kd = r / H2;
+      const int ihp = r % H2;            // padded row index: ih = ihp - 1
+      const int id = od - 1 + kd;
+      const int ih = ihp - 1;
+      __bf16* dst = sX + ((int64_t)kd * H2 + ihp) * W2;
+      if ((unsigned)id >= (unsigned)cd.D || (unsigned)ih >= (unsigned)cd.H) {
+        for (int c = 0; c < W2; ++c) dst[c] = (__bf16)0.f;
+        continue;
+      }
+      const __bf16* src = x + ((int64_t)n * cd.D + id) * HW +
+                          (int64_t)ih * cd.W;
+      dst[0] = (__bf16)0.f;
+      for (int v = 0; v + 8 <= cd.W; v += 8)
+        *reinterpret_cast<bf16x8*>(dst + 1 + v) =
+            *reinterpret_cast<const bf16x8*>(src + v);
+      for (int c = (cd.W / 8) * 8; c < cd.W; ++c) dst[1 + c] = src[c];
+      dst[1 + cd.W] = (__bf16)0.f;
+    }
+
+    for (int oh0 = 0; oh0 < cd.OH; oh0 += CI1_OHT) {
+      // ---- stage go tile [co32][CI1_OHT * OW] --------------------------
+      __syncthreads();
+      for (int r = tid; r < 32 * CI1_OHT; r += 256) {
+        const int ohl = r % CI1_OHT;
+        const int co = r / CI1_OHT;
+        __bf16* dst = sGo + (int64_t)co * GOL + ohl * cd.OW;
+        const int oh = oh0 + ohl;
+        if ((co0 + co) >= cd.Cout || oh >= cd.OH) {
+          for (int c = 0; c < cd.OW; ++c) dst[c] = (__bf16)0.f;
+          continue;
+        }
+        const __bf16* src = go + (((int64_t)n * cd.Cout + co0 + co) *
+                                  cd.OD + od) * OHW + (int64_t)oh * cd.OW;
+        for (int v = 0; v < cd.OW; v += 8)
+          *reinterpret_cast<bf16x8*>(dst + v) =
+              *reinterpret_cast<const bf16x8*>(src + v);
+      }
+      __syncthreads();
+
+      // ---- MFMA: waves stride the m subchunks --------------------------
+      const int nms = (CI1_OHT * cd.OW) / 32;
+      for (int ms = wave; ms < nms; ms += 4) {
+        const int m0 = ms * 32 + kg * 8;     // 8 consecutive m, no row wrap
+        const int ohl = m0 / cd.OW;
+        const int ow0 = m0 % cd.OW;
+        bf16x8 afrag[2];
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+          const __bf16* src = sGo + (int64_t)(i * 16 + row) * GOL + m0;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) afrag[i][j] = src[j];
+        }
+        bf16x8 bfrag[2];
+#pragma unroll
+        for (int t = 0; t < 2; ++t) {
+          const int tap = t * 16 + row;      // 0..31; taps 27..31 padded
+          const int kd = tap / 9, r9 = tap - kd * 9;
+          const int kh = r9 / 3, kw = r9 % 3;
+          if (tap < 27) {
+            const __bf16* src = sX + ((int64_t)kd * H2 + (oh0 + ohl + kh)) *
+                                W2 + ow0 + kw;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) bfrag[t][j] = src[j];
+          } else {
+            bfrag[t] = bf16x8{};
+          }
+        }
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+          for (int t = 0; t < 2; ++t)
+            acc[i][t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[i], bfrag[t], acc[i][t], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- cross-wave reduce through LDS, one atomic set per block ----------
+  __syncthreads();
+  float* red = reinterpret_cast<float*>(smem_ci1);  // reuse: 4*4*256 floats
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      const int fi = i * 2 + t;
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        red[((int64_t)wave * 4 + fi) * 256 + lane * 4 + r] = acc[i][t][r];
+    }
+  __syncthreads();
+  if (wave == 0) {
+    // lane l, reg r of fragment fi: co = i*16 + (l>>4)*4+r? no — C layout:
+    // row = (l>>4)*4+r (co), col = l&15 (tap)
+#pragma unroll
+    for (int fi = 0; fi < 4; ++fi) {
+      const int i = fi >> 1, t = fi & 1;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float v = 0.f;
+#pragma unroll
+        for (int w = 0; w < 4; ++w)
+          v += red[((int64_t)w * 4 + fi) * 256 + lane * 4 + r];
+        const int co = co0 + i * 16 + (lane >> 4) * 4 + r;
+        const int tap = t * 16 + (lane & 15);
+        if (co < cd.Cout && tap < 27)
+          atomicAdd(&dw[(int64_t)co * 27 + tap], v);
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Stride-2 DGRAD, parity-decomposed: the dense mask formulation wastes 7/8
 // of the MFMA work (only taps with (id+1-kd) even contribute). Decompose
@@ -1043,6 +1202,22 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
     abp = ab.data_ptr<float>();
   }
 
+  if (cd.Cin == 1 && stride == 1 && cd.OW % 32 == 0 && !fuse) {
+    // Cin=1 first-layer specialization (slab tap-reuse, see kernel doc)
+    const int od_groups = std::max(1, std::min(cd.OD, 512 / cd.N));
+    const int W2 = cd.W + 2, H2 = cd.H + 2;
+    const int GOL = CI1_OHT * cd.OW + 8;
+    size_t lds = (size_t)(3 * H2 * W2 + 32 * GOL) * sizeof(__bf16);
+    if (lds <= 64 * 1024) {
+      dim3 grid(cd.N, od_groups, (cd.Cout + 31) / 32);
+      hipLaunchKernelGGL(conv3d_wgrad_ci1_kernel, grid, dim3(256), lds,
+                         current_stream(),
+                         reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                         reinterpret_cast<const __bf16*>(g.data_ptr()),
+                         dw.data_ptr<float>(), cd, od_groups);
+      return dw.view({cd.Cout, cd.Cin, 3, 3, 3});
+    }
+  }
   if ((cd.OW % 8) == 0 && cd.Cin >= 16 && cd.OH * cd.OW >= 64 &&
       (stride == 1 || stride == 2)) {
     // tap-reuse path (stride-templated)

@@ -179,3 +179,23 @@ def test_conv3d_wgrad_double_buffered(dev, case):
                                atol=5e-2 * m * 0.1)
     torch.testing.assert_close(dw_db.float(), dw_sb.float(), rtol=2e-2,
                                atol=2e-2 * m * 0.1)
+
+
+@pytest.mark.parametrize('case', [(2, 32, 8, 16, 32),
+                                  (1, 16, 6, 10, 64),
+                                  (1, 8, 9, 7, 32)])   # OH % OHT != 0
+def test_conv3d_wgrad_ci1_specialized(dev, case):
+    """Cin=1 slab tap-reuse wgrad (first-layer specialization) vs torch
+    fp32; exercises the dedicated conv3d_wgrad_ci1_kernel routing."""
+    N, Co, D, H, W = case
+    torch.manual_seed(12)
+    x = torch.randn(N, 1, D, H, W, device=dev, dtype=torch.bfloat16)
+    go = torch.randn(N, Co, D, H, W, device=dev, dtype=torch.bfloat16) * 0.1
+    go[0, 0, 0, 0, 3] += 1.5
+    dw = C.conv3d_wgrad(x, go, 1, 0)
+    xf = x.float().requires_grad_(True)
+    w0 = torch.zeros(Co, 1, 3, 3, 3, device=dev, requires_grad=True)
+    torch.nn.functional.conv3d(xf, w0, padding=1).backward(go.float())
+    m = (N * D * H * W) ** 0.5
+    torch.testing.assert_close(dw.float(), w0.grad, rtol=5e-2,
+                               atol=5e-2 * m * 0.1)
