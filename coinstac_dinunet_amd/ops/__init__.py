@@ -3,15 +3,21 @@
 Kernel inventory (MI355X equivalents of the ops the reference implicitly
 runs through stock PyTorch — SURVEY.md §2.9):
   - Conv3d fwd/dgrad/wgrad (K1): MFMA spatial-slab tap-reuse kernels
-    (stride-1 + stride-2 incl. parity-decomposed dgrad) with implicit-GEMM
-    fallbacks for odd shapes — conv3d_spatial.hip / conv3d.hip
-  - fused BatchNorm3d(+ReLU) fwd/bwd              — bnorm.hip
-  - fused multi-tensor Adam / SGD step            (K4) — adam.hip
+    (stride-1 + stride-2 incl. parity-decomposed dgrad, fused-BN
+    normalize-on-load instances, Cin=1 wgrad specialization) with
+    implicit-GEMM fallbacks — conv3d_spatial.hip / conv3d.hip
+  - Conv2d fwd/dgrad/wgrad (ResNet): igemm (3x3 + 7x7 stem), spatial
+    slab, tap-reuse + split-K wgrad, parity s2 dgrad — conv2d.hip
+  - pointwise 1x1 convs as batched MFMA GEMMs     — pointwise.hip
+  - fused BatchNorm(+ReLU[+residual]) fwd/bwd, stats-only — bnorm.hip
+  - single-launch multi-tensor Adam / SGD         (K4) — adam.hip
   - flat gradient bucket pack / unpack            (K5/K7) — pack.hip
   - fused log_softmax + NLL loss fwd/bwd + argmax (K3/K16) — lsnll.hip
   - Prf1a confusion counts / KxK histogram        (K12/K13) — metrics.hip
-  - MFMA linear with fused bias+ReLU              (K2) — linear.hip
+  - MFMA linear fwd (fused bias+ReLU) + dgrad + split-K wgrad (K2),
+    parallel colsum — linear.hip
   - fused Gram-Schmidt for PowerSGD               (K8) — linear.hip
+  - fused deflated power iteration + rowsum       (K10/K11) — rankdad.hip
 
 The extension is built in-tree (`python setup.py build_ext --inplace` or
 __graft_entry__.build()) for gfx950 only. On a GPU box the native path is

@@ -55,7 +55,6 @@ class _Conv3dFn(torch.autograd.Function):
         ctx.save_for_backward(xb, wb)
         ctx.stride = stride
         ctx.has_bias = bias is not None
-        ctx.x_requires = x.requires_grad
         ctx.in_dtype = x.dtype
         ctx.w_dtype = weight.dtype
         return out

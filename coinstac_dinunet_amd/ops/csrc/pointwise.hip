@@ -1,14 +1,15 @@
-// Pointwise (1x1x1) convolution kernels — the UNet3D segmentation head and
-// any channel-mixing conv. A 1x1 conv in NCDHW is a per-position channel
-// GEMM: out[n,co,s] = sum_ci w[co,ci] * x[n,ci,s]. At the shapes that hit
-// this path (head: Cin<=64, Cout=num_class) the op is purely HBM-bound —
-// read x once, write out once — so the kernels are bandwidth-shaped
-// (coalesced along s, weights staged in LDS, fp32 accumulate), not MFMA:
-// the arithmetic intensity (Cout MACs per loaded element, Cout ~ 2) is far
-// below the memory roofline crossover.
-// Reference counterpart: the stock nn.Conv3d head the external UNet
-// computations use (reference utils/tensorutils.py:10-25 safe_concat and
-// metrics/loss.py:1 dice_loss exist to serve that model family).
+// Pointwise (1x1) convolution kernels — 2D and 3D channel-mixing convs
+// (ResNet downsamples, the UNet3D segmentation head). A 1x1 conv in
+// channels-first layout is, per batch element, a plain GEMM over the
+// channel dim:
+//   fwd  : out_n[Co, S] = W[Co, Ci] @ x_n[Ci, S]
+//   dgrad: gx_n[Ci, S]  = W^T       @ go_n[Co, S]
+//   wgrad: gw[Co, Ci]   = sum_n go_n @ x_n^T
+// All three run as batched MFMA GEMMs (weights shared across the batch,
+// grid = row-tiles x col-tiles x batch). An earlier streaming form that
+// parallelized only over spatial positions collapsed to a handful of
+// blocks at ResNet downsample shapes (S as small as 49) and left the
+// chip ~1% occupied (r2 profile: 629 of 774 ms) — see git history.
 #include "common.h"
 
 // out[n,co,s] (+= bias) from x[n,ci,s]; TRANSPOSE_W=true computes
@@ -17,123 +18,9 @@
 // Each thread accumulates 4 spatial points (spaced blockDim apart so each
 // of the 4 loads per ci stays coalesced) — 4 FMAs per LDS weight read
 // instead of 1; co_chunk <= 16 keeps acc in 64 VGPRs.
-#define PW_SPT 4
-template <bool TRANSPOSE_W>
-__global__ void pw_conv_kernel(const __bf16* __restrict__ x,
-                               const __bf16* __restrict__ w,
-                               const float* __restrict__ bias,
-                               __bf16* __restrict__ out, int N, int Cin,
-                               int Cout, int64_t S, int co0, int co_chunk) {
-  extern __shared__ __bf16 wlds[];
-  for (int i = threadIdx.x; i < co_chunk * Cin; i += blockDim.x) {
-    int co = co0 + i / Cin, ci = i % Cin;
-    wlds[i] = TRANSPOSE_W ? w[(int64_t)ci * Cout + co]
-                          : w[(int64_t)co * Cin + ci];
-  }
-  __syncthreads();
-
-  const int64_t total = (int64_t)N * S;
-  const int64_t wstep = (int64_t)gridDim.x * blockDim.x * PW_SPT;
-  for (int64_t w0 = (int64_t)blockIdx.x * blockDim.x * PW_SPT; w0 < total;
-       w0 += wstep) {
-    int64_t nn[PW_SPT], ss[PW_SPT];
-    bool ok[PW_SPT];
-#pragma unroll
-    for (int p = 0; p < PW_SPT; ++p) {
-      const int64_t idx = w0 + threadIdx.x + (int64_t)p * blockDim.x;
-      ok[p] = idx < total;
-      nn[p] = ok[p] ? idx / S : 0;
-      ss[p] = ok[p] ? idx % S : 0;
-    }
-    float acc[16][PW_SPT];
-#pragma unroll
-    for (int j = 0; j < 16; ++j)
-#pragma unroll
-      for (int p = 0; p < PW_SPT; ++p) acc[j][p] = 0.f;
-    for (int ci = 0; ci < Cin; ++ci) {
-      float xv[PW_SPT];
-#pragma unroll
-      for (int p = 0; p < PW_SPT; ++p)
-        xv[p] = ok[p] ? (float)x[(nn[p] * Cin + ci) * S + ss[p]] : 0.f;
-      for (int j = 0; j < co_chunk; ++j) {
-        const float wv = (float)wlds[j * Cin + ci];
-#pragma unroll
-        for (int p = 0; p < PW_SPT; ++p) acc[j][p] += wv * xv[p];
-      }
-    }
-    for (int j = 0; j < co_chunk; ++j) {
-      const float bv = (bias != nullptr) ? bias[co0 + j] : 0.f;
-#pragma unroll
-      for (int p = 0; p < PW_SPT; ++p)
-        if (ok[p])
-          out[(nn[p] * Cout + co0 + j) * S + ss[p]] =
-              (__bf16)(acc[j][p] + bv);
-    }
-  }
-}
-
-// gw[co,ci] = sum_{n,s} go[n,co,s] * x[n,ci,s]. Output is tiny (Co x Ci);
-// each block accumulates a per-block fp32 partial tile over its slice of
-// (n,s) in registers, reduces through LDS, then atomically adds into gw.
-// CO_T x CI_T accumulators per thread; grid.y tiles (co, ci) chunks.
-template <int CO_T, int CI_T>
-__global__ void pw_wgrad_kernel(const __bf16* __restrict__ x,
-                                const __bf16* __restrict__ go,
-                                float* __restrict__ gw, int N, int Cin,
-                                int Cout, int64_t S) {
-  const int nco = (Cout + CO_T - 1) / CO_T;
-  const int co0 = (blockIdx.y % nco) * CO_T;
-  const int ci0 = (blockIdx.y / nco) * CI_T;
-  const int cot = min(CO_T, Cout - co0), cit = min(CI_T, Cin - ci0);
-
-  float acc[CO_T][CI_T];
-#pragma unroll
-  for (int a = 0; a < CO_T; ++a)
-#pragma unroll
-    for (int b = 0; b < CI_T; ++b) acc[a][b] = 0.f;
-
-  const int64_t total = (int64_t)N * S;
-  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
-    const int64_t n = idx / S, s = idx % S;
-    float gv[CO_T], xv[CI_T];
-    for (int a = 0; a < cot; ++a)
-      gv[a] = (float)go[((int64_t)n * Cout + co0 + a) * S + s];
-    for (int b = 0; b < cit; ++b)
-      xv[b] = (float)x[((int64_t)n * Cin + ci0 + b) * S + s];
-    for (int a = 0; a < cot; ++a)
-      for (int b = 0; b < cit; ++b) acc[a][b] += gv[a] * xv[b];
-  }
-
-  // block reduction: waves fold their lanes, wave leaders sum in LDS
-  __shared__ float part[4][CO_T][CI_T];
-  for (int a = 0; a < cot; ++a)
-    for (int b = 0; b < cit; ++b) {
-      float v = acc[a][b];
-      for (int off = WAVE_SIZE / 2; off > 0; off >>= 1)
-        v += __shfl_down(v, off);
-      if ((threadIdx.x & 63) == 0) part[threadIdx.x >> 6][a][b] = v;
-    }
-  __syncthreads();
-  if (threadIdx.x < (unsigned)(cot * cit)) {
-    const int a = threadIdx.x / cit, b = threadIdx.x % cit;
-    float t = 0.f;
-    for (int wv = 0; wv < (int)(blockDim.x >> 6); ++wv) t += part[wv][a][b];
-    atomicAdd(&gw[(int64_t)(co0 + a) * Cin + ci0 + b], t);
-  }
-}
-
 // ---------------------------------------------------------------------------
-// Batched MFMA GEMM for the channel-mixing form: per batch n,
-//   fwd  : out_n[Co, S] = W[Co, Ci]      @ x_n[Ci, S]
-//   dgrad: gx_n[Ci, S]  = W^T[Ci, Co]    @ go_n[Co, S]
-// One kernel: C_n[i, j] = sum_c opA(i, c) * B_n[c, j], weights shared
-// across the batch (strideA = 0), grid.z = n. This replaces the
-// streaming pw kernel for real channel counts: that kernel's grid
-// collapses to a handful of blocks at ResNet downsample shapes (S as
-// small as 49) and left the chip ~1% occupied (r2 profile: 629 of
-// 774 ms in pw_conv) — the MFMA form gets its parallelism from
-// (row tiles x col tiles x batch).
+// Batched MFMA GEMM: C_n[i, j] = sum_c opA(i, c) * B_n[c, j], weights
+// shared across the batch, grid.z = n.
 // ---------------------------------------------------------------------------
 #include <hip/hip_bf16.h>
 typedef __attribute__((ext_vector_type(8))) __bf16 pwbf16x8;
@@ -312,12 +199,6 @@ __global__ __launch_bounds__(256) void pw_wgrad_batched_kernel(
 // ---------------------------------------------------------------------------
 // host wrappers
 // ---------------------------------------------------------------------------
-static int pw_grid(int64_t total) {
-  int64_t blocks = (total + ELEM_BLOCK * 4 - 1) / (ELEM_BLOCK * 4);
-  if (blocks > 4096) blocks = 4096;  // grid-stride covers the rest
-  if (blocks < 1) blocks = 1;
-  return (int)blocks;
-}
 
 torch::Tensor conv3d_pw_fwd(torch::Tensor x, torch::Tensor w,
                             torch::Tensor bias) {
