@@ -31,9 +31,13 @@ from ..distrib.learner import COINNLearner
 from ..distrib.reducer import COINNReducer
 
 # xGMI: 7 p2p links x ~153 GB/s; ring all-reduce is per-link bound. Big
-# buckets amortize launch + ring latency; 50 MB keeps >= 4 buckets in
-# flight for typical CNNs so comm still overlaps backward.
-DEFAULT_BUCKET_BYTES = 50 * 1024 * 1024
+# buckets amortize launch + ring latency, but the arena must split into
+# >= ~4 buckets or nothing overlaps backward (the r1 50 MB default put
+# every BASELINE model in ONE bucket: VBM's whole arena is 14 MB). The
+# default is now adaptive: arena/4, clamped to [2 MB, 50 MB].
+DEFAULT_BUCKET_BYTES = None
+_BUCKET_MIN = 2 * 1024 * 1024
+_BUCKET_MAX = 50 * 1024 * 1024
 
 
 def init_distributed(backend=None, timeout_s=300):
@@ -76,6 +80,11 @@ class FlatGradBuffer:
         self.params = [p for p in params if p.requires_grad]
         self.world_size = world_size
         self.comm_stream = comm_stream
+        if bucket_bytes is None:
+            total_bytes = sum(p.numel() for p in self.params) * 4
+            bucket_bytes = min(max(total_bytes // 4, _BUCKET_MIN),
+                               _BUCKET_MAX)
+        self.bucket_bytes = bucket_bytes
         # evidence/debug knob: run the collectives even at world 1 (a
         # 1-rank RCCL all-reduce launches real RCCL kernels — used by
         # tools/r2_multirank.sh to trace RCCL overlapping backward)
