@@ -32,6 +32,10 @@ def parse_args():
                     choices=['vbm', 'mlp', 'resnet18'])
     ap.add_argument('--stock', action='store_true',
                     help='stock torch ops (MIOpen conv/BN) for comparison')
+    ap.add_argument('--graph', type=int, default=1, choices=[0, 1],
+                    help='hipGraph-capture the fwd+loss+bwd inner loop '
+                         '(single-rank only; collectives and the optimizer '
+                         'stay outside the capture)')
     return ap.parse_args()
 
 
@@ -118,23 +122,46 @@ def main():
 
     import torch.distributed as dist
 
-    def one_step():
+    def fwd_bwd(sync):
         buf.zero_()
+        buf.begin_round(sync=sync)
+        if on_gpu:
+            with torch.autocast('cuda', dtype=torch.bfloat16):
+                out = net(data)
+            loss = ops.cross_entropy(out.float(), labels)
+        else:
+            loss = ops.cross_entropy(net(data), labels)
+        loss.backward()
+
+    def one_step():
         for li in range(args.local_iterations):
-            buf.begin_round(sync=(li == args.local_iterations - 1))
-            if on_gpu:
-                with torch.autocast('cuda', dtype=torch.bfloat16):
-                    out = net(data)
-                loss = ops.cross_entropy(out.float(), labels)
-            else:
-                loss = ops.cross_entropy(net(data), labels)
-            loss.backward()
+            fwd_bwd(sync=(li == args.local_iterations - 1))
         buf.finish_round()
         opt.step()
 
-    # warmup
+    # warmup (also primes autotuners before any graph capture)
     for _ in range(args.warmup):
         one_step()
+
+    # hipGraph capture of the launch-bound fwd+loss+bwd region: ResNet-18
+    # dispatches ~1100 kernels/step, VBM ~465 — replay removes the launch
+    # gaps. Single-rank + single-micro-batch only (collectives inside a
+    # capture need a cooperating RCCL setup; optimizer stays eager so its
+    # per-step bias correction keeps advancing).
+    graphed = (args.graph and on_gpu and world == 1
+               and args.local_iterations == 1)
+    if graphed:
+        g = torch.cuda.CUDAGraph()
+        torch.cuda.synchronize()
+        with torch.cuda.graph(g):
+            fwd_bwd(sync=False)
+
+        def one_step():  # noqa: F811 — replay + eager optimizer
+            g.replay()
+            opt.step()
+
+        for _ in range(3):  # graph warmup
+            one_step()
 
     if dist.is_initialized():
         dist.barrier()
@@ -182,6 +209,7 @@ def main():
                 'volume': f'{args.vol}^3' if args.model == 'vbm' else None,
                 'local_iterations': args.local_iterations,
                 'parallelism': f'dsgd-dp{world}',
+                'hipgraph': bool(graphed),
                 # BASELINE.json names "wall-clock/epoch" too: derived for a
                 # nominal 1024-sample per-site epoch at this step time
                 'wallclock_per_epoch_s_1024spp': round(
