@@ -702,26 +702,37 @@ torch::Tensor conv3d_dgrad_s2_spatial(torch::Tensor go, torch::Tensor w,
   sd.TD = (int)in_shape[2]; sd.TH = (int)in_shape[3];
   sd.TW = (int)in_shape[4];
 
-  // per-class tap-gathered weights: WBc[ci][co*T + r]
+  // per-class tap-gathered weights: WBc[ci][co*T + r]. The tap index
+  // tensors are compile-time constants — cached per device so no H2D
+  // copy happens on the hot path (pageable H2D is illegal inside a
+  // hipGraph capture).
   auto w3 = wc.reshape({Cout, Cin, 27}).permute({1, 0, 2}).contiguous();
+  static std::vector<torch::Tensor> s_idx;       // [8], device-resident
+  static torch::Device s_dev = torch::kCPU;
+  if (s_idx.empty() || s_dev != wc.device()) {
+    s_idx.clear();
+    for (int cls = 0; cls < 8; ++cls) {
+      int a = (cls >> 2) & 1, b = (cls >> 1) & 1, c = cls & 1;
+      std::vector<int64_t> taps;
+      for (int td_i = 0; td_i < (a ? 2 : 1); ++td_i)
+        for (int th_i = 0; th_i < (b ? 2 : 1); ++th_i)
+          for (int tw_i = 0; tw_i < (c ? 2 : 1); ++tw_i) {
+            int kd = a ? td_i * 2 : 1;
+            int kh = b ? th_i * 2 : 1;
+            int kw = c ? tw_i * 2 : 1;
+            taps.push_back(kd * 9 + kh * 3 + kw);
+          }
+      s_idx.push_back(torch::tensor(taps, torch::TensorOptions()
+                                              .dtype(torch::kLong))
+                          .to(wc.device()));
+    }
+    s_dev = wc.device();
+  }
   std::vector<torch::Tensor> blocks;
   std::vector<int64_t> offs(8, 0);
   int64_t cur = 0;
   for (int cls = 0; cls < 8; ++cls) {
-    int a = (cls >> 2) & 1, b = (cls >> 1) & 1, c = cls & 1;
-    std::vector<int64_t> taps;
-    for (int td_i = 0; td_i < (a ? 2 : 1); ++td_i)
-      for (int th_i = 0; th_i < (b ? 2 : 1); ++th_i)
-        for (int tw_i = 0; tw_i < (c ? 2 : 1); ++tw_i) {
-          int kd = a ? td_i * 2 : 1;
-          int kh = b ? th_i * 2 : 1;
-          int kw = c ? tw_i * 2 : 1;
-          taps.push_back(kd * 9 + kh * 3 + kw);
-        }
-    auto idx = torch::tensor(taps, torch::TensorOptions()
-                                       .dtype(torch::kLong)
-                                       .device(wc.device()));
-    auto blk = w3.index_select(2, idx).reshape({Cin, -1});  // [Cin][Co*T]
+    auto blk = w3.index_select(2, s_idx[cls]).reshape({Cin, -1});
     offs[cls] = cur;
     cur += blk.size(1);
     blocks.push_back(blk);
