@@ -1233,8 +1233,15 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
     int COT = stride == 1 ? 32 : 64, CT = stride == 1 ? 32 : 16;
     int co_t = (cd.Cout + COT - 1) / COT, ci_t = (cd.Cin + CT - 1) / CT;
     int64_t nchunks = (int64_t)cd.N * cd.OD * htiles * wtiles;
+    // blocks-in-flight target: swept on MI355X (r2) — 512 won (41.0
+    // ms/step vs 42.1 at the old 768; >=1024 loses to atomic-fold
+    // contention, <=256 underfills the chip). COINN_WGRAD_Z overrides.
+    static const int zbase = []() {
+      const char* e = getenv("COINN_WGRAD_Z");
+      return e ? atoi(e) : 512;
+    }();
     int64_t zstride = std::max<int64_t>(
-        1, std::min<int64_t>(nchunks, 768 / std::max(co_t * ci_t, 1)));
+        1, std::min<int64_t>(nchunks, zbase / std::max(co_t * ci_t, 1)));
     dim3 grid(co_t, ci_t, (unsigned)zstride);
     auto L = [&](auto kern) {
       hipLaunchKernelGGL(kern, grid, dim3(256), 0, current_stream(),
