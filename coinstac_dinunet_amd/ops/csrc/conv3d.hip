@@ -757,7 +757,12 @@ static torch::Tensor conv3d_dgrad_s2(torch::Tensor g, torch::Tensor wc,
 // Blocks grid-stride over spatial chunks; partials fold into dw by fp32
 // atomics once at the end.
 // ---------------------------------------------------------------------------
-template <int OWT, int STRIDE, int CHUNK = 128, bool FUSE_BN = false>
+// SLICED: each z-block stores its partial (co, ci*27) tile into its own
+// slice of dw[zstride][Cout][K] with plain stores; a cheap sum over z
+// replaces the atomic fold (the r2 zstride sweep showed atomics cap the
+// useful parallelism at ~512 blocks).
+template <int OWT, int STRIDE, int CHUNK = 128, bool FUSE_BN = false,
+          bool SLICED = false>
 __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
     const __bf16* __restrict__ x, const __bf16* __restrict__ go,
     float* __restrict__ dw, ConvDims cd, int64_t nchunks, int64_t zstride,
@@ -898,6 +903,7 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
 
   // ---- fold partials into dw[co][ci*27 + tap] -------------------------
   const int K = cd.Cin * 27;
+  float* out = SLICED ? dw + (int64_t)blockIdx.z * cd.Cout * K : dw;
   const int ccol = lane & 15;          // ci col within fragment
   const int crow0 = (lane >> 4) * 4;   // co row
 #pragma unroll 1
@@ -906,8 +912,12 @@ __global__ __launch_bounds__(256) void conv3d_wgrad_s1_kernel(
     for (int r = 0; r < 4; ++r) {
       const int co = co0 + wi * 16 + crow0 + r;
       const int ci = ci0 + wj * 16 + ccol;
-      if (co < cd.Cout && ci < cd.Cin)
-        atomicAdd(&dw[(int64_t)co * K + ci * 27 + tp], acc[tp][r]);
+      if (co < cd.Cout && ci < cd.Cin) {
+        if (SLICED)
+          out[(int64_t)co * K + ci * 27 + tp] = acc[tp][r];
+        else
+          atomicAdd(&out[(int64_t)co * K + ci * 27 + tp], acc[tp][r]);
+      }
     }
   }
 }
@@ -1242,13 +1252,60 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
     }();
     int64_t zstride = std::max<int64_t>(
         1, std::min<int64_t>(nchunks, zbase / std::max(co_t * ci_t, 1)));
+    // sliced partial buffers pay when z-parallelism is high relative to
+    // the (co, ci) tile count (atomic-fold contention regime)
+    const bool sliced = (variant == 0) && zstride >= 64 &&
+                        (int64_t)zstride * cd.Cout * K * 4 <=
+                            (int64_t)512 * 1024 * 1024;
+    torch::Tensor part;
+    float* outp = dw.data_ptr<float>();
+    if (sliced) {
+      part = torch::empty({(int64_t)zstride, (int64_t)cd.Cout, (int64_t)K},
+                          xc.options().dtype(torch::kFloat32));
+      outp = part.data_ptr<float>();
+    }
     dim3 grid(co_t, ci_t, (unsigned)zstride);
     auto L = [&](auto kern) {
       hipLaunchKernelGGL(kern, grid, dim3(256), 0, current_stream(),
                          reinterpret_cast<const __bf16*>(xc.data_ptr()),
                          reinterpret_cast<const __bf16*>(g.data_ptr()),
-                         dw.data_ptr<float>(), cd, nchunks, zstride, abp);
+                         outp, cd, nchunks, zstride, abp);
     };
+    if (sliced) {
+      if (fuse) {
+        if (stride == 1) {
+          if (chunk == 64) L(conv3d_wgrad_s1_kernel<8, 1, 64, true, true>);
+          else if (OWT == 32)
+            L(conv3d_wgrad_s1_kernel<32, 1, 128, true, true>);
+          else if (OWT == 16)
+            L(conv3d_wgrad_s1_kernel<16, 1, 128, true, true>);
+          else L(conv3d_wgrad_s1_kernel<8, 1, 128, true, true>);
+        } else {
+          if (chunk == 64) L(conv3d_wgrad_s1_kernel<8, 2, 64, true, true>);
+          else if (OWT == 32)
+            L(conv3d_wgrad_s1_kernel<32, 2, 128, true, true>);
+          else if (OWT == 16)
+            L(conv3d_wgrad_s1_kernel<16, 2, 128, true, true>);
+          else L(conv3d_wgrad_s1_kernel<8, 2, 128, true, true>);
+        }
+      } else if (stride == 1) {
+        if (chunk == 64) L(conv3d_wgrad_s1_kernel<8, 1, 64, false, true>);
+        else if (OWT == 32)
+          L(conv3d_wgrad_s1_kernel<32, 1, 128, false, true>);
+        else if (OWT == 16)
+          L(conv3d_wgrad_s1_kernel<16, 1, 128, false, true>);
+        else L(conv3d_wgrad_s1_kernel<8, 1, 128, false, true>);
+      } else {
+        if (chunk == 64) L(conv3d_wgrad_s1_kernel<8, 2, 64, false, true>);
+        else if (OWT == 32)
+          L(conv3d_wgrad_s1_kernel<32, 2, 128, false, true>);
+        else if (OWT == 16)
+          L(conv3d_wgrad_s1_kernel<16, 2, 128, false, true>);
+        else L(conv3d_wgrad_s1_kernel<8, 2, 128, false, true>);
+      }
+      auto dwsum = part.sum(0);
+      return dwsum.view({cd.Cout, cd.Cin, 3, 3, 3});
+    }
     if (fuse) {
       if (stride == 1) {
         if (chunk == 64) L(conv3d_wgrad_s1_kernel<8, 1, 64, true>);
