@@ -521,7 +521,7 @@ __global__ __launch_bounds__(256) void conv2d_spatial_kernel(
 // redundancy at the 256/512-channel stages and doubles the MFMA work
 // per barrier (acc[9][2] = 72 VGPRs).
 template <int OWT, int STRIDE, int CHUNK = 128, bool FUSE_BN = false,
-          bool WIDE = false>
+          bool WIDE = false, bool SLICED = false>
 __global__ __launch_bounds__(256) void conv2d_wgrad_sp_kernel(
     const __bf16* __restrict__ x, const __bf16* __restrict__ go,
     float* __restrict__ dw, Conv2dDims cd, int64_t nchunks, int64_t zstride,
@@ -669,6 +669,7 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_sp_kernel(
   }
 
   const int K = cd.Cin * 9;
+  float* out = SLICED ? dw + (int64_t)blockIdx.z * cd.Cout * K : dw;
   const int ccol = lane & 15;
   const int crow0 = (lane >> 4) * 4;
 #pragma unroll 1
@@ -679,8 +680,12 @@ __global__ __launch_bounds__(256) void conv2d_wgrad_sp_kernel(
       for (int r = 0; r < 4; ++r) {
         const int co = co0 + wi * 16 + crow0 + r;
         const int ci = ci0 + (wj + f) * 16 + ccol;
-        if (co < cd.Cout && ci < cd.Cin)
-          atomicAdd(&dw[(int64_t)co * K + ci * 9 + tp], acc[tp][f][r]);
+        if (co < cd.Cout && ci < cd.Cin) {
+          if (SLICED)
+            out[(int64_t)co * K + ci * 9 + tp] = acc[tp][f][r];
+          else
+            atomicAdd(&out[(int64_t)co * K + ci * 9 + tp], acc[tp][f][r]);
+        }
       }
     }
   }
@@ -1048,12 +1053,60 @@ torch::Tensor conv2d_wgrad(torch::Tensor x, torch::Tensor go,
     int64_t zstride = std::max<int64_t>(
         1, std::min<int64_t>(nchunks, 768 / std::max(co_t * ci_t, 1)));
     dim3 grid(co_t, ci_t, (unsigned)zstride);
+    const bool sliced = zstride >= 16 &&
+                        (int64_t)zstride * cd.Cout * K * 4 <=
+                            (int64_t)512 * 1024 * 1024;
+    torch::Tensor part;
+    float* outp = dw.data_ptr<float>();
+    if (sliced) {
+      part = torch::empty({(int64_t)zstride, (int64_t)cd.Cout, (int64_t)K},
+                          xc.options().dtype(torch::kFloat32));
+      outp = part.data_ptr<float>();
+    }
     auto L = [&](auto kern) {
       hipLaunchKernelGGL(kern, grid, dim3(256), 0, current_stream(),
                          reinterpret_cast<const __bf16*>(xc.data_ptr()),
                          reinterpret_cast<const __bf16*>(g.data_ptr()),
-                         dw.data_ptr<float>(), cd, nchunks, zstride, abp);
+                         outp, cd, nchunks, zstride, abp);
     };
+    if (sliced) {
+      if (fuse) {
+        if (stride == 1) {
+          if (chunk == 64)
+            L(conv2d_wgrad_sp_kernel<8, 1, 64, true, false, true>);
+          else if (OWT == 32)
+            L(conv2d_wgrad_sp_kernel<32, 1, 128, true, false, true>);
+          else if (OWT == 16)
+            L(conv2d_wgrad_sp_kernel<16, 1, 128, true, false, true>);
+          else L(conv2d_wgrad_sp_kernel<8, 1, 128, true, false, true>);
+        } else {
+          if (chunk == 64)
+            L(conv2d_wgrad_sp_kernel<8, 2, 64, true, false, true>);
+          else if (OWT == 32)
+            L(conv2d_wgrad_sp_kernel<32, 2, 128, true, false, true>);
+          else if (OWT == 16)
+            L(conv2d_wgrad_sp_kernel<16, 2, 128, true, false, true>);
+          else L(conv2d_wgrad_sp_kernel<8, 2, 128, true, false, true>);
+        }
+      } else if (stride == 1) {
+        if (chunk == 64)
+          L(conv2d_wgrad_sp_kernel<8, 1, 64, false, false, true>);
+        else if (OWT == 32)
+          L(conv2d_wgrad_sp_kernel<32, 1, 128, false, false, true>);
+        else if (OWT == 16)
+          L(conv2d_wgrad_sp_kernel<16, 1, 128, false, false, true>);
+        else L(conv2d_wgrad_sp_kernel<8, 1, 128, false, false, true>);
+      } else {
+        if (chunk == 64)
+          L(conv2d_wgrad_sp_kernel<8, 2, 64, false, false, true>);
+        else if (OWT == 32)
+          L(conv2d_wgrad_sp_kernel<32, 2, 128, false, false, true>);
+        else if (OWT == 16)
+          L(conv2d_wgrad_sp_kernel<16, 2, 128, false, false, true>);
+        else L(conv2d_wgrad_sp_kernel<8, 2, 128, false, false, true>);
+      }
+      return part.sum(0).view({cd.Cout, cd.Cin, 3, 3});
+    }
     if (fuse) {
       // fused-BN wgrad keeps the 32-wide form (no wide instances)
       if (stride == 1) {
