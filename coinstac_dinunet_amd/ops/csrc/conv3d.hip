@@ -1254,7 +1254,7 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
         1, std::min<int64_t>(nchunks, zbase / std::max(co_t * ci_t, 1)));
     // sliced partial buffers pay when z-parallelism is high relative to
     // the (co, ci) tile count (atomic-fold contention regime)
-    const bool sliced = (variant == 0) && zstride >= 64 &&
+    const bool sliced = (variant == 0) && zstride >= 32 &&
                         (int64_t)zstride * cd.Cout * K * 4 <=
                             (int64_t)512 * 1024 * 1024;
     torch::Tensor part;
