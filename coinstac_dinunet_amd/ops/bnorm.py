@@ -51,6 +51,21 @@ class _OpsBNMixin:
         if self.training:
             if self.num_batches_tracked is not None:
                 self.num_batches_tracked.add_(1)
+            if getattr(x, '_coinn_bn_stats', None) is not None:
+                # stats came free from the producing conv's epilogue
+                from .conv import bn_stats_of
+                # pass x itself: .to() would drop the attached stats
+                mean, var, mean_rstd = bn_stats_of(x, self.eps)
+                if self.running_mean is not None:
+                    with torch.no_grad():
+                        n = x.numel() // x.size(1)
+                        unbiased = var * (n / max(n - 1, 1))
+                        self.running_mean.mul_(1 - self.momentum).add_(
+                            mean, alpha=self.momentum)
+                        self.running_var.mul_(1 - self.momentum).add_(
+                            unbiased, alpha=self.momentum)
+                return _BN3dPreFn.apply(x, self.weight, self.bias,
+                                        mean_rstd, self.relu)
             return _BN3dFn.apply(x, self.weight, self.bias,
                                  self.running_mean, self.running_var,
                                  self.momentum, self.eps, self.relu)
@@ -58,6 +73,31 @@ class _OpsBNMixin:
         return C.bn3d_infer(x.to(torch.bfloat16), self.weight, self.bias,
                             self.running_mean, self.running_var, self.eps,
                             self.relu)
+
+
+class _BN3dPreFn(torch.autograd.Function):
+    """BN normalize with externally computed batch statistics (the
+    producing conv's epilogue-stats path): one normalize pass instead of
+    reduce + normalize. Backward is the standard bn3d_bwd."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, mean_rstd, relu):
+        C = require_native()
+        xb = x.to(torch.bfloat16)
+        y = C.bn3d_normalize(xb, mean_rstd, gamma, beta, relu)
+        ctx.save_for_backward(xb, mean_rstd, gamma, beta)
+        ctx.relu = relu
+        ctx.in_dtype = x.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        C = require_native()
+        xb, mean_rstd, gamma, beta = ctx.saved_tensors
+        dx, dgamma, dbeta = C.bn3d_bwd(dy.to(torch.bfloat16), xb, mean_rstd,
+                                       gamma, beta, ctx.relu)
+        return (dx.to(ctx.in_dtype), dgamma.to(gamma.dtype),
+                dbeta.to(beta.dtype), None, None)
 
 
 class _BNAddReluFn(torch.autograd.Function):

@@ -112,8 +112,14 @@ class _ConvBNFn(torch.autograd.Function):
             ab = torch.stack([a, b], 1).contiguous()
         ow = (xb.size(4) + 2 - 3) // stride + 1
         oh = (xb.size(3) + 2 - 3) // stride + 1
+        stats = None
         if ow % 8 == 0 and xb.size(1) >= 16 and oh * ow >= 64:
-            out = C.conv3d_fwd_spatial(xb, wb, stride, 0, ab)
+            if bias is None:
+                # epilogue also emits this output's BN statistics — the
+                # next fused pair skips its bn_reduce pass entirely
+                out, stats = C.conv3d_fwd_spatial_stats(xb, wb, stride, ab)
+            else:
+                out = C.conv3d_fwd_spatial(xb, wb, stride, 0, ab)
         else:
             out = C.conv3d_fwd(xb, wb, stride, ab)
         if bias is not None:
@@ -123,10 +129,13 @@ class _ConvBNFn(torch.autograd.Function):
         ctx.has_bias = bias is not None
         ctx.in_dtype = x_raw.dtype
         ctx.w_dtype = weight.dtype
-        return out
+        if stats is None:
+            stats = torch.empty(0, device=out.device)
+        ctx.mark_non_differentiable(stats)
+        return out, stats
 
     @staticmethod
-    def backward(ctx, grad_out):
+    def backward(ctx, grad_out, _grad_stats):
         C = require_native()
         xb, wb, gamma, beta, mean_rstd, ab = ctx.saved_tensors
         go = grad_out.to(torch.bfloat16).contiguous()
@@ -163,7 +172,7 @@ def conv_bn3d(x_raw, bn, conv):
     xb = x_raw if x_raw.dtype == torch.bfloat16 \
         else x_raw.to(torch.bfloat16)
     if bn.training:
-        mean, var, mean_rstd = C.bn3d_stats(xb.detach(), bn.eps)
+        mean, var, mean_rstd = bn_stats_of(xb, bn.eps)
         if bn.num_batches_tracked is not None:
             bn.num_batches_tracked.add_(1)
         if bn.running_mean is not None:
@@ -178,8 +187,28 @@ def conv_bn3d(x_raw, bn, conv):
         mean = bn.running_mean.float()
         rstd = torch.rsqrt(bn.running_var.float() + bn.eps)
         mean_rstd = torch.stack([mean, rstd], 1).contiguous()
-    return _ConvBNFn.apply(x_raw, bn.weight, bn.bias, mean_rstd,
-                           conv.weight, conv.bias, int(conv.stride[0]))
+    out, stats = _ConvBNFn.apply(x_raw, bn.weight, bn.bias, mean_rstd,
+                                 conv.weight, conv.bias,
+                                 int(conv.stride[0]))
+    if stats.numel() > 0:
+        out._coinn_bn_stats = stats
+    return out
+
+
+def bn_stats_of(xb, eps):
+    """(mean, var, mean_rstd) of a bf16 activation — from the producing
+    conv's epilogue partials when attached, else one reduce pass."""
+    C = require_native()
+    stats = getattr(xb, '_coinn_bn_stats', None)
+    if stats is not None and stats.numel() > 0:
+        with torch.no_grad():
+            sums = stats.sum(0)                      # [C][2]
+            per_ch = xb.numel() // xb.size(1)
+            mean = sums[:, 0] / per_ch
+            var = (sums[:, 1] / per_ch - mean * mean).clamp_min(0)
+            rstd = torch.rsqrt(var + eps)
+            return mean, var, torch.stack([mean, rstd], 1).contiguous()
+    return C.bn3d_stats(xb.detach(), eps)
 
 
 def can_fuse_bn_conv(bn, conv, x):

@@ -51,6 +51,9 @@ torch::Tensor channel_sum(torch::Tensor go);
 torch::Tensor conv3d_fwd_spatial(torch::Tensor x, torch::Tensor w,
                                  int64_t stride, int64_t ctile_opt,
                                  c10::optional<torch::Tensor> bn_ab);
+std::vector<torch::Tensor> conv3d_fwd_spatial_stats(
+    torch::Tensor x, torch::Tensor w, int64_t stride,
+    c10::optional<torch::Tensor> bn_ab);
 torch::Tensor conv3d_dgrad_spatial(torch::Tensor go, torch::Tensor w,
                                    std::vector<int64_t> in_shape);
 torch::Tensor conv3d_dgrad_s2_spatial(torch::Tensor go, torch::Tensor w,
@@ -77,6 +80,9 @@ torch::Tensor conv3d_pw_wgrad(torch::Tensor x, torch::Tensor go);
 std::vector<torch::Tensor> bn3d_fwd(torch::Tensor x, torch::Tensor gamma,
                                     torch::Tensor beta, double eps, bool relu);
 std::vector<torch::Tensor> bn3d_stats(torch::Tensor x, double eps);
+torch::Tensor bn3d_normalize(torch::Tensor x, torch::Tensor mean_rstd,
+                             torch::Tensor gamma, torch::Tensor beta,
+                             bool relu);
 std::vector<torch::Tensor> bn3d_fwd_res(torch::Tensor x, torch::Tensor res,
                                         torch::Tensor gamma,
                                         torch::Tensor beta, double eps);
@@ -120,6 +126,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3d_fwd_spatial", &conv3d_fwd_spatial,
         py::arg("x"), py::arg("w"), py::arg("stride"),
         py::arg("ctile_opt") = 0, py::arg("bn_ab") = py::none());
+  m.def("conv3d_fwd_spatial_stats", &conv3d_fwd_spatial_stats,
+        py::arg("x"), py::arg("w"), py::arg("stride"),
+        py::arg("bn_ab") = py::none());
   m.def("conv3d_dgrad_spatial", &conv3d_dgrad_spatial);
   m.def("conv3d_dgrad_s2_spatial", &conv3d_dgrad_s2_spatial);
   m.def("conv2d_fwd", &conv2d_fwd, py::arg("x"), py::arg("w"),
@@ -135,6 +144,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3d_pw_wgrad", &conv3d_pw_wgrad);
   m.def("bn3d_fwd", &bn3d_fwd);
   m.def("bn3d_stats", &bn3d_stats);
+  m.def("bn3d_normalize", &bn3d_normalize);
   m.def("bn3d_fwd_res", &bn3d_fwd_res);
   m.def("bn3d_bwd_res", &bn3d_bwd_res);
   m.def("bn3d_infer", &bn3d_infer);

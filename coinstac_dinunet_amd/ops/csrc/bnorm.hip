@@ -257,6 +257,34 @@ std::vector<torch::Tensor> bn3d_fwd(torch::Tensor x, torch::Tensor gamma,
   return {y, mean, var, mean_rstd};
 }
 
+// normalize-only (training form with externally computed batch stats —
+// the conv-epilogue-stats path): y = [relu](affine(x)).
+torch::Tensor bn3d_normalize(torch::Tensor x, torch::Tensor mean_rstd,
+                             torch::Tensor gamma, torch::Tensor beta,
+                             bool relu) {
+  CHECK_GPU(x);
+  auto xc = x.contiguous();
+  int N = (int)xc.size(0), C = (int)xc.size(1);
+  int64_t spatial = xc.numel() / ((int64_t)N * C);
+  auto g = gamma.to(torch::kFloat32).contiguous();
+  auto b = beta.to(torch::kFloat32).contiguous();
+  auto mr = mean_rstd.to(torch::kFloat32).contiguous();
+  auto y = torch::empty_like(xc);
+  dim3 grid, blk;
+  bn_grid(spatial, N * C, grid, blk);
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, blk, 0, current_stream(),
+                       reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                       reinterpret_cast<__bf16*>(y.data_ptr()),
+                       mr.data_ptr<float>(), g.data_ptr<float>(),
+                       b.data_ptr<float>(), C, spatial,
+                       (const __bf16*)nullptr);
+  };
+  if (relu) launch(bn_normalize_kernel<true>);
+  else launch(bn_normalize_kernel<false>);
+  return y;
+}
+
 // fused residual form: y = relu(affine(x) + res); returns stats too.
 std::vector<torch::Tensor> bn3d_fwd_res(torch::Tensor x, torch::Tensor res,
                                         torch::Tensor gamma,

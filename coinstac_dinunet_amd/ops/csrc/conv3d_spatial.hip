@@ -38,12 +38,18 @@ struct SpDims {
 // staging (a = gamma*rstd, b = beta - mean*a, exactly bn_normalize's
 // folding) so the normalized tensor never exists in HBM. Padding halo
 // stays 0 (conv pads the BN OUTPUT with zeros).
+// STATS: the epilogue also folds per-output-channel (sum, sumsq) of the
+// written elements into stats[hash_slice][NCOL][2] — the NEXT block's BN
+// statistics come for free (no standalone bn_reduce pass over the
+// activation). Hash slices (blockIdx.x & (SLICES-1)) spread the atomics.
 template <int OWT, int STRIDE, int CTILE,
-          int CHUNK = (STRIDE == 1 ? 256 : 128), bool FUSE_BN = false>
+          int CHUNK = (STRIDE == 1 ? 256 : 128), bool FUSE_BN = false,
+          bool STATS = false>
 __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
     const __bf16* __restrict__ in, const __bf16* __restrict__ wb,
     __bf16* __restrict__ out, SpDims sd, int64_t nchunks,
-    const float* __restrict__ bn_ab = nullptr) {
+    const float* __restrict__ bn_ab = nullptr,
+    float* __restrict__ stats = nullptr) {
   constexpr int OHT = CHUNK / OWT;
   constexpr int IW = STRIDE * OWT;                  // staged interior width
   constexpr int W2 = IW + (STRIDE == 1 ? 4 : 2);
@@ -178,6 +184,7 @@ __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
   const int64_t out_n = (int64_t)n * sd.NCOL * sd.TD * THW;
   const int ccol = lane & 15;
   const int crow0 = (lane >> 4) * 4;
+  float cs[2] = {0.f, 0.f}, css[2] = {0.f, 0.f};  // per-col-frag partials
 #pragma unroll
   for (int i = 0; i < MPW; ++i) {
 #pragma unroll
@@ -189,9 +196,53 @@ __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
         const int m = (wave * MPW + i) * 16 + crow0 + r;
         const int oh = oh0 + m / OWT;
         const int ow = ow0 + m % OWT;
-        if (oh < sd.TH && ow < sd.TW)
+        if (oh < sd.TH && ow < sd.TW) {
+          const float v = acc[i][j][r];
           out[out_n + ((int64_t)col * sd.TD + td) * THW +
-              (int64_t)oh * sd.TW + ow] = (__bf16)(acc[i][j][r]);
+              (int64_t)oh * sd.TW + ow] = (__bf16)v;
+          if (STATS) {
+            // accumulate the ROUNDED value: bit-matches bn3d_stats on
+            // the stored bf16 tensor
+            const float vb = (float)(__bf16)v;
+            cs[j] += vb;
+            css[j] += vb * vb;
+          }
+        }
+      }
+    }
+  }
+  if (STATS) {
+    // lanes l, l+16, l+32, l+48 share ccol: fold over lane bits 4-5
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      cs[j] += __shfl_xor(cs[j], 16);
+      cs[j] += __shfl_xor(cs[j], 32);
+      css[j] += __shfl_xor(css[j], 16);
+      css[j] += __shfl_xor(css[j], 32);
+    }
+    __shared__ float sred[4][2][16][2];
+    if ((lane >> 4) == 0) {
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        sred[wave][j][ccol][0] = cs[j];
+        sred[wave][j][ccol][1] = css[j];
+      }
+    }
+    __syncthreads();
+    // 64 (col, stat) pairs: threads 0..63 fold the 4 waves and emit one
+    // atomicAdd each into the hashed stats slice
+    if (tid < 64) {
+      const int j = tid >> 5;           // col fragment
+      const int cc = (tid >> 1) & 15;   // ccol
+      const int st = tid & 1;           // 0 = sum, 1 = sumsq
+      const int col = ncol0 + j * 16 + cc;
+      if (col < sd.NCOL) {
+        float t = 0.f;
+#pragma unroll
+        for (int w = 0; w < 4; ++w) t += sred[w][j][cc][st];
+        float* slice = stats +
+            ((int64_t)(blockIdx.x & 63) * sd.NCOL + col) * 2 + st;
+        atomicAdd(slice, t);
       }
     }
   }
@@ -378,7 +429,8 @@ static torch::Tensor prep_wb(torch::Tensor w_flat2d, int KCH, int ctile) {
 
 static void launch_spatial(torch::Tensor in, torch::Tensor wb,
                            torch::Tensor out, SpDims sd, int stride,
-                           int ctile = 0, const float* bn_ab = nullptr) {
+                           int ctile = 0, const float* bn_ab = nullptr,
+                           float* stats = nullptr) {
   int OWT = sd.TW % 32 == 0 ? 32 : (sd.TW % 16 == 0 ? 16 : 8);
   int chunk = stride == 1 ? 256 : 128;
   // small images (d8-class): 64-position chunks keep the grid dense
@@ -402,12 +454,31 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
   };
   auto L = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, dim3(256), 0, s, ip, wp, op, sd, nchunks,
-                       (const float*)nullptr);
+                       (const float*)nullptr, (float*)nullptr);
   };
   auto LF = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, dim3(256), 0, s, ip, wp, op, sd, nchunks,
-                       bn_ab);
+                       bn_ab, stats);
   };
+  if (bn_ab != nullptr && stats != nullptr) {
+    // fused normalize-on-load + epilogue-stats instances
+    if (stride == 1) {
+      if (chunk == 64) LF(conv3d_spatial_kernel<8, 1, 32, 64, true, true>);
+      else if (OWT == 32)
+        LF(conv3d_spatial_kernel<32, 1, 32, 256, true, true>);
+      else if (OWT == 16)
+        LF(conv3d_spatial_kernel<16, 1, 32, 256, true, true>);
+      else LF(conv3d_spatial_kernel<8, 1, 32, 256, true, true>);
+    } else {
+      if (chunk == 64) LF(conv3d_spatial_kernel<8, 2, 16, 64, true, true>);
+      else if (OWT == 32)
+        LF(conv3d_spatial_kernel<32, 2, 16, 128, true, true>);
+      else if (OWT == 16)
+        LF(conv3d_spatial_kernel<16, 2, 16, 128, true, true>);
+      else LF(conv3d_spatial_kernel<8, 2, 16, 128, true, true>);
+    }
+    return;
+  }
   if (bn_ab != nullptr) {
     // fused normalize-on-load instances (default tilings only)
     if (stride == 1) {
@@ -486,6 +557,40 @@ torch::Tensor conv3d_fwd_spatial(torch::Tensor x, torch::Tensor w,
   launch_spatial(xc, wb, out, sd, (int)stride, ctile,
                  fuse ? ab.data_ptr<float>() : nullptr);
   return out;
+}
+
+// fused fwd + epilogue BN statistics: returns {out, stats[64][Cout][2]}
+// (hash-sliced partial sums; callers fold with stats.sum(0)).
+std::vector<torch::Tensor> conv3d_fwd_spatial_stats(
+    torch::Tensor x, torch::Tensor w, int64_t stride,
+    c10::optional<torch::Tensor> bn_ab_opt) {
+  torch::Tensor bn_ab = bn_ab_opt.value_or(torch::Tensor());
+  CHECK_GPU(x);
+  auto xc = x.contiguous();
+  auto wc = w.to(torch::kBFloat16).contiguous();
+  TORCH_CHECK(xc.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(stride == 1 || stride == 2);
+  TORCH_CHECK(bn_ab.defined() && bn_ab.numel() > 0,
+              "stats form requires the fused-BN path");
+  auto ab = bn_ab.to(torch::kFloat32).contiguous();
+  SpDims sd;
+  sd.N = (int)xc.size(0); sd.KCH = (int)xc.size(1);
+  sd.D = (int)xc.size(2); sd.H = (int)xc.size(3); sd.W = (int)xc.size(4);
+  sd.NCOL = (int)wc.size(0);
+  sd.TD = (sd.D + 2 - 3) / (int)stride + 1;
+  sd.TH = (sd.H + 2 - 3) / (int)stride + 1;
+  sd.TW = (sd.W + 2 - 3) / (int)stride + 1;
+  int ctile = stride == 1 ? 32 : 16;
+  auto wb = prep_wb(wc.reshape({sd.NCOL, (int64_t)sd.KCH * 27}), sd.KCH,
+                    ctile);
+  sd.Kpad = (int)wb.size(1);
+  auto out = torch::empty({sd.N, sd.NCOL, sd.TD, sd.TH, sd.TW},
+                          xc.options());
+  auto stats = torch::zeros({64, sd.NCOL, 2},
+                            xc.options().dtype(torch::kFloat32));
+  launch_spatial(xc, wb, out, sd, (int)stride, ctile, ab.data_ptr<float>(),
+                 stats.data_ptr<float>());
+  return {out, stats};
 }
 
 torch::Tensor conv3d_dgrad_spatial(torch::Tensor go, torch::Tensor w,
