@@ -12,7 +12,7 @@ remain on the library path (documented scope).
 import torch.nn as nn
 
 from ..ops.bnorm import OpsBatchNorm2d, bn_add_relu
-from ..ops.conv import OpsConv2d
+from ..ops.conv import OpsConv2d, can_fuse_bn_conv2d, conv_bn2d
 
 
 class BasicBlock(nn.Module):
@@ -33,9 +33,14 @@ class BasicBlock(nn.Module):
 
     def forward(self, x):
         idt = x if self.down is None else self.down(x)
-        y = self.bn1(self.conv1(x))
+        y = self.conv1(x)
+        if self.bn1.training and can_fuse_bn_conv2d(self.bn1, self.conv2, y):
+            # bn1(+ReLU) folds into conv2's input load
+            y = conv_bn2d(y, self.bn1, self.conv2)
+        else:
+            y = self.conv2(self.bn1(y))
         # bn2 -> +identity -> ReLU fused into one pass each way
-        return bn_add_relu(self.conv2(y), idt, self.bn2)
+        return bn_add_relu(y, idt, self.bn2)
 
 
 class ResNet18(nn.Module):

@@ -211,6 +211,77 @@ def bn_stats_of(xb, eps):
     return C.bn3d_stats(xb.detach(), eps)
 
 
+class _ConvBN2dFn(torch.autograd.Function):
+    """2D twin of _ConvBNFn: the previous BN(+ReLU) folds into this 3x3
+    conv's input load (ResNet BasicBlock's bn1 -> conv2 pair)."""
+
+    @staticmethod
+    def forward(ctx, x_raw, gamma, beta, mean_rstd, weight, stride):
+        C = require_native()
+        xb = x_raw.to(torch.bfloat16)
+        wb = weight.to(torch.bfloat16)
+        with torch.no_grad():
+            a = gamma.float() * mean_rstd[:, 1]
+            b = beta.float() - mean_rstd[:, 0] * a
+            ab = torch.stack([a, b], 1).contiguous()
+        out = C.conv2d_fwd(xb, wb, stride, ab)
+        ctx.save_for_backward(xb, wb, gamma, beta, mean_rstd, ab)
+        ctx.stride = stride
+        ctx.in_dtype = x_raw.dtype
+        ctx.w_dtype = weight.dtype
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        C = require_native()
+        xb, wb, gamma, beta, mean_rstd, ab = ctx.saved_tensors
+        go = grad_out.to(torch.bfloat16).contiguous()
+        dz = C.conv2d_dgrad(go, wb, list(xb.shape), ctx.stride)
+        gw = C.conv2d_wgrad(xb, go, ctx.stride, ab).to(ctx.w_dtype) \
+            if ctx.needs_input_grad[4] else None
+        dx, dgamma, dbeta = C.bn3d_bwd(dz, xb, mean_rstd, gamma, beta, True)
+        return (dx.to(ctx.in_dtype), dgamma.to(gamma.dtype),
+                dbeta.to(beta.dtype), None, gw, None)
+
+
+def conv_bn2d(x_raw, bn, conv):
+    """Fused BN(+ReLU) -> 3x3 Conv2d (the 2D twin of conv_bn3d)."""
+    C = require_native()
+    xb = x_raw if x_raw.dtype == torch.bfloat16 \
+        else x_raw.to(torch.bfloat16)
+    if bn.training:
+        mean, var, mean_rstd = bn_stats_of(xb, bn.eps)
+        if bn.num_batches_tracked is not None:
+            bn.num_batches_tracked.add_(1)
+        if bn.running_mean is not None:
+            with torch.no_grad():
+                n = xb.numel() // xb.size(1)
+                unbiased = var * (n / max(n - 1, 1))
+                bn.running_mean.mul_(1 - bn.momentum).add_(
+                    mean, alpha=bn.momentum)
+                bn.running_var.mul_(1 - bn.momentum).add_(
+                    unbiased, alpha=bn.momentum)
+    else:
+        mean = bn.running_mean.float()
+        rstd = torch.rsqrt(bn.running_var.float() + bn.eps)
+        mean_rstd = torch.stack([mean, rstd], 1).contiguous()
+    return _ConvBN2dFn.apply(x_raw, bn.weight, bn.bias, mean_rstd,
+                             conv.weight, int(conv.stride[0]))
+
+
+def can_fuse_bn_conv2d(bn, conv, x):
+    """True when the 2D (bn -> conv) pair routes onto the fused kernels."""
+    from .bnorm import OpsBatchNorm2d
+    return (x.is_cuda and native_available()
+            and isinstance(bn, OpsBatchNorm2d) and bn.relu
+            and isinstance(conv, OpsConv2d) and conv.bias is None
+            and conv.kernel_size == (3, 3)
+            and conv.padding == (1, 1)
+            and conv.stride[0] in (1, 2)
+            and conv.stride[0] == conv.stride[1]
+            and conv.dilation == (1, 1) and conv.groups == 1)
+
+
 def can_fuse_bn_conv(bn, conv, x):
     """True when the (bn -> conv) pair routes onto the fused kernels."""
     from .bnorm import OpsBatchNorm3d

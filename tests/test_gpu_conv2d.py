@@ -134,3 +134,40 @@ def test_resnet18_trains_on_hip_kernels(dev):
     assert losses[-1] < losses[0] * 0.8, losses[::5]
     for n, p in net.named_parameters():
         assert p.grad is not None, n
+
+
+def test_conv_bn2d_fused_matches_composed(dev):
+    """2D BN(+ReLU)->conv normalize-on-load vs the composed ops."""
+    from coinstac_dinunet_amd.ops.bnorm import OpsBatchNorm2d
+    from coinstac_dinunet_amd.ops.conv import OpsConv2d, conv_bn2d
+    torch.manual_seed(93)
+    for Ci, Co, H, W, s in [(32, 32, 28, 28, 1), (64, 128, 28, 28, 2),
+                            (64, 64, 14, 14, 1)]:
+        bn = OpsBatchNorm2d(Ci, relu=True).to(dev)
+        with torch.no_grad():
+            bn.weight.mul_(0).add_(torch.rand(Ci, device=dev) + 0.5)
+            bn.bias.add_(torch.randn(Ci, device=dev) * 0.1)
+        conv = OpsConv2d(Ci, Co, 3, stride=s, padding=1, bias=False).to(dev)
+        bn.train(), conv.train()
+        x0 = torch.randn(2, Ci, H, W, device=dev, dtype=torch.bfloat16)
+        go = None
+
+        def run(fused):
+            nonlocal go
+            bn.running_mean.zero_(), bn.running_var.fill_(1.0)
+            x = x0.clone().requires_grad_(True)
+            for p in list(bn.parameters()) + list(conv.parameters()):
+                p.grad = None
+            out = conv_bn2d(x, bn, conv) if fused else conv(bn(x))
+            if go is None:
+                go = torch.randn_like(out) * 0.1
+            out.backward(go)
+            return (out.float(), x.grad.float(), conv.weight.grad.clone())
+
+        yf, gxf, gwf = run(True)
+        yu, gxu, gwu = run(False)
+        tol = 5e-2 * (Ci * 9) ** 0.5 * 0.3
+        torch.testing.assert_close(yf, yu, rtol=3e-2, atol=tol)
+        torch.testing.assert_close(gxf, gxu, rtol=5e-2, atol=5e-2)
+        m = (2 * H * W) ** 0.5
+        torch.testing.assert_close(gwf, gwu, rtol=5e-2, atol=5e-2 * m * 0.1)
