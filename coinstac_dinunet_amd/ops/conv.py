@@ -141,12 +141,19 @@ class _ConvBNFn(torch.autograd.Function):
         go = grad_out.to(torch.bfloat16).contiguous()
         stride = ctx.stride
         # dz = conv dgrad wrt the (virtual) normalized input — same
-        # routing as the unfused path
+        # routing as the unfused path. The stride-1 spatial dgrad ALSO
+        # folds the BN-backward reduction into its epilogue, so
+        # bn3d_bwd's standalone reduce pass disappears.
         wsub = (xb.size(4) + 1) // 2
         hsub = (xb.size(3) + 1) // 2
+        sums = None
         if (stride == 1 and xb.size(4) % 8 == 0 and go.size(1) >= 16
                 and xb.size(3) * xb.size(4) >= 64):
-            dz = C.conv3d_dgrad_spatial(go, wb, list(xb.shape))
+            prm = torch.stack([mean_rstd[:, 0], mean_rstd[:, 1],
+                               gamma.float(), beta.float()], 1).contiguous()
+            dz, bsums = C.conv3d_dgrad_spatial_bnbwd(
+                go, wb, list(xb.shape), xb, prm)
+            sums = bsums.sum(0)
         elif (stride == 2 and wsub % 8 == 0 and go.size(1) >= 32
                 and hsub * wsub >= 128):
             dz = C.conv3d_dgrad_s2_spatial(go, wb, list(xb.shape))
@@ -156,7 +163,12 @@ class _ConvBNFn(torch.autograd.Function):
             if ctx.needs_input_grad[4] else None
         gb = C.channel_sum(go) if (ctx.has_bias
                                    and ctx.needs_input_grad[5]) else None
-        dx, dgamma, dbeta = C.bn3d_bwd(dz, xb, mean_rstd, gamma, beta, True)
+        if sums is not None:
+            dx, dgamma, dbeta = C.bn3d_bwd_pre(dz, xb, mean_rstd, gamma,
+                                               beta, True, sums)
+        else:
+            dx, dgamma, dbeta = C.bn3d_bwd(dz, xb, mean_rstd, gamma, beta,
+                                           True)
         return (dx.to(ctx.in_dtype), dgamma.to(gamma.dtype),
                 dbeta.to(beta.dtype), None, gw, gb, None)
 

@@ -56,6 +56,9 @@ std::vector<torch::Tensor> conv3d_fwd_spatial_stats(
     c10::optional<torch::Tensor> bn_ab);
 torch::Tensor conv3d_dgrad_spatial(torch::Tensor go, torch::Tensor w,
                                    std::vector<int64_t> in_shape);
+std::vector<torch::Tensor> conv3d_dgrad_spatial_bnbwd(
+    torch::Tensor go, torch::Tensor w, std::vector<int64_t> in_shape,
+    torch::Tensor xraw, torch::Tensor bn_prm);
 torch::Tensor conv3d_dgrad_s2_spatial(torch::Tensor go, torch::Tensor w,
                                       std::vector<int64_t> in_shape);
 // conv2d.hip
@@ -80,6 +83,11 @@ torch::Tensor conv3d_pw_wgrad(torch::Tensor x, torch::Tensor go);
 std::vector<torch::Tensor> bn3d_fwd(torch::Tensor x, torch::Tensor gamma,
                                     torch::Tensor beta, double eps, bool relu);
 std::vector<torch::Tensor> bn3d_stats(torch::Tensor x, double eps);
+std::vector<torch::Tensor> bn3d_bwd_pre(torch::Tensor dy, torch::Tensor x,
+                                        torch::Tensor mean_rstd,
+                                        torch::Tensor gamma,
+                                        torch::Tensor beta, bool relu,
+                                        torch::Tensor sums2);
 torch::Tensor bn3d_normalize(torch::Tensor x, torch::Tensor mean_rstd,
                              torch::Tensor gamma, torch::Tensor beta,
                              bool relu);
@@ -130,6 +138,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("w"), py::arg("stride"),
         py::arg("bn_ab") = py::none());
   m.def("conv3d_dgrad_spatial", &conv3d_dgrad_spatial);
+  m.def("conv3d_dgrad_spatial_bnbwd", &conv3d_dgrad_spatial_bnbwd);
   m.def("conv3d_dgrad_s2_spatial", &conv3d_dgrad_s2_spatial);
   m.def("conv2d_fwd", &conv2d_fwd, py::arg("x"), py::arg("w"),
         py::arg("stride"), py::arg("bn_ab") = py::none());
@@ -145,6 +154,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn3d_fwd", &bn3d_fwd);
   m.def("bn3d_stats", &bn3d_stats);
   m.def("bn3d_normalize", &bn3d_normalize);
+  m.def("bn3d_bwd_pre", &bn3d_bwd_pre);
   m.def("bn3d_fwd_res", &bn3d_fwd_res);
   m.def("bn3d_bwd_res", &bn3d_bwd_res);
   m.def("bn3d_infer", &bn3d_infer);

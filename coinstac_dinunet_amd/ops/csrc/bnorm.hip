@@ -415,6 +415,42 @@ std::vector<torch::Tensor> bn3d_bwd(torch::Tensor dy, torch::Tensor x,
   return {dx, dgamma, dbeta};
 }
 
+// backward with the reduction precomputed (conv-epilogue bnbwd path):
+// runs ONLY the dx pass; sums2 = [C][2] (sum dz*mask, sum dz*xhat*mask).
+std::vector<torch::Tensor> bn3d_bwd_pre(torch::Tensor dy, torch::Tensor x,
+                                        torch::Tensor mean_rstd,
+                                        torch::Tensor gamma,
+                                        torch::Tensor beta, bool relu,
+                                        torch::Tensor sums2) {
+  CHECK_GPU(dy);
+  auto dyc = dy.contiguous();
+  auto xc = x.contiguous();
+  int N = (int)dyc.size(0), C = (int)dyc.size(1);
+  int64_t spatial = dyc.numel() / ((int64_t)N * C);
+  auto g = gamma.to(torch::kFloat32).contiguous();
+  auto b = beta.to(torch::kFloat32).contiguous();
+  auto sums = sums2.to(torch::kFloat32).contiguous();
+  TORCH_CHECK(sums.numel() == 2 * C, "sums2 must be [C,2]");
+  dim3 grid, blk;
+  bn_grid(spatial, N * C, grid, blk);
+  auto dx = torch::empty_like(dyc);
+  int64_t per_ch = (int64_t)N * spatial;
+  auto launch_d = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, blk, 0, current_stream(),
+                       reinterpret_cast<const __bf16*>(dyc.data_ptr()),
+                       reinterpret_cast<const __bf16*>(xc.data_ptr()),
+                       reinterpret_cast<__bf16*>(dx.data_ptr()),
+                       mean_rstd.data_ptr<float>(), sums.data_ptr<float>(),
+                       g.data_ptr<float>(), b.data_ptr<float>(), per_ch, C,
+                       spatial, (const __bf16*)nullptr, (__bf16*)nullptr);
+  };
+  if (relu) launch_d(bn_bwd_dx_kernel<true>);
+  else launch_d(bn_bwd_dx_kernel<false>);
+  auto dbeta = sums.select(1, 0).clone();
+  auto dgamma = sums.select(1, 1).clone();
+  return {dx, dgamma, dbeta};
+}
+
 // backward of the fused residual form; extra output dres = masked dy.
 std::vector<torch::Tensor> bn3d_bwd_res(torch::Tensor dy, torch::Tensor x,
                                         torch::Tensor res,

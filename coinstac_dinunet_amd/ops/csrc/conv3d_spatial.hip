@@ -38,18 +38,25 @@ struct SpDims {
 // staging (a = gamma*rstd, b = beta - mean*a, exactly bn_normalize's
 // folding) so the normalized tensor never exists in HBM. Padding halo
 // stays 0 (conv pads the BN OUTPUT with zeros).
-// STATS: the epilogue also folds per-output-channel (sum, sumsq) of the
-// written elements into stats[hash_slice][NCOL][2] — the NEXT block's BN
-// statistics come for free (no standalone bn_reduce pass over the
-// activation). Hash slices (blockIdx.x & (SLICES-1)) spread the atomics.
+// SMODE 1: the epilogue also folds per-output-channel (sum, sumsq) of
+// the written elements into stats[hash_slice][NCOL][2] — the NEXT
+// block's BN statistics come for free (no standalone bn_reduce pass).
+// SMODE 2 (dgrad use): the written elements are dz of a BN(+ReLU)
+// output; fold the BN-BACKWARD reduction (sum(dz*mask), sum(dz*xhat*
+// mask)) instead, reading x_raw at the same offsets with the mask
+// recomputed from bn_prm = [mean, rstd, gamma, beta] per channel —
+// replaces the standalone bn_bwd_reduce pass.
+// Hash slices (blockIdx.x & 63) spread the atomics.
 template <int OWT, int STRIDE, int CTILE,
           int CHUNK = (STRIDE == 1 ? 256 : 128), bool FUSE_BN = false,
-          bool STATS = false>
+          int SMODE = 0>
 __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
     const __bf16* __restrict__ in, const __bf16* __restrict__ wb,
     __bf16* __restrict__ out, SpDims sd, int64_t nchunks,
     const float* __restrict__ bn_ab = nullptr,
-    float* __restrict__ stats = nullptr) {
+    float* __restrict__ stats = nullptr,
+    const __bf16* __restrict__ xraw = nullptr,
+    const float* __restrict__ bn_prm = nullptr) {
   constexpr int OHT = CHUNK / OWT;
   constexpr int IW = STRIDE * OWT;                  // staged interior width
   constexpr int W2 = IW + (STRIDE == 1 ? 4 : 2);
@@ -191,6 +198,13 @@ __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
     for (int j = 0; j < 2; ++j) {
       const int col = ncol0 + j * 16 + ccol;
       if (col >= sd.NCOL) continue;
+      float p_mean = 0.f, p_rstd = 0.f, p_g = 0.f, p_b = 0.f;
+      if (SMODE == 2) {
+        p_mean = bn_prm[col * 4 + 0];
+        p_rstd = bn_prm[col * 4 + 1];
+        p_g = bn_prm[col * 4 + 2];
+        p_b = bn_prm[col * 4 + 3];
+      }
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int m = (wave * MPW + i) * 16 + crow0 + r;
@@ -198,20 +212,28 @@ __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
         const int ow = ow0 + m % OWT;
         if (oh < sd.TH && ow < sd.TW) {
           const float v = acc[i][j][r];
-          out[out_n + ((int64_t)col * sd.TD + td) * THW +
-              (int64_t)oh * sd.TW + ow] = (__bf16)v;
-          if (STATS) {
+          const int64_t off = out_n + ((int64_t)col * sd.TD + td) * THW +
+                              (int64_t)oh * sd.TW + ow;
+          out[off] = (__bf16)v;
+          if (SMODE == 1) {
             // accumulate the ROUNDED value: bit-matches bn3d_stats on
             // the stored bf16 tensor
             const float vb = (float)(__bf16)v;
             cs[j] += vb;
             css[j] += vb * vb;
+          } else if (SMODE == 2) {
+            const float xh = ((float)xraw[off] - p_mean) * p_rstd;
+            if (p_g * xh + p_b > 0.f) {  // ReLU mask
+              const float d = (float)(__bf16)v;
+              cs[j] += d;
+              css[j] += d * xh;
+            }
           }
         }
       }
     }
   }
-  if (STATS) {
+  if (SMODE != 0) {
     // lanes l, l+16, l+32, l+48 share ccol: fold over lane bits 4-5
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
@@ -454,28 +476,30 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
   };
   auto L = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, dim3(256), 0, s, ip, wp, op, sd, nchunks,
-                       (const float*)nullptr, (float*)nullptr);
+                       (const float*)nullptr, (float*)nullptr,
+                       (const __bf16*)nullptr, (const float*)nullptr);
   };
   auto LF = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, dim3(256), 0, s, ip, wp, op, sd, nchunks,
-                       bn_ab, stats);
+                       bn_ab, stats, (const __bf16*)nullptr,
+                       (const float*)nullptr);
   };
   if (bn_ab != nullptr && stats != nullptr) {
     // fused normalize-on-load + epilogue-stats instances
     if (stride == 1) {
-      if (chunk == 64) LF(conv3d_spatial_kernel<8, 1, 32, 64, true, true>);
+      if (chunk == 64) LF(conv3d_spatial_kernel<8, 1, 32, 64, true, 1>);
       else if (OWT == 32)
-        LF(conv3d_spatial_kernel<32, 1, 32, 256, true, true>);
+        LF(conv3d_spatial_kernel<32, 1, 32, 256, true, 1>);
       else if (OWT == 16)
-        LF(conv3d_spatial_kernel<16, 1, 32, 256, true, true>);
-      else LF(conv3d_spatial_kernel<8, 1, 32, 256, true, true>);
+        LF(conv3d_spatial_kernel<16, 1, 32, 256, true, 1>);
+      else LF(conv3d_spatial_kernel<8, 1, 32, 256, true, 1>);
     } else {
-      if (chunk == 64) LF(conv3d_spatial_kernel<8, 2, 16, 64, true, true>);
+      if (chunk == 64) LF(conv3d_spatial_kernel<8, 2, 16, 64, true, 1>);
       else if (OWT == 32)
-        LF(conv3d_spatial_kernel<32, 2, 16, 128, true, true>);
+        LF(conv3d_spatial_kernel<32, 2, 16, 128, true, 1>);
       else if (OWT == 16)
-        LF(conv3d_spatial_kernel<16, 2, 16, 128, true, true>);
-      else LF(conv3d_spatial_kernel<8, 2, 16, 128, true, true>);
+        LF(conv3d_spatial_kernel<16, 2, 16, 128, true, 1>);
+      else LF(conv3d_spatial_kernel<8, 2, 16, 128, true, 1>);
     }
     return;
   }
@@ -612,6 +636,60 @@ torch::Tensor conv3d_dgrad_spatial(torch::Tensor go, torch::Tensor w,
   auto dx = torch::empty(in_shape, g.options());
   launch_spatial(g, wb, dx, sd, 1);
   return dx;
+}
+
+// stride-1 dgrad that ALSO folds the BN-backward reduction of the dz it
+// writes: returns {dz, bsums[64][Cin][2]} with bsums = hash-sliced
+// partials of (sum dz*mask, sum dz*xhat*mask). bn_prm packs
+// [mean, rstd, gamma, beta] per channel of the BN being backpropped.
+std::vector<torch::Tensor> conv3d_dgrad_spatial_bnbwd(
+    torch::Tensor go, torch::Tensor w, std::vector<int64_t> in_shape,
+    torch::Tensor xraw, torch::Tensor bn_prm) {
+  CHECK_GPU(go);
+  auto g = go.to(torch::kBFloat16).contiguous();
+  auto wc = w.to(torch::kBFloat16).contiguous();
+  auto xr = xraw.contiguous();
+  TORCH_CHECK(xr.scalar_type() == torch::kBFloat16, "xraw must be bf16");
+  auto prm = bn_prm.to(torch::kFloat32).contiguous();
+  SpDims sd;
+  sd.N = (int)g.size(0); sd.KCH = (int)g.size(1);
+  sd.D = (int)g.size(2); sd.H = (int)g.size(3); sd.W = (int)g.size(4);
+  sd.NCOL = (int)in_shape[1];
+  sd.TD = (int)in_shape[2]; sd.TH = (int)in_shape[3];
+  sd.TW = (int)in_shape[4];
+  TORCH_CHECK(prm.numel() == 4 * sd.NCOL, "bn_prm must be [Cin,4]");
+  int Cout = (int)wc.size(0), Cin = (int)wc.size(1);
+  auto wf = wc.reshape({Cout, Cin, 27}).flip(-1).permute({1, 0, 2})
+                .reshape({Cin, (int64_t)Cout * 27}).contiguous();
+  auto wb = prep_wb(wf, Cout, 32);
+  sd.Kpad = (int)wb.size(1);
+  auto dx = torch::empty(in_shape, g.options());
+  auto bsums = torch::zeros({64, sd.NCOL, 2},
+                            g.options().dtype(torch::kFloat32));
+
+  int OWT = sd.TW % 32 == 0 ? 32 : (sd.TW % 16 == 0 ? 16 : 8);
+  int chunk = 256;
+  if (sd.TH * sd.TW < chunk) { chunk = 64; OWT = 8; }
+  int OHT = chunk / OWT;
+  int wtiles = (sd.TW + OWT - 1) / OWT;
+  int htiles = (sd.TH + OHT - 1) / OHT;
+  int64_t nchunks = (int64_t)sd.N * sd.TD * htiles * wtiles;
+  dim3 grid((unsigned)nchunks, (sd.NCOL + 31) / 32);
+  auto st = current_stream();
+  auto LB = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, st,
+                       reinterpret_cast<const __bf16*>(g.data_ptr()),
+                       reinterpret_cast<const __bf16*>(wb.data_ptr()),
+                       reinterpret_cast<__bf16*>(dx.data_ptr()), sd, nchunks,
+                       (const float*)nullptr, bsums.data_ptr<float>(),
+                       reinterpret_cast<const __bf16*>(xr.data_ptr()),
+                       prm.data_ptr<float>());
+  };
+  if (chunk == 64) LB(conv3d_spatial_kernel<8, 1, 32, 64, false, 2>);
+  else if (OWT == 32) LB(conv3d_spatial_kernel<32, 1, 32, 256, false, 2>);
+  else if (OWT == 16) LB(conv3d_spatial_kernel<16, 1, 32, 256, false, 2>);
+  else LB(conv3d_spatial_kernel<8, 1, 32, 256, false, 2>);
+  return {dx, bsums};
 }
 
 // ---------------------------------------------------------------------------
