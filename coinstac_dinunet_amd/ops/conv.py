@@ -26,6 +26,9 @@ _SPATIAL_DB = os.environ.get('COINN_SPATIAL_DB', '0') == '1'
 # (45.89 ms/step — LDS doubling costs co-residency more than intra-block
 # overlap buys). Kept compiled behind the env flag.
 _WGRAD_DB = os.environ.get('COINN_WGRAD_DB', '0') == '1'
+# BN-backward reduction inside the dgrad epilogue — measured worse on the
+# flagship (see _ConvBNFn.backward); kernels kept, off by default.
+_BNBWD_FUSE = os.environ.get('COINN_BNBWD_FUSE', '0') == '1'
 
 
 class _Conv3dFn(torch.autograd.Function):
@@ -147,13 +150,19 @@ class _ConvBNFn(torch.autograd.Function):
         wsub = (xb.size(4) + 1) // 2
         hsub = (xb.size(3) + 1) // 2
         sums = None
-        if (stride == 1 and xb.size(4) % 8 == 0 and go.size(1) >= 16
-                and xb.size(3) * xb.size(4) >= 64):
+        if (_BNBWD_FUSE and stride == 1 and xb.size(4) % 8 == 0
+                and go.size(1) >= 16 and xb.size(3) * xb.size(4) >= 64):
+            # MEASURED WORSE by ~0.5 ms/step on the flagship (r2): the
+            # epilogue x_raw gathers slow the dgrad more than the removed
+            # reduce pass saves — off by default, COINN_BNBWD_FUSE=1
             prm = torch.stack([mean_rstd[:, 0], mean_rstd[:, 1],
                                gamma.float(), beta.float()], 1).contiguous()
             dz, bsums = C.conv3d_dgrad_spatial_bnbwd(
                 go, wb, list(xb.shape), xb, prm)
             sums = bsums.sum(0)
+        elif (stride == 1 and xb.size(4) % 8 == 0 and go.size(1) >= 16
+                and xb.size(3) * xb.size(4) >= 64):
+            dz = C.conv3d_dgrad_spatial(go, wb, list(xb.shape))
         elif (stride == 2 and wsub % 8 == 0 and go.size(1) >= 32
                 and hsub * wsub >= 128):
             dz = C.conv3d_dgrad_s2_spatial(go, wb, list(xb.shape))
