@@ -175,6 +175,7 @@ void fused_adam(std::vector<torch::Tensor> params,
                                    torch::dtype(torch::kInt32)).clone();
   auto bmap = bmap_cpu.to(dev, /*non_blocking=*/true);
   const unsigned G = (unsigned)(bmap_v.size() / 2);
+  if (G == 0) return;  // all params empty
   hipLaunchKernelGGL(adam_mt_kernel, dim3(G), dim3(ELEM_BLOCK), 0,
                      current_stream(), ptrs.data_ptr<int64_t>(),
                      bmap.data_ptr<int>(), chunk, h);
