@@ -180,10 +180,9 @@ void fused_adam(std::vector<torch::Tensor> params,
     key ^= (size_t)v;
     key *= 1099511628211ull;
   };
-  for (int t = 0; t < T; ++t) {
-    mix(pc[t * 5 + 0]);
-    mix(pc[t * 5 + 4]);
-  }
+  // hash ALL pointers (p, g, m, v) + sizes: a fresh optimizer on the
+  // same params re-creates state tensors and must miss the cache
+  for (int64_t i = 0; i < (int64_t)T * 5; ++i) mix(pc[i]);
   static std::unordered_map<size_t, std::pair<torch::Tensor, torch::Tensor>>
       s_cache;
   auto it = s_cache.find(key);
