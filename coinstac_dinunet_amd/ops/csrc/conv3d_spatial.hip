@@ -49,7 +49,7 @@ struct SpDims {
 // Hash slices (blockIdx.x & 63) spread the atomics.
 template <int OWT, int STRIDE, int CTILE,
           int CHUNK = (STRIDE == 1 ? 256 : 128), bool FUSE_BN = false,
-          int SMODE = 0>
+          int SMODE = 0, int NCOLT = 32>
 __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
     const __bf16* __restrict__ in, const __bf16* __restrict__ wb,
     __bf16* __restrict__ out, SpDims sd, int64_t nchunks,
@@ -64,10 +64,11 @@ __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
   constexpr int MPW = CHUNK / 64;                   // m-frags per wave
   static_assert(MPW >= 1, "chunk too small");
   constexpr int KT_PAD = ((CTILE * 27 + 31) / 32) * 32;
+  constexpr int NCF = NCOLT / 16;  // ncol fragments per wave
   __shared__ __bf16 sX[CTILE][3][H2][W2];
   __shared__ unsigned short sKtab[KT_PAD + 8];
 
-  const int ncol0 = blockIdx.y * 32;
+  const int ncol0 = blockIdx.y * NCOLT;
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
   const int row = lane & 15, kg = lane >> 4;
@@ -96,11 +97,11 @@ __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
     sKtab[k] = off;
   }
 
-  f32x4 acc[MPW][2];
+  f32x4 acc[MPW][NCF];
 #pragma unroll
   for (int i = 0; i < MPW; ++i)
 #pragma unroll
-    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < NCF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const int64_t HW = (int64_t)sd.H * sd.W;
   const int64_t in_n = (int64_t)n * sd.KCH * sd.D * HW;
@@ -167,9 +168,9 @@ __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
             afrag[i][j] = slab[base + kt8[j]];
         }
       }
-      bf16x8 bfrag[2];
+      bf16x8 bfrag[NCF];
 #pragma unroll
-      for (int i = 0; i < 2; ++i) {
+      for (int i = 0; i < NCF; ++i) {
         const int col = ncol0 + i * 16 + row;
         const int64_t off =
             (int64_t)col * sd.Kpad + kbase_g + ks * 32 + kg * 8;
@@ -180,7 +181,7 @@ __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
 #pragma unroll
       for (int i = 0; i < MPW; ++i)
 #pragma unroll
-        for (int j = 0; j < 2; ++j)
+        for (int j = 0; j < NCF; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     }
@@ -191,11 +192,12 @@ __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
   const int64_t out_n = (int64_t)n * sd.NCOL * sd.TD * THW;
   const int ccol = lane & 15;
   const int crow0 = (lane >> 4) * 4;
-  float cs[2] = {0.f, 0.f}, css[2] = {0.f, 0.f};  // per-col-frag partials
+  float cs[4] = {}, css[4] = {};  // per-col-frag partials (NCF <= 4)
+  static_assert(NCF <= 4, "stats buffers sized for NCF <= 4");
 #pragma unroll
   for (int i = 0; i < MPW; ++i) {
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
+    for (int j = 0; j < NCF; ++j) {
       const int col = ncol0 + j * 16 + ccol;
       if (col >= sd.NCOL) continue;
       float p_mean = 0.f, p_rstd = 0.f, p_g = 0.f, p_b = 0.f;
@@ -236,24 +238,24 @@ __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
   if (SMODE != 0) {
     // lanes l, l+16, l+32, l+48 share ccol: fold over lane bits 4-5
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
+    for (int j = 0; j < NCF; ++j) {
       cs[j] += __shfl_xor(cs[j], 16);
       cs[j] += __shfl_xor(cs[j], 32);
       css[j] += __shfl_xor(css[j], 16);
       css[j] += __shfl_xor(css[j], 32);
     }
-    __shared__ float sred[4][2][16][2];
+    __shared__ float sred[4][NCF][16][2];
     if ((lane >> 4) == 0) {
 #pragma unroll
-      for (int j = 0; j < 2; ++j) {
+      for (int j = 0; j < NCF; ++j) {
         sred[wave][j][ccol][0] = cs[j];
         sred[wave][j][ccol][1] = css[j];
       }
     }
     __syncthreads();
-    // 64 (col, stat) pairs: threads 0..63 fold the 4 waves and emit one
-    // atomicAdd each into the hashed stats slice
-    if (tid < 64) {
+    // NCF*32 (col, stat) pairs: low threads fold the 4 waves and emit
+    // one atomicAdd each into the hashed stats slice
+    if (tid < NCF * 32) {
       const int j = tid >> 5;           // col fragment
       const int cc = (tid >> 1) & 15;   // ccol
       const int st = tid & 1;           // 0 = sum, 1 = sumsq
@@ -281,6 +283,7 @@ template <int OWT, int STRIDE, int CTILE,
 __global__ __launch_bounds__(256) void conv3d_spatial_db_kernel(
     const __bf16* __restrict__ in, const __bf16* __restrict__ wb,
     __bf16* __restrict__ out, SpDims sd, int64_t nchunks) {
+  constexpr int NCF = 2;  // fixed 32-col tiles in the DB variant
   constexpr int OHT = CHUNK / OWT;
   constexpr int IW = STRIDE * OWT;
   constexpr int W2 = IW + (STRIDE == 1 ? 4 : 2);
@@ -319,11 +322,11 @@ __global__ __launch_bounds__(256) void conv3d_spatial_db_kernel(
     sKtab[k] = off;
   }
 
-  f32x4 acc[MPW][2];
+  f32x4 acc[MPW][NCF];
 #pragma unroll
   for (int i = 0; i < MPW; ++i)
 #pragma unroll
-    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < NCF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const int64_t HW = (int64_t)sd.H * sd.W;
   const int64_t in_n = (int64_t)n * sd.KCH * sd.D * HW;
@@ -382,9 +385,9 @@ __global__ __launch_bounds__(256) void conv3d_spatial_db_kernel(
             afrag[i][j] = slab[base + kt8[j]];
         }
       }
-      bf16x8 bfrag[2];
+      bf16x8 bfrag[NCF];
 #pragma unroll
-      for (int i = 0; i < 2; ++i) {
+      for (int i = 0; i < NCF; ++i) {
         const int col = ncol0 + i * 16 + row;
         const int64_t off =
             (int64_t)col * sd.Kpad + kbase_g + ks * 32 + kg * 8;
@@ -395,7 +398,7 @@ __global__ __launch_bounds__(256) void conv3d_spatial_db_kernel(
 #pragma unroll
       for (int i = 0; i < MPW; ++i)
 #pragma unroll
-        for (int j = 0; j < 2; ++j)
+        for (int j = 0; j < NCF; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     }
@@ -455,16 +458,20 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
                            float* stats = nullptr) {
   int OWT = sd.TW % 32 == 0 ? 32 : (sd.TW % 16 == 0 ? 16 : 8);
   int chunk = stride == 1 ? 256 : 128;
-  // small images (d8-class): 64-position chunks keep the grid dense
+  int ncolt = 32;
+  // small images (d8-class): 64-position chunks keep the grid dense;
+  // wide-column instances halve the slab-staging redundancy there
+  // (PMC: 86% SQ_WAIT on the 32-col chunk-64 forms)
   if (sd.TH * sd.TW < chunk) {
     chunk = 64;
     OWT = 8;  // the chunk-64 instances are OWT=8
+    if (sd.NCOL >= 64 && ctile != 1) ncolt = 64;
   }
   int OHT = chunk / OWT;
   int wtiles = (sd.TW + OWT - 1) / OWT;
   int htiles = (sd.TH + OHT - 1) / OHT;
   int64_t nchunks = (int64_t)sd.N * sd.TD * htiles * wtiles;
-  dim3 grid((unsigned)nchunks, (sd.NCOL + 31) / 32);
+  dim3 grid((unsigned)nchunks, (sd.NCOL + ncolt - 1) / ncolt);
   auto s = current_stream();
   const __bf16* ip = reinterpret_cast<const __bf16*>(in.data_ptr());
   const __bf16* wp = reinterpret_cast<const __bf16*>(wb.data_ptr());
@@ -487,14 +494,18 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
   if (bn_ab != nullptr && stats != nullptr) {
     // fused normalize-on-load + epilogue-stats instances
     if (stride == 1) {
-      if (chunk == 64) LF(conv3d_spatial_kernel<8, 1, 32, 64, true, 1>);
+      if (chunk == 64 && ncolt == 64)
+        LF(conv3d_spatial_kernel<8, 1, 32, 64, true, 1, 64>);
+      else if (chunk == 64) LF(conv3d_spatial_kernel<8, 1, 32, 64, true, 1>);
       else if (OWT == 32)
         LF(conv3d_spatial_kernel<32, 1, 32, 256, true, 1>);
       else if (OWT == 16)
         LF(conv3d_spatial_kernel<16, 1, 32, 256, true, 1>);
       else LF(conv3d_spatial_kernel<8, 1, 32, 256, true, 1>);
     } else {
-      if (chunk == 64) LF(conv3d_spatial_kernel<8, 2, 16, 64, true, 1>);
+      if (chunk == 64 && ncolt == 64)
+        LF(conv3d_spatial_kernel<8, 2, 16, 64, true, 1, 64>);
+      else if (chunk == 64) LF(conv3d_spatial_kernel<8, 2, 16, 64, true, 1>);
       else if (OWT == 32)
         LF(conv3d_spatial_kernel<32, 2, 16, 128, true, 1>);
       else if (OWT == 16)
@@ -506,19 +517,27 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
   if (bn_ab != nullptr) {
     // fused normalize-on-load instances (default tilings only)
     if (stride == 1) {
-      if (chunk == 64) LF(conv3d_spatial_kernel<8, 1, 32, 64, true>);
+      if (chunk == 64 && ncolt == 64)
+        LF(conv3d_spatial_kernel<8, 1, 32, 64, true, 0, 64>);
+      else if (chunk == 64) LF(conv3d_spatial_kernel<8, 1, 32, 64, true>);
       else if (OWT == 32) LF(conv3d_spatial_kernel<32, 1, 32, 256, true>);
       else if (OWT == 16) LF(conv3d_spatial_kernel<16, 1, 32, 256, true>);
       else LF(conv3d_spatial_kernel<8, 1, 32, 256, true>);
     } else {
-      if (chunk == 64) LF(conv3d_spatial_kernel<8, 2, 16, 64, true>);
+      if (chunk == 64 && ncolt == 64)
+        LF(conv3d_spatial_kernel<8, 2, 16, 64, true, 0, 64>);
+      else if (chunk == 64) LF(conv3d_spatial_kernel<8, 2, 16, 64, true>);
       else if (OWT == 32) LF(conv3d_spatial_kernel<32, 2, 16, 128, true>);
       else if (OWT == 16) LF(conv3d_spatial_kernel<16, 2, 16, 128, true>);
       else LF(conv3d_spatial_kernel<8, 2, 16, 128, true>);
     }
     return;
   }
-  if (stride == 1 && ctile == 16) {
+  if (stride == 1 && chunk == 64 && ncolt == 64 && ctile == 32) {
+    L(conv3d_spatial_kernel<8, 1, 32, 64, false, 0, 64>);
+  } else if (stride == 2 && chunk == 64 && ncolt == 64) {
+    L(conv3d_spatial_kernel<8, 2, 16, 64, false, 0, 64>);
+  } else if (stride == 1 && ctile == 16) {
     // experimental double-buffered CTILE=16 instances (ctile_opt=16)
     if (chunk == 64) LDB(conv3d_spatial_db_kernel<8, 1, 16, 64>);
     else if (OWT == 32) LDB(conv3d_spatial_db_kernel<32, 1, 16>);
