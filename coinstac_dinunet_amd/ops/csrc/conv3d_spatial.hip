@@ -192,8 +192,8 @@ __global__ __launch_bounds__(256) void conv3d_spatial_kernel(
   const int64_t out_n = (int64_t)n * sd.NCOL * sd.TD * THW;
   const int ccol = lane & 15;
   const int crow0 = (lane >> 4) * 4;
-  float cs[4] = {}, css[4] = {};  // per-col-frag partials (NCF <= 4)
-  static_assert(NCF <= 4, "stats buffers sized for NCF <= 4");
+  float cs[8] = {}, css[8] = {};  // per-col-frag partials (NCF <= 8)
+  static_assert(NCF <= 8, "stats buffers sized for NCF <= 8");
 #pragma unroll
   for (int i = 0; i < MPW; ++i) {
 #pragma unroll
@@ -465,7 +465,9 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
   if (sd.TH * sd.TW < chunk) {
     chunk = 64;
     OWT = 8;  // the chunk-64 instances are OWT=8
-    if (sd.NCOL >= 64 && ctile != 1) ncolt = 64;
+    // swept on MI355X (r2): 128-col instances won at the deep layers
+    // (37.7 vs 38.2 ms/step flagship) — slab re-reads drop 4x vs 32-col
+    if (sd.NCOL >= 64 && ctile != 1) ncolt = sd.NCOL >= 128 ? 128 : 64;
   }
   int OHT = chunk / OWT;
   int wtiles = (sd.TW + OWT - 1) / OWT;
@@ -494,7 +496,9 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
   if (bn_ab != nullptr && stats != nullptr) {
     // fused normalize-on-load + epilogue-stats instances
     if (stride == 1) {
-      if (chunk == 64 && ncolt == 64)
+      if (chunk == 64 && ncolt == 128)
+        LF(conv3d_spatial_kernel<8, 1, 32, 64, true, 1, 128>);
+      else if (chunk == 64 && ncolt == 64)
         LF(conv3d_spatial_kernel<8, 1, 32, 64, true, 1, 64>);
       else if (chunk == 64) LF(conv3d_spatial_kernel<8, 1, 32, 64, true, 1>);
       else if (OWT == 32)
@@ -503,7 +507,9 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
         LF(conv3d_spatial_kernel<16, 1, 32, 256, true, 1>);
       else LF(conv3d_spatial_kernel<8, 1, 32, 256, true, 1>);
     } else {
-      if (chunk == 64 && ncolt == 64)
+      if (chunk == 64 && ncolt == 128)
+        LF(conv3d_spatial_kernel<8, 2, 16, 64, true, 1, 128>);
+      else if (chunk == 64 && ncolt == 64)
         LF(conv3d_spatial_kernel<8, 2, 16, 64, true, 1, 64>);
       else if (chunk == 64) LF(conv3d_spatial_kernel<8, 2, 16, 64, true, 1>);
       else if (OWT == 32)
@@ -517,14 +523,18 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
   if (bn_ab != nullptr) {
     // fused normalize-on-load instances (default tilings only)
     if (stride == 1) {
-      if (chunk == 64 && ncolt == 64)
+      if (chunk == 64 && ncolt == 128)
+        LF(conv3d_spatial_kernel<8, 1, 32, 64, true, 0, 128>);
+      else if (chunk == 64 && ncolt == 64)
         LF(conv3d_spatial_kernel<8, 1, 32, 64, true, 0, 64>);
       else if (chunk == 64) LF(conv3d_spatial_kernel<8, 1, 32, 64, true>);
       else if (OWT == 32) LF(conv3d_spatial_kernel<32, 1, 32, 256, true>);
       else if (OWT == 16) LF(conv3d_spatial_kernel<16, 1, 32, 256, true>);
       else LF(conv3d_spatial_kernel<8, 1, 32, 256, true>);
     } else {
-      if (chunk == 64 && ncolt == 64)
+      if (chunk == 64 && ncolt == 128)
+        LF(conv3d_spatial_kernel<8, 2, 16, 64, true, 0, 128>);
+      else if (chunk == 64 && ncolt == 64)
         LF(conv3d_spatial_kernel<8, 2, 16, 64, true, 0, 64>);
       else if (chunk == 64) LF(conv3d_spatial_kernel<8, 2, 16, 64, true>);
       else if (OWT == 32) LF(conv3d_spatial_kernel<32, 2, 16, 128, true>);
@@ -533,7 +543,11 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
     }
     return;
   }
-  if (stride == 1 && chunk == 64 && ncolt == 64 && ctile == 32) {
+  if (stride == 1 && chunk == 64 && ncolt == 128 && ctile == 32) {
+    L(conv3d_spatial_kernel<8, 1, 32, 64, false, 0, 128>);
+  } else if (stride == 2 && chunk == 64 && ncolt == 128) {
+    L(conv3d_spatial_kernel<8, 2, 16, 64, false, 0, 128>);
+  } else if (stride == 1 && chunk == 64 && ncolt == 64 && ctile == 32) {
     L(conv3d_spatial_kernel<8, 1, 32, 64, false, 0, 64>);
   } else if (stride == 2 && chunk == 64 && ncolt == 64) {
     L(conv3d_spatial_kernel<8, 2, 16, 64, false, 0, 64>);
