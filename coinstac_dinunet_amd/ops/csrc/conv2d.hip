@@ -847,7 +847,8 @@ static void launch_spatial2(torch::Tensor in, torch::Tensor wb,
   if (sd.TH * sd.TW < chunk) {
     chunk = 64;
     OWT = 8;
-    if (sd.NCOL >= 64) ncolt = 64;  // small-image wide-column instances
+    // wide columns cut slab re-reads; 128 won the 3D sweep at deep layers
+    if (sd.NCOL >= 64) ncolt = sd.NCOL >= 128 ? 128 : 64;
   }
   int OHT = chunk / OWT;
   int wtiles = (sd.TW + OWT - 1) / OWT;
@@ -864,14 +865,18 @@ static void launch_spatial2(torch::Tensor in, torch::Tensor wb,
   };
   if (bn_ab != nullptr) {
     if (stride == 1) {
-      if (chunk == 64 && ncolt == 64)
+      if (chunk == 64 && ncolt == 128)
+        L(conv2d_spatial_kernel<8, 1, 32, 64, true, 128>);
+      else if (chunk == 64 && ncolt == 64)
         L(conv2d_spatial_kernel<8, 1, 32, 64, true, 64>);
       else if (chunk == 64) L(conv2d_spatial_kernel<8, 1, 32, 64, true>);
       else if (OWT == 32) L(conv2d_spatial_kernel<32, 1, 32, 256, true>);
       else if (OWT == 16) L(conv2d_spatial_kernel<16, 1, 32, 256, true>);
       else L(conv2d_spatial_kernel<8, 1, 32, 256, true>);
     } else {
-      if (chunk == 64 && ncolt == 64)
+      if (chunk == 64 && ncolt == 128)
+        L(conv2d_spatial_kernel<8, 2, 16, 64, true, 128>);
+      else if (chunk == 64 && ncolt == 64)
         L(conv2d_spatial_kernel<8, 2, 16, 64, true, 64>);
       else if (chunk == 64) L(conv2d_spatial_kernel<8, 2, 16, 64, true>);
       else if (OWT == 32) L(conv2d_spatial_kernel<32, 2, 16, 128, true>);
@@ -879,14 +884,18 @@ static void launch_spatial2(torch::Tensor in, torch::Tensor wb,
       else L(conv2d_spatial_kernel<8, 2, 16, 128, true>);
     }
   } else if (stride == 1) {
-    if (chunk == 64 && ncolt == 64)
+    if (chunk == 64 && ncolt == 128)
+      L(conv2d_spatial_kernel<8, 1, 32, 64, false, 128>);
+    else if (chunk == 64 && ncolt == 64)
       L(conv2d_spatial_kernel<8, 1, 32, 64, false, 64>);
     else if (chunk == 64) L(conv2d_spatial_kernel<8, 1, 32, 64>);
     else if (OWT == 32) L(conv2d_spatial_kernel<32, 1, 32, 256>);
     else if (OWT == 16) L(conv2d_spatial_kernel<16, 1, 32, 256>);
     else L(conv2d_spatial_kernel<8, 1, 32, 256>);
   } else {
-    if (chunk == 64 && ncolt == 64)
+    if (chunk == 64 && ncolt == 128)
+      L(conv2d_spatial_kernel<8, 2, 16, 64, false, 128>);
+    else if (chunk == 64 && ncolt == 64)
       L(conv2d_spatial_kernel<8, 2, 16, 64, false, 64>);
     else if (chunk == 64) L(conv2d_spatial_kernel<8, 2, 16, 64>);
     else if (OWT == 32) L(conv2d_spatial_kernel<32, 2, 16, 128>);
