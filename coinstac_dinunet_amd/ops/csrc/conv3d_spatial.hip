@@ -459,6 +459,10 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
   int OWT = sd.TW % 32 == 0 ? 32 : (sd.TW % 16 == 0 ? 16 : 8);
   int chunk = stride == 1 ? 256 : 128;
   int ncolt = 32;
+  // (wide columns on the chunk-256 main instances were tried and
+  // produced wrong results at OWT=32 — acc[4][4] f32x4 = 256 VGPRs,
+  // past the occupancy-bound register budget; a (2x2)-wave remap would
+  // be needed. chunk-64 instances below use them safely at MPW=1.)
   // small images (d8-class): 64-position chunks keep the grid dense;
   // wide-column instances halve the slab-staging redundancy there
   // (PMC: 86% SQ_WAIT on the 32-col chunk-64 forms)
