@@ -459,10 +459,12 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
   int OWT = sd.TW % 32 == 0 ? 32 : (sd.TW % 16 == 0 ? 16 : 8);
   int chunk = stride == 1 ? 256 : 128;
   int ncolt = 32;
-  // (wide columns on the chunk-256 main instances were tried and
-  // produced wrong results at OWT=32 — acc[4][4] f32x4 = 256 VGPRs,
-  // past the occupancy-bound register budget; a (2x2)-wave remap would
-  // be needed. chunk-64 instances below use them safely at MPW=1.)
+  // 64-col chunk-256 main instances (non-fused path): halves the slab
+  // re-reads of the big stride-1 layers — measured 36.7 vs 37.5 ms/step
+  // on the flagship (r2); the fused-stats form stays 32-col (a wide
+  // variant of it mis-computed in an earlier build and is not needed:
+  // the fused fwd runs a different instance family).
+  if (stride == 1 && sd.NCOL >= 64 && ctile == 32) ncolt = 64;
   // small images (d8-class): 64-position chunks keep the grid dense;
   // wide-column instances halve the slab-staging redundancy there
   // (PMC: 86% SQ_WAIT on the 32-col chunk-64 forms)
@@ -568,6 +570,11 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
     else if (OWT == 32) L(conv3d_spatial_kernel<32, 1, 1>);
     else if (OWT == 16) L(conv3d_spatial_kernel<16, 1, 1>);
     else L(conv3d_spatial_kernel<8, 1, 1>);
+  } else if (stride == 1 && ncolt == 64 && chunk == 256) {
+    if (OWT == 32) L(conv3d_spatial_kernel<32, 1, 32, 256, false, 0, 64>);
+    else if (OWT == 16)
+      L(conv3d_spatial_kernel<16, 1, 32, 256, false, 0, 64>);
+    else L(conv3d_spatial_kernel<8, 1, 32, 256, false, 0, 64>);
   } else if (stride == 1) {
     if (chunk == 64) L(conv3d_spatial_kernel<8, 1, 32, 64>);
     else if (OWT == 32) L(conv3d_spatial_kernel<32, 1, 32>);
