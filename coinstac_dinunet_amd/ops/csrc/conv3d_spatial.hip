@@ -465,6 +465,10 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
   // variant of it mis-computed in an earlier build and is not needed:
   // the fused fwd runs a different instance family).
   if (stride == 1 && sd.NCOL >= 64 && ctile == 32) ncolt = 64;
+  // fused-stats wide form: numerically validated but measured SLOWER
+  // (37.3 vs 36.7 ms/step — the stats reduction on 4 col-fragments adds
+  // register pressure the fused fwd cannot afford); off by default.
+  const bool wide_fused = getenv("COINN_WIDE256F") != nullptr;
   // small images (d8-class): 64-position chunks keep the grid dense;
   // wide-column instances halve the slab-staging redundancy there
   // (PMC: 86% SQ_WAIT on the 32-col chunk-64 forms)
@@ -507,8 +511,12 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
       else if (chunk == 64 && ncolt == 64)
         LF(conv3d_spatial_kernel<8, 1, 32, 64, true, 1, 64>);
       else if (chunk == 64) LF(conv3d_spatial_kernel<8, 1, 32, 64, true, 1>);
+      else if (OWT == 32 && wide_fused && ncolt == 64)
+        LF(conv3d_spatial_kernel<32, 1, 32, 256, true, 1, 64>);
       else if (OWT == 32)
         LF(conv3d_spatial_kernel<32, 1, 32, 256, true, 1>);
+      else if (OWT == 16 && wide_fused && ncolt == 64)
+        LF(conv3d_spatial_kernel<16, 1, 32, 256, true, 1, 64>);
       else if (OWT == 16)
         LF(conv3d_spatial_kernel<16, 1, 32, 256, true, 1>);
       else LF(conv3d_spatial_kernel<8, 1, 32, 256, true, 1>);
