@@ -849,6 +849,8 @@ static void launch_spatial2(torch::Tensor in, torch::Tensor wb,
     OWT = 8;
     // wide columns cut slab re-reads; 128 won the 3D sweep at deep layers
     if (sd.NCOL >= 64) ncolt = sd.NCOL >= 128 ? 128 : 64;
+  } else if (stride == 1 && sd.NCOL >= 64 && bn_ab == nullptr) {
+    ncolt = 64;  // non-fused chunk-256 wide (same win as 3D: dgrad path)
   }
   int OHT = chunk / OWT;
   int wtiles = (sd.TW + OWT - 1) / OWT;
@@ -889,8 +891,14 @@ static void launch_spatial2(torch::Tensor in, torch::Tensor wb,
     else if (chunk == 64 && ncolt == 64)
       L(conv2d_spatial_kernel<8, 1, 32, 64, false, 64>);
     else if (chunk == 64) L(conv2d_spatial_kernel<8, 1, 32, 64>);
+    else if (OWT == 32 && ncolt == 64)
+      L(conv2d_spatial_kernel<32, 1, 32, 256, false, 64>);
     else if (OWT == 32) L(conv2d_spatial_kernel<32, 1, 32, 256>);
+    else if (OWT == 16 && ncolt == 64)
+      L(conv2d_spatial_kernel<16, 1, 32, 256, false, 64>);
     else if (OWT == 16) L(conv2d_spatial_kernel<16, 1, 32, 256>);
+    else if (ncolt == 64)
+      L(conv2d_spatial_kernel<8, 1, 32, 256, false, 64>);
     else L(conv2d_spatial_kernel<8, 1, 32, 256>);
   } else {
     if (chunk == 64 && ncolt == 128)
