@@ -465,6 +465,9 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
   // variant of it mis-computed in an earlier build and is not needed:
   // the fused fwd runs a different instance family).
   if (stride == 1 && sd.NCOL >= 64 && ctile == 32) ncolt = 64;
+  // chunk-512/CTILE-16 instances for thin-channel stride-1 layers
+  // (L2-class): caller prepped wb with ctile=16 and passes ctile=165
+  const bool c512 = (ctile == 165);
   // fused-stats wide form: numerically validated but measured SLOWER
   // (37.3 vs 36.7 ms/step — the stats reduction on 4 col-fragments adds
   // register pressure the fused fwd cannot afford); off by default.
@@ -578,6 +581,16 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
     else if (OWT == 32) L(conv3d_spatial_kernel<32, 1, 1>);
     else if (OWT == 16) L(conv3d_spatial_kernel<16, 1, 1>);
     else L(conv3d_spatial_kernel<8, 1, 1>);
+  } else if (c512 && stride == 1) {
+    // grid must be recomputed for the larger chunk
+    {
+      int OHT5 = 512 / 32;
+      int ht5 = (sd.TH + OHT5 - 1) / OHT5;
+      int wt5 = (sd.TW + 31) / 32;
+      nchunks = (int64_t)sd.N * sd.TD * ht5 * wt5;
+      grid = dim3((unsigned)nchunks, (sd.NCOL + 31) / 32);
+    }
+    L(conv3d_spatial_kernel<32, 1, 16, 512>);
   } else if (stride == 1 && ncolt == 64 && chunk == 256) {
     if (OWT == 32) L(conv3d_spatial_kernel<32, 1, 32, 256, false, 0, 64>);
     else if (OWT == 16)
@@ -683,10 +696,14 @@ torch::Tensor conv3d_dgrad_spatial(torch::Tensor go, torch::Tensor w,
   int Cout = (int)wc.size(0), Cin = (int)wc.size(1);
   auto wf = wc.reshape({Cout, Cin, 27}).flip(-1).permute({1, 0, 2})
                 .reshape({Cin, (int64_t)Cout * 27}).contiguous();
-  auto wb = prep_wb(wf, Cout, 32);
+  // thin-channel large-plane layers: the chunk-512/CTILE-16 instance
+  // doubles the m amortizing each staged slab (A/B: COINN_C512)
+  const bool c512 = getenv("COINN_C512") && sd.KCH <= 32 &&
+                    sd.TW % 32 == 0 && sd.TH * sd.TW >= 512;
+  auto wb = prep_wb(wf, Cout, c512 ? 16 : 32);
   sd.Kpad = (int)wb.size(1);
   auto dx = torch::empty(in_shape, g.options());
-  launch_spatial(g, wb, dx, sd, 1);
+  launch_spatial(g, wb, dx, sd, 1, c512 ? 165 : 0);
   return dx;
 }
 
