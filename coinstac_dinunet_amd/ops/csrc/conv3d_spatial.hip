@@ -697,8 +697,10 @@ torch::Tensor conv3d_dgrad_spatial(torch::Tensor go, torch::Tensor w,
   auto wf = wc.reshape({Cout, Cin, 27}).flip(-1).permute({1, 0, 2})
                 .reshape({Cin, (int64_t)Cout * 27}).contiguous();
   // thin-channel large-plane layers: the chunk-512/CTILE-16 instance
-  // doubles the m amortizing each staged slab (A/B: COINN_C512)
-  const bool c512 = getenv("COINN_C512") && sd.KCH <= 32 &&
+  // doubles the m amortizing each staged slab (measured 36.2 vs 36.6
+  // ms/step on the flagship, r2; COINN_C512=0 reverts)
+  const char* c512_env = getenv("COINN_C512");
+  const bool c512 = (!c512_env || c512_env[0] != '0') && sd.KCH <= 32 &&
                     sd.TW % 32 == 0 && sd.TH * sd.TW >= 512;
   auto wb = prep_wb(wf, Cout, c512 ? 16 : 32);
   sd.Kpad = (int)wb.size(1);
