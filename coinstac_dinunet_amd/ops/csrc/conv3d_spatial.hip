@@ -508,6 +508,15 @@ static void launch_spatial(torch::Tensor in, torch::Tensor wb,
   };
   if (bn_ab != nullptr && stats != nullptr) {
     // fused normalize-on-load + epilogue-stats instances
+    if (c512 && stride == 1) {
+      int OHT5 = 512 / 32;
+      int ht5 = (sd.TH + OHT5 - 1) / OHT5;
+      int wt5 = (sd.TW + 31) / 32;
+      nchunks = (int64_t)sd.N * sd.TD * ht5 * wt5;
+      grid = dim3((unsigned)nchunks, (sd.NCOL + 31) / 32);
+      LF(conv3d_spatial_kernel<32, 1, 16, 512, true, 1>);
+      return;
+    }
     if (stride == 1) {
       if (chunk == 64 && ncolt == 128)
         LF(conv3d_spatial_kernel<8, 1, 32, 64, true, 1, 128>);
@@ -670,6 +679,10 @@ std::vector<torch::Tensor> conv3d_fwd_spatial_stats(
   sd.TH = (sd.H + 2 - 3) / (int)stride + 1;
   sd.TW = (sd.W + 2 - 3) / (int)stride + 1;
   int ctile = stride == 1 ? 32 : 16;
+  const char* c5e = getenv("COINN_C512F");
+  const bool c512 = c5e && c5e[0] != '0' && stride == 1 && sd.KCH <= 32 &&
+                    sd.TW % 32 == 0 && sd.TH * sd.TW >= 512;
+  if (c512) ctile = 16;
   auto wb = prep_wb(wc.reshape({sd.NCOL, (int64_t)sd.KCH * 27}), sd.KCH,
                     ctile);
   sd.Kpad = (int)wb.size(1);
@@ -677,8 +690,8 @@ std::vector<torch::Tensor> conv3d_fwd_spatial_stats(
                           xc.options());
   auto stats = torch::zeros({64, sd.NCOL, 2},
                             xc.options().dtype(torch::kFloat32));
-  launch_spatial(xc, wb, out, sd, (int)stride, ctile, ab.data_ptr<float>(),
-                 stats.data_ptr<float>());
+  launch_spatial(xc, wb, out, sd, (int)stride, c512 ? 165 : ctile,
+                 ab.data_ptr<float>(), stats.data_ptr<float>());
   return {out, stats};
 }
 
