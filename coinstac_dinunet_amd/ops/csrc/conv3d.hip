@@ -1230,19 +1230,10 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
   }
   if ((cd.OW % 8) == 0 && cd.Cin >= 16 && cd.OH * cd.OW >= 64 &&
       (stride == 1 || stride == 2)) {
-    // tap-reuse path (stride-templated). COINN_WG256: OWT=16/CHUNK=256
-    // instances — twice the m amortizing each staged slab (60 KB LDS).
-    static const bool wg256 = []() {
-      const char* e = getenv("COINN_WG256");
-      return e && e[0] != '0';
-    }();
+    // tap-reuse path (stride-templated)
     int OWT = cd.OW % 32 == 0 ? 32 : (cd.OW % 16 == 0 ? 16 : 8);
     int chunk = 128;
-    if (wg256 && stride == 1 && variant == 0 && cd.OW % 16 == 0 &&
-        cd.OH * cd.OW >= 256) {
-      OWT = 16;
-      chunk = 256;
-    } else if (cd.OH * cd.OW < 128) {
+    if (cd.OH * cd.OW < 128) {
       chunk = 64;
       OWT = 8;
     }
@@ -1283,10 +1274,7 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
     if (sliced) {
       if (fuse) {
         if (stride == 1) {
-          if (chunk == 256)
-            L(conv3d_wgrad_s1_kernel<16, 1, 256, true, true>);
-          else if (chunk == 64)
-            L(conv3d_wgrad_s1_kernel<8, 1, 64, true, true>);
+          if (chunk == 64) L(conv3d_wgrad_s1_kernel<8, 1, 64, true, true>);
           else if (OWT == 32)
             L(conv3d_wgrad_s1_kernel<32, 1, 128, true, true>);
           else if (OWT == 16)
@@ -1301,8 +1289,7 @@ torch::Tensor conv3d_wgrad(torch::Tensor x, torch::Tensor go,
           else L(conv3d_wgrad_s1_kernel<8, 2, 128, true, true>);
         }
       } else if (stride == 1) {
-        if (chunk == 256) L(conv3d_wgrad_s1_kernel<16, 1, 256, false, true>);
-        else if (chunk == 64) L(conv3d_wgrad_s1_kernel<8, 1, 64, false, true>);
+        if (chunk == 64) L(conv3d_wgrad_s1_kernel<8, 1, 64, false, true>);
         else if (OWT == 32)
           L(conv3d_wgrad_s1_kernel<32, 1, 128, false, true>);
         else if (OWT == 16)
