@@ -134,3 +134,22 @@ def test_rccl_cluster_three_ranks_gloo(tmp_path):
     for r in (1, 2):
         wr = np.load(os.path.join(result_dir, f'weights_rank{r}.npy'))
         np.testing.assert_allclose(w0, wr, rtol=1e-5, atol=1e-6)
+
+
+def test_adaptive_bucket_sizing():
+    """Default bucket size = arena/4 clamped to [2, 50] MB (r2: a fixed
+    50 MB put every BASELINE model in ONE bucket — no overlap)."""
+    import torch
+    from coinstac_dinunet_amd.parallel.engine import FlatGradBuffer
+
+    # 14 MB arena (VBM-like) -> ~4 buckets
+    params = [torch.nn.Parameter(torch.zeros(875_000)) for _ in range(4)]
+    buf = FlatGradBuffer(params)
+    assert 3 <= len(buf.buckets) <= 5, len(buf.buckets)
+    # tiny arena -> single bucket at the 2 MB floor
+    small = [torch.nn.Parameter(torch.zeros(1000))]
+    buf2 = FlatGradBuffer(small)
+    assert len(buf2.buckets) == 1
+    # explicit size still honored
+    buf3 = FlatGradBuffer(params, bucket_bytes=64 * 1024 * 1024)
+    assert len(buf3.buckets) == 1
