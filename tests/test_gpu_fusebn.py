@@ -246,3 +246,36 @@ def test_dgrad_bnbwd_sums_match_standalone_reduce(dev):
     torch.testing.assert_close(db_p, db_r, rtol=1e-3, atol=1e-1)
     torch.testing.assert_close(dx_p.float(), dx_r.float(),
                                rtol=2e-2, atol=2e-2)
+
+
+def test_conv_bn3d_eval_uses_running_stats(dev):
+    """Eval-mode fused pair must normalize with RUNNING stats and match
+    the composed eval path."""
+    bn, conv = _pair(32, 32, 1, dev, 57)
+    x = torch.randn(2, 32, 8, 8, 8, device=dev, dtype=torch.bfloat16)
+    with torch.no_grad():
+        bn.running_mean.add_(torch.randn(32, device=dev) * 0.1)
+        bn.running_var.mul_(0).add_(torch.rand(32, device=dev) + 0.5)
+    bn.eval(), conv.eval()
+    with torch.no_grad():
+        fused = conv_bn3d(x, bn, conv)
+        composed = conv(bn(x))
+    torch.testing.assert_close(fused.float(), composed.float(),
+                               rtol=3e-2, atol=3e-2 * (32 * 27) ** 0.5 * 0.3)
+
+
+def test_chain_attached_stats_agree_with_bn3d_stats(dev):
+    """Stats attached by the fused chain must match a direct reduce of
+    the same activation."""
+    from coinstac_dinunet_amd.ops.conv import bn_stats_of
+    bn, conv = _pair(32, 32, 1, dev, 58)
+    bn.train(), conv.train()
+    x = torch.randn(2, 32, 16, 16, 16, device=dev, dtype=torch.bfloat16)
+    out = conv_bn3d(x, bn, conv)
+    assert getattr(out, '_coinn_bn_stats', None) is not None
+    m_a, v_a, _ = bn_stats_of(out.detach(), 1e-5)
+    stripped = out.detach().clone()  # clone drops the attribute
+    assert getattr(stripped, '_coinn_bn_stats', None) is None
+    m_d, v_d, _ = bn_stats_of(stripped, 1e-5)
+    torch.testing.assert_close(m_a, m_d, rtol=1e-3, atol=1e-3)
+    torch.testing.assert_close(v_a, v_d, rtol=1e-3, atol=1e-3)

@@ -231,3 +231,34 @@ def test_gram_schmidt_kernel(dev):
     gram = m.t() @ m
     torch.testing.assert_close(gram, torch.eye(4, device=dev),
                                rtol=1e-3, atol=1e-3)
+
+
+def test_forced_1rank_allreduce_is_identity(dev):
+    """COINN_FORCE_ALLREDUCE runs the real RCCL collectives at world 1;
+    an all-reduce over one rank must leave gradients unchanged."""
+    import os
+    import torch.distributed as tdist
+    from coinstac_dinunet_amd.parallel.engine import (FlatGradBuffer,
+                                                      init_distributed)
+    init_distributed()
+    os.environ['COINN_FORCE_ALLREDUCE'] = '1'
+    try:
+        net = torch.nn.Sequential(torch.nn.Linear(64, 64),
+                                  torch.nn.Linear(64, 8)).to(dev)
+        comm = torch.cuda.Stream()
+        buf = FlatGradBuffer(net.parameters(), bucket_bytes=4096,
+                             world_size=1, comm_stream=comm)
+        assert buf.force_collectives and len(buf.buckets) >= 2
+        x = torch.randn(16, 64, device=dev)
+        buf.zero_()
+        buf.begin_round(sync=False)
+        net(x).sum().backward()
+        ref = buf.flat.detach().clone()
+        buf.zero_()
+        buf.begin_round(sync=True)
+        net(x).sum().backward()
+        buf.finish_round()
+        torch.cuda.synchronize()
+        torch.testing.assert_close(buf.flat, ref)
+    finally:
+        os.environ.pop('COINN_FORCE_ALLREDUCE', None)
