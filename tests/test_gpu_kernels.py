@@ -246,7 +246,7 @@ def test_forced_1rank_allreduce_is_identity(dev):
         net = torch.nn.Sequential(torch.nn.Linear(64, 64),
                                   torch.nn.Linear(64, 8)).to(dev)
         comm = torch.cuda.Stream()
-        buf = FlatGradBuffer(net.parameters(), bucket_bytes=4096,
+        buf = FlatGradBuffer(net.parameters(), bucket_bytes=2048,
                              world_size=1, comm_stream=comm)
         assert buf.force_collectives and len(buf.buckets) >= 2
         x = torch.randn(16, 64, device=dev)
