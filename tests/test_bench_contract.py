@@ -76,3 +76,21 @@ def test_bench_torchrun_eight_ranks():
     assert out['n_gpus'] == 8
     assert out['config']['parallelism'] == 'dsgd-dp8'
     assert out['config']['global_batch'] == 16
+
+
+def test_bench_stock_model_builders():
+    """--stock builders (library-op VBM/ResNet) must keep constructing:
+    they are the measured comparison baseline."""
+    import argparse
+    import torch
+    import bench
+
+    for model in ['vbm', 'resnet18']:
+        args = argparse.Namespace(model=model, stock=True, batch=2, vol=8)
+        net, data, labels = bench.build_model(args, torch.device('cpu'))
+        out = net(data)
+        assert out.shape[0] == 2
+        # stock means NO Ops modules anywhere
+        from coinstac_dinunet_amd.ops.conv import OpsConv2d, OpsConv3d
+        for m in net.modules():
+            assert not isinstance(m, (OpsConv2d, OpsConv3d)), type(m)
