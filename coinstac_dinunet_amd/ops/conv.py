@@ -1,8 +1,14 @@
-"""Conv3d module on the hand-written implicit-GEMM MFMA kernels (K1).
+"""Conv modules on the hand-written implicit-GEMM MFMA kernels (K1).
 
-GPU path (bf16): ops._hip_ops.conv3d_{fwd,dgrad,wgrad}; CPU falls back to
-torch.nn.functional.conv3d. Restricted to the VBM workload's conv family:
-3x3x3 kernels, padding 1, stride 1 or 2, NCDHW.
+GPU paths (bf16):
+  - OpsConv3d: 3x3x3/pad-1/stride-{1,2} NCDHW on the spatial-slab/igemm
+    kernels; 1x1x1 on the batched-MFMA pointwise kernels.
+  - OpsConv2d: 3x3/pad-1/stride-{1,2} (+ the 7x7 stem, fwd+wgrad) on the
+    2D family; 1x1 (incl. stride-2 downsamples) on the pointwise kernels.
+  - conv_bn3d / conv_bn2d: the PREVIOUS block's BN(+ReLU) folded into the
+    conv's input staging (normalize-on-load), with the conv epilogue
+    emitting the next BN's statistics.
+CPU falls back to torch.nn.functional convs.
 """
 import os
 

@@ -233,9 +233,8 @@ std::vector<torch::Tensor> power_iter_bc(torch::Tensor B, torch::Tensor C,
   return {Bf, Cf, ncomp};
 }
 
-// K11 helper — per-component row sums: out[r] = sum_i M[i,r] over rows of a
-// [n, r] factor (bias gradient of the reconstructed layer: gf.sum over n?
-// no — gf [out, r].sum(dim=1) => out[i] = sum_r gf[i, r]).
+// K11 helper — per-row sums of a [out, r] factor:
+// out[i] = sum_r M[i, r] (the reconstructed layer's bias gradient).
 __global__ void rowsum_kernel(const float* __restrict__ M,
                               float* __restrict__ out, int64_t rows,
                               int64_t cols) {
