@@ -133,3 +133,36 @@ def test_fused_adam_buckets_by_step():
         opt.step()                      # p1 at 2, p2 at 1: TWO buckets
     assert calls[0] == (1, 1)
     assert sorted(calls[1:]) == [(1, 1), (1, 2)], calls
+
+
+def test_auc_fallback_property_sweep():
+    """Randomized tie-heavy sweep: the sklearn-free AUC must track
+    sklearn within rounding for any tie structure."""
+    import unittest.mock as mock
+    import builtins
+    from sklearn.metrics import roc_curve, auc as _auc
+    from coinstac_dinunet_amd.metrics.metrics import AUCROCMetrics
+
+    rng = np.random.RandomState(7)
+    real_import = builtins.__import__
+
+    def no_sklearn(name, *a, **k):
+        if name.startswith('sklearn'):
+            raise ImportError('forced')
+        return real_import(name, *a, **k)
+
+    for trial in range(20):
+        n = rng.randint(5, 60)
+        # quantized probabilities force ties; ensure both classes present
+        p = np.round(rng.rand(n), 1).tolist()
+        y = rng.randint(0, 2, n)
+        if y.sum() in (0, n):
+            y[0] = 1 - y[0]
+        y = y.tolist()
+        fpr, tpr, _ = roc_curve(y, p)
+        expect = float(_auc(fpr, tpr))
+        m = AUCROCMetrics()
+        m.probabilities, m.labels = list(p), list(y)
+        with mock.patch.object(builtins, '__import__', no_sklearn):
+            got = m.auc
+        assert abs(got - expect) < 1e-4, (trial, got, expect)
